@@ -1,0 +1,455 @@
+// bind.cpp — pybind11 module `hippt._C`: scene assembly, CPU reference
+// renderer entry points, GPU upload + kernel launches, BVH builders.
+//
+// Python (hippt/scene/scene.py) parses scenes and calls the setters here;
+// the extension owns all device-side scene buffers (hipMalloc) while output
+// images live in torch tensors whose data_ptr() is passed in for kernels to
+// accumulate into (zero-copy handoff back to PyTorch-ROCm / RCCL).
+#include <pybind11/pybind11.h>
+#include <pybind11/numpy.h>
+#include <pybind11/stl.h>
+#include <cstring>
+#include <stdexcept>
+
+#include "core/scene_view.h"
+#include "core/integrator.h"
+#include "cpu/bvh_build.h"
+#include "hip/kernels.h"
+
+namespace py = pybind11;
+using namespace hippt;
+
+namespace hippt {
+void render_cpu(const SceneView& sv, float* accum, float* var,
+                int spp0, int nspp, uint32_t seed, int renderer, int n_threads);
+void render_lt_cpu(const SceneView& sv, float* accum, int spp0, int nspp, uint32_t seed,
+                   int spec_constraint, float caustic_scaling, int n_threads);
+}
+
+namespace {
+
+using farr = py::array_t<float, py::array::c_style | py::array::forcecast>;
+using iarr = py::array_t<int32_t, py::array::c_style | py::array::forcecast>;
+using uarr = py::array_t<uint32_t, py::array::c_style | py::array::forcecast>;
+
+#define HIP_OK(expr)                                                                   \
+    do {                                                                               \
+        int _e = (expr);                                                               \
+        if (_e != 0) throw std::runtime_error(std::string("HIP error ") +              \
+                                              std::to_string(_e) + " at " #expr);      \
+    } while (0)
+
+Vec3 to_vec3(const std::vector<float>& v) { return {v[0], v[1], v[2]}; }
+
+struct SceneHolder {
+    // ----- host data (kept alive / owned)
+    farr np_prims, np_attrs, np_nodes;
+    uarr np_prim_obj;
+    iarr np_objs;
+    std::vector<BsdfParams> bsdfs;
+    std::vector<EmitterParams> emitters;
+    std::vector<PhaseParams> phases;
+    std::vector<MediumParams> media;          // host pointers inside
+    std::vector<farr> media_density, media_temp;
+    std::vector<farr> tex_data;
+    std::vector<TexView> tex_host;
+    std::vector<int> emitter_prims;
+    std::vector<float> emitter_cdf;
+    Camera cam{};
+    MaxDepthParams md{};
+    int env_emitter = -1;
+    int cam_medium = -1;
+
+    // ----- device mirrors
+    bool has_dev = false;
+    int device_id = 0;
+    std::vector<void*> dev_bufs;              // everything hipMalloc'ed
+    std::vector<TexView> tex_dev;
+    std::vector<MediumParams> media_dev;
+    // offsets of updateable device arrays (hot reload)
+    void* dev_bsdfs = nullptr;
+    void* dev_emitters = nullptr;
+    void* dev_media = nullptr;
+
+    SceneView host_sv{};
+    SceneView dev_sv{};
+
+    ~SceneHolder() { release(); }
+
+    void release() {
+        for (void* p : dev_bufs) dev_free(p);
+        dev_bufs.clear();
+        has_dev = false;
+    }
+
+    void set_geometry(farr prims, farr attrs, uarr prim_obj, farr nodes) {
+        if (prims.ndim() != 2 || prims.shape(1) != 12) throw std::runtime_error("prims must be (n,12)");
+        if (attrs.ndim() != 2 || attrs.shape(1) != 16) throw std::runtime_error("attrs must be (n,16)");
+        if (nodes.ndim() != 2 || nodes.shape(1) != 8) throw std::runtime_error("nodes must be (m,8)");
+        np_prims = std::move(prims);
+        np_attrs = std::move(attrs);
+        np_prim_obj = std::move(prim_obj);
+        np_nodes = std::move(nodes);
+    }
+
+    void set_objects(iarr objs) {
+        if (objs.ndim() != 2 || objs.shape(1) != 8) throw std::runtime_error("objs must be (n,8)");
+        np_objs = std::move(objs);
+    }
+
+    int add_bsdf(int type, std::vector<float> kd, std::vector<float> ks, std::vector<float> kg,
+                 float ior, float extra0, float extra1, std::vector<int> tex) {
+        BsdfParams b = BsdfParams::make(type);
+        b.kd = Vec4(kd[0], kd[1], kd[2], kd.size() > 3 ? kd[3] : 0.f);
+        b.ks = Vec4(ks[0], ks[1], ks[2], ks.size() > 3 ? ks[3] : 0.f);
+        b.kg = Vec4(kg[0], kg[1], kg[2], kg.size() > 3 ? kg[3] : 0.f);
+        b.ior = ior; b.extra0 = extra0; b.extra1 = extra1;
+        for (size_t i = 0; i < tex.size() && i < 8; ++i) b.tex[i] = (int16_t)tex[i];
+        bsdfs.push_back(b);
+        return (int)bsdfs.size() - 1;
+    }
+
+    void update_bsdf(int i, int type, std::vector<float> kd, std::vector<float> ks,
+                     std::vector<float> kg, float ior, float extra0, float extra1,
+                     std::vector<int> tex) {
+        if (i < 0 || i >= (int)bsdfs.size()) throw std::runtime_error("bad bsdf index");
+        BsdfParams b = BsdfParams::make(type);
+        b.kd = Vec4(kd[0], kd[1], kd[2], kd.size() > 3 ? kd[3] : 0.f);
+        b.ks = Vec4(ks[0], ks[1], ks[2], ks.size() > 3 ? ks[3] : 0.f);
+        b.kg = Vec4(kg[0], kg[1], kg[2], kg.size() > 3 ? kg[3] : 0.f);
+        b.ior = ior; b.extra0 = extra0; b.extra1 = extra1;
+        for (size_t k = 0; k < tex.size() && k < 8; ++k) b.tex[k] = (int16_t)tex[k];
+        bsdfs[i] = b;
+        if (has_dev && dev_bsdfs)
+            HIP_OK(dev_upload((char*)dev_bsdfs + sizeof(BsdfParams) * i, &bsdfs[i], sizeof(BsdfParams)));
+    }
+
+    int add_emitter(int type, std::vector<float> emission, float scale, std::vector<float> aux,
+                    int obj_id, int tex_id, int prim_base, int prim_cnt, float inv_area) {
+        EmitterParams e{};
+        e.emission = Vec4(emission[0], emission[1], emission[2], scale);
+        e.aux = Vec4(aux[0], aux[1], aux[2], aux.size() > 3 ? aux[3] : 0.f);
+        e.type = type; e.obj_id = obj_id; e.tex_id = tex_id;
+        e.prim_base = prim_base; e.prim_cnt = prim_cnt; e.inv_area = inv_area;
+        emitters.push_back(e);
+        if (type == EM_ENVMAP) env_emitter = (int)emitters.size() - 1;
+        return (int)emitters.size() - 1;
+    }
+
+    void update_emitter(int i, std::vector<float> emission, float scale, std::vector<float> aux) {
+        if (i < 0 || i >= (int)emitters.size()) throw std::runtime_error("bad emitter index");
+        emitters[i].emission = Vec4(emission[0], emission[1], emission[2], scale);
+        emitters[i].aux = Vec4(aux[0], aux[1], aux[2], aux.size() > 3 ? aux[3] : 0.f);
+        if (has_dev && dev_emitters)
+            HIP_OK(dev_upload((char*)dev_emitters + sizeof(EmitterParams) * i, &emitters[i],
+                              sizeof(EmitterParams)));
+    }
+
+    void set_emitter_prims(iarr eprims, farr ecdf) {
+        emitter_prims.assign(eprims.data(), eprims.data() + eprims.size());
+        emitter_cdf.assign(ecdf.data(), ecdf.data() + ecdf.size());
+    }
+
+    int add_phase(int type, float g1, float g2, float wmix) {
+        phases.push_back({type, g1, g2, wmix});
+        return (int)phases.size() - 1;
+    }
+
+    int add_medium(int type, std::vector<float> sigma_a, std::vector<float> sigma_s,
+                   int phase_id, std::vector<float> grid_lo, std::vector<float> grid_hi,
+                   py::object density, py::object temperature,
+                   float scale, float emission_scale, float temp_scale) {
+        MediumParams m{};
+        m.sigma_a = Vec4(sigma_a[0], sigma_a[1], sigma_a[2], 0.f);
+        m.sigma_s = Vec4(sigma_s[0], sigma_s[1], sigma_s[2], 0.f);
+        m.type = type; m.phase_id = phase_id;
+        m.scale = scale; m.emission_scale = emission_scale; m.temp_scale = temp_scale;
+        m.density = nullptr; m.temperature = nullptr;
+        m.nx = m.ny = m.nz = 0;
+        if (type == MED_GRID) {
+            farr d = density.cast<farr>();
+            if (d.ndim() != 3) throw std::runtime_error("density must be (nz,ny,nx)");
+            media_density.push_back(d);
+            m.nz = (int)d.shape(0); m.ny = (int)d.shape(1); m.nx = (int)d.shape(2);
+            m.density = media_density.back().data();
+            float mx = 0.f; double sum = 0.0;
+            const float* dd = m.density;
+            size_t n = (size_t)m.nx * m.ny * m.nz;
+            for (size_t i = 0; i < n; ++i) { mx = fmaxf(mx, dd[i]); sum += dd[i]; }
+            m.majorant = mx * scale;
+            m.avg_density = (float)(sum / std::max<size_t>(n, 1)) * scale;
+            Vec3 lo = to_vec3(grid_lo), hi = to_vec3(grid_hi);
+            m.grid_lo = Vec4(lo, 0.f);
+            Vec3 ext = hi - lo;
+            m.grid_inv_extent = Vec4(1.f / ext.x, 1.f / ext.y, 1.f / ext.z, 0.f);
+            if (!temperature.is_none()) {
+                farr t = temperature.cast<farr>();
+                media_temp.push_back(t);
+                m.temperature = media_temp.back().data();
+            }
+        }
+        media.push_back(m);
+        return (int)media.size() - 1;
+    }
+
+    void update_medium(int i, std::vector<float> sigma_a, std::vector<float> sigma_s,
+                       float scale, float emission_scale) {
+        if (i < 0 || i >= (int)media.size()) throw std::runtime_error("bad medium index");
+        MediumParams& m = media[i];
+        float old_scale = m.scale;
+        m.sigma_a = Vec4(sigma_a[0], sigma_a[1], sigma_a[2], 0.f);
+        m.sigma_s = Vec4(sigma_s[0], sigma_s[1], sigma_s[2], 0.f);
+        if (m.type == MED_GRID && old_scale > 0.f) {
+            m.majorant *= scale / old_scale;
+            m.avg_density *= scale / old_scale;
+        }
+        m.scale = scale; m.emission_scale = emission_scale;
+        if (has_dev && dev_media) {
+            MediumParams md2 = media_dev[i];
+            md2.sigma_a = m.sigma_a; md2.sigma_s = m.sigma_s;
+            md2.scale = m.scale; md2.majorant = m.majorant;
+            md2.avg_density = m.avg_density; md2.emission_scale = m.emission_scale;
+            media_dev[i] = md2;
+            HIP_OK(dev_upload((char*)dev_media + sizeof(MediumParams) * i, &media_dev[i],
+                              sizeof(MediumParams)));
+        }
+    }
+
+    int add_texture(farr rgba) {
+        if (rgba.ndim() != 3 || rgba.shape(2) != 4) throw std::runtime_error("texture must be (h,w,4)");
+        tex_data.push_back(std::move(rgba));
+        const farr& t = tex_data.back();
+        TexView v{};
+        v.tex_obj = 0;
+        v.data = t.data();
+        v.w = (int)t.shape(1);
+        v.h = (int)t.shape(0);
+        tex_host.push_back(v);
+        return (int)tex_host.size() - 1;
+    }
+
+    void set_camera(std::vector<float> pos, std::vector<float> R_rows, float focal,
+                    int w, int h, float aperture, float focal_dist, int ortho, float ortho_scale) {
+        cam.pos = to_vec3(pos);
+        cam.R = Mat3({R_rows[0], R_rows[1], R_rows[2]},
+                     {R_rows[3], R_rows[4], R_rows[5]},
+                     {R_rows[6], R_rows[7], R_rows[8]});
+        cam.focal = focal; cam.w = w; cam.h = h;
+        cam.aperture = aperture; cam.focal_dist = focal_dist;
+        cam.ortho = ortho; cam.ortho_scale = ortho_scale;
+    }
+
+    void set_depths(int max_depth, int max_diffuse, int max_specular, int max_transmit,
+                    int max_volume, float min_time, float max_time, int use_tof) {
+        md = {max_depth, max_diffuse, max_specular, max_transmit, max_volume,
+              min_time, max_time, use_tof};
+    }
+
+    void fill_common(SceneView& sv) {
+        sv.n_nodes = (int)np_nodes.shape(0);
+        sv.n_prims = (int)np_prims.shape(0);
+        sv.n_objs = np_objs.ndim() == 2 ? (int)np_objs.shape(0) : 0;
+        sv.n_bsdfs = (int)bsdfs.size();
+        sv.n_emitters = (int)emitters.size();
+        sv.n_textures = (int)tex_host.size();
+        sv.n_media = (int)media.size();
+        sv.env_emitter = env_emitter;
+        sv.cam_medium = cam_medium;
+        sv.cam = cam;
+        sv.md = md;
+    }
+
+    void finalize() {
+        fill_common(host_sv);
+        host_sv.nodes = (const BVHNode*)np_nodes.data();
+        host_sv.prims = (const Prim*)np_prims.data();
+        host_sv.attrs = (const PrimAttr*)np_attrs.data();
+        host_sv.prim_obj = np_prim_obj.data();
+        host_sv.objs = (const ObjInfo*)np_objs.data();
+        host_sv.bsdfs = bsdfs.data();
+        host_sv.emitters = emitters.data();
+        host_sv.emitter_prims = emitter_prims.data();
+        host_sv.emitter_cdf = emitter_cdf.data();
+        host_sv.textures = tex_host.data();
+        host_sv.media = media.data();
+        host_sv.phases = phases.data();
+    }
+
+    template <typename T>
+    T* upload_vec(const T* src, size_t count) {
+        if (count == 0) return nullptr;
+        void* p = nullptr;
+        HIP_OK(dev_malloc(&p, sizeof(T) * count));
+        dev_bufs.push_back(p);
+        HIP_OK(dev_upload(p, src, sizeof(T) * count));
+        return (T*)p;
+    }
+
+    void upload(int device) {
+        release();
+        device_id = device;
+        HIP_OK(dev_set_device(device));
+        finalize();
+        fill_common(dev_sv);
+        dev_sv.nodes = upload_vec((const BVHNode*)np_nodes.data(), np_nodes.shape(0));
+        dev_sv.prims = upload_vec((const Prim*)np_prims.data(), np_prims.shape(0));
+        dev_sv.attrs = upload_vec((const PrimAttr*)np_attrs.data(), np_attrs.shape(0));
+        dev_sv.prim_obj = upload_vec(np_prim_obj.data(), np_prim_obj.size());
+        dev_sv.objs = upload_vec((const ObjInfo*)np_objs.data(), np_objs.shape(0));
+        dev_sv.emitter_prims = upload_vec(emitter_prims.data(), emitter_prims.size());
+        dev_sv.emitter_cdf = upload_vec(emitter_cdf.data(), emitter_cdf.size());
+        // textures: RGBA32F rows in device global memory (gfx950 has no
+        // device texture units — software bilinear is the CDNA-native path)
+        tex_dev.clear();
+        for (size_t i = 0; i < tex_host.size(); ++i) {
+            TexView v = tex_host[i];
+            v.data = upload_vec(tex_data[i].data(), (size_t)v.w * v.h * 4);
+            tex_dev.push_back(v);
+        }
+        dev_sv.textures = upload_vec(tex_dev.data(), tex_dev.size());
+        // media: re-point grids at device copies
+        media_dev = media;
+        for (size_t i = 0; i < media.size(); ++i) {
+            if (media[i].density) {
+                size_t n = (size_t)media[i].nx * media[i].ny * media[i].nz;
+                media_dev[i].density = upload_vec(media[i].density, n);
+            }
+            if (media[i].temperature) {
+                size_t n = (size_t)media[i].nx * media[i].ny * media[i].nz;
+                media_dev[i].temperature = upload_vec(media[i].temperature, n);
+            }
+        }
+        dev_sv.media = upload_vec(media_dev.data(), media_dev.size());
+        dev_sv.phases = upload_vec(phases.data(), phases.size());
+        dev_sv.bsdfs = upload_vec(bsdfs.data(), bsdfs.size());
+        dev_sv.emitters = upload_vec(emitters.data(), emitters.size());
+        dev_bsdfs = (void*)dev_sv.bsdfs;
+        dev_emitters = (void*)dev_sv.emitters;
+        dev_media = (void*)dev_sv.media;
+        has_dev = true;
+    }
+
+    void render_host(farr accum, py::object var, int spp0, int nspp, uint32_t seed,
+                     int renderer, int spec_constraint, float caustic_scaling, int n_threads) {
+        finalize();
+        float* vp = nullptr;
+        farr var_arr;
+        if (!var.is_none()) { var_arr = var.cast<farr>(); vp = var_arr.mutable_data(); }
+        py::gil_scoped_release rel;
+        if (renderer == R_LIGHT_TRACE)
+            render_lt_cpu(host_sv, accum.mutable_data(), spp0, nspp, seed,
+                          spec_constraint, caustic_scaling, n_threads);
+        else
+            render_cpu(host_sv, accum.mutable_data(), vp, spp0, nspp, seed, renderer, n_threads);
+    }
+
+    void render_device(uintptr_t accum_ptr, uintptr_t var_ptr, int spp0, int nspp,
+                       uint32_t seed, int renderer, int spec_constraint,
+                       float caustic_scaling, uintptr_t stream) {
+        if (!has_dev) throw std::runtime_error("scene not uploaded to device");
+        dev_sv.cam = cam;   // camera / depth params may have changed (hot reload)
+        dev_sv.md = md;
+        dev_sv.cam_medium = cam_medium;
+        HIP_OK(launch_render(dev_sv, (float*)accum_ptr, (float*)var_ptr, spp0, nspp, seed,
+                             renderer, spec_constraint, caustic_scaling, (void*)stream));
+    }
+
+    py::dict info() {
+        py::dict d;
+        d["n_prims"] = np_prims.ndim() == 2 ? (int)np_prims.shape(0) : 0;
+        d["n_nodes"] = np_nodes.ndim() == 2 ? (int)np_nodes.shape(0) : 0;
+        d["n_bsdfs"] = (int)bsdfs.size();
+        d["n_emitters"] = (int)emitters.size();
+        d["n_media"] = (int)media.size();
+        d["n_textures"] = (int)tex_host.size();
+        d["has_dev"] = has_dev;
+        return d;
+    }
+};
+
+// ------------------------------------------------------------- BVH builder
+py::tuple py_build_bvh(farr prims, uarr prim_obj, int max_leaf, float overlap_w,
+                       bool use_sbvh, bool ref_unsplit) {
+    int n = (int)prims.shape(0);
+    BVHBuildConfig cfg;
+    cfg.max_leaf_prims = max_leaf;
+    cfg.overlap_w = overlap_w;
+    cfg.use_sbvh = use_sbvh;
+    cfg.ref_unsplit = ref_unsplit;
+    BVHBuildResult res;
+    {
+        py::gil_scoped_release rel;
+        res = use_sbvh ? build_sbvh((const Prim*)prims.data(), prim_obj.data(), n, cfg)
+                       : build_bvh((const Prim*)prims.data(), prim_obj.data(), n, cfg);
+    }
+    farr nodes({(py::ssize_t)res.nodes.size(), (py::ssize_t)8});
+    std::memcpy(nodes.mutable_data(), res.nodes.data(), res.nodes.size() * sizeof(BVHNode));
+    iarr order((py::ssize_t)res.prim_order.size());
+    std::memcpy(order.mutable_data(), res.prim_order.data(), res.prim_order.size() * sizeof(int));
+    py::dict stats;
+    stats["n_leaves"] = res.n_leaves;
+    stats["max_depth"] = res.max_depth;
+    stats["sah_cost"] = res.sah_cost;
+    return py::make_tuple(nodes, order, stats);
+}
+
+} // namespace
+
+PYBIND11_MODULE(_C, m) {
+    m.doc() = "hippt native core: MI355X path tracing kernels + scene runtime";
+
+    py::class_<SceneHolder>(m, "Scene")
+        .def(py::init<>())
+        .def("set_geometry", &SceneHolder::set_geometry)
+        .def("set_objects", &SceneHolder::set_objects)
+        .def("add_bsdf", &SceneHolder::add_bsdf)
+        .def("update_bsdf", &SceneHolder::update_bsdf)
+        .def("add_emitter", &SceneHolder::add_emitter)
+        .def("update_emitter", &SceneHolder::update_emitter)
+        .def("set_emitter_prims", &SceneHolder::set_emitter_prims)
+        .def("add_phase", &SceneHolder::add_phase)
+        .def("add_medium", &SceneHolder::add_medium)
+        .def("update_medium", &SceneHolder::update_medium)
+        .def("add_texture", &SceneHolder::add_texture)
+        .def("set_camera", &SceneHolder::set_camera)
+        .def("set_depths", &SceneHolder::set_depths)
+        .def_readwrite("cam_medium", &SceneHolder::cam_medium)
+        .def_readwrite("env_emitter", &SceneHolder::env_emitter)
+        .def("finalize", &SceneHolder::finalize)
+        .def("upload", &SceneHolder::upload)
+        .def("release", &SceneHolder::release)
+        .def("render_host", &SceneHolder::render_host)
+        .def("render_device", &SceneHolder::render_device)
+        .def("info", &SceneHolder::info);
+
+    m.def("build_bvh", &py_build_bvh, py::arg("prims"), py::arg("prim_obj"),
+          py::arg("max_leaf") = 4, py::arg("overlap_w") = 0.f,
+          py::arg("use_sbvh") = false, py::arg("ref_unsplit") = true);
+
+    m.def("dev_synchronize", [] { HIP_OK(dev_synchronize()); });
+    m.def("dev_set_device", [](int d) { HIP_OK(dev_set_device(d)); });
+
+    m.def("struct_sizes", [] {
+        py::dict d;
+        d["BVHNode"] = (int)sizeof(BVHNode);
+        d["Prim"] = (int)sizeof(Prim);
+        d["PrimAttr"] = (int)sizeof(PrimAttr);
+        d["ObjInfo"] = (int)sizeof(ObjInfo);
+        d["BsdfParams"] = (int)sizeof(BsdfParams);
+        d["EmitterParams"] = (int)sizeof(EmitterParams);
+        d["MediumParams"] = (int)sizeof(MediumParams);
+        d["PhaseParams"] = (int)sizeof(PhaseParams);
+        d["TexView"] = (int)sizeof(TexView);
+        d["Camera"] = (int)sizeof(Camera);
+        d["SceneView"] = (int)sizeof(SceneView);
+        return d;
+    });
+
+    m.attr("R_MEGAKERNEL_PT") = (int)R_MEGAKERNEL_PT;
+    m.attr("R_WAVEFRONT_PT") = (int)R_WAVEFRONT_PT;
+    m.attr("R_VOLUME_PT") = (int)R_VOLUME_PT;
+    m.attr("R_LIGHT_TRACE") = (int)R_LIGHT_TRACE;
+    m.attr("R_DEPTH") = (int)R_DEPTH;
+    m.attr("R_BVH_COST") = (int)R_BVH_COST;
+    m.attr("WAVE_SIZE") = 64;
+}

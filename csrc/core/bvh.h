@@ -1,0 +1,111 @@
+// bvh.h — linearized BVH node layout + stackless device/host traversal.
+//
+// Capability parity: reference src/core/bvh.cuh (LinearNode, 2 float4/node,
+// negative-offset skip jumps) + src/renderer/tracing_func.cuh:44-181
+// (ray_intersect_bvh / occlusion_test_bvh).
+//
+// MI355X-native design: nodes are stored in DFS order with an explicit
+// skip-link (the reference encodes the same thing as a negative subtree
+// offset).  Traversal is a single while loop with no stack, no texture
+// fetches (CDNA has no texture-cache path worth using for this): two 16-byte
+// vector loads per node, branchless index advance, early-out on tmax.  The
+// megakernel caches the top of the tree in LDS (see pt_kernels.hip).
+#pragma once
+#include "geometry.h"
+
+namespace hippt {
+
+// Node = 32 bytes:
+//   lo.xyz / hi.xyz  — AABB
+//   lo.w  (int)      — leaf: first primitive index;  internal: unused (-1)
+//   hi.w  (int)      — leaf: primitive count (>0);   internal: -skip_index
+// "skip_index" = node index of the next subtree in DFS order (where to jump
+// on an AABB miss).  For a leaf the skip target is node_idx+1, so only
+// internal nodes store it.
+struct alignas(16) BVHNode {
+    Vec4 lo;
+    Vec4 hi;
+
+    HD bool is_leaf() const { return float_as_int(hi.w) > 0; }
+    HD int prim_base() const { return float_as_int(lo.w); }
+    HD int prim_cnt() const { return float_as_int(hi.w); }
+    HD int skip() const { return -float_as_int(hi.w); }
+    HD AABB aabb() const { return AABB(lo.xyz(), hi.xyz()); }
+};
+
+struct HitRecord {
+    float t;
+    float u, v;
+    int prim_idx;   // -1 = miss
+    HD HitRecord() : t(MAX_DIST), u(0), v(0), prim_idx(-1) {}
+};
+
+// Closest-hit traversal.  `prim_obj[i]` carries PRIM_SPHERE_BIT.
+// Templated on node-fetch so kernels can interpose an LDS-cached top tree.
+HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
+                               const Prim* prims, const uint32_t* prim_obj,
+                               const Ray& ray, float tmax = MAX_DIST) {
+    HitRecord rec;
+    rec.t = tmax;
+    Vec3 inv_d = ray.d.rcp();
+    Vec3 o_div = ray.o * inv_d;
+    int i = 0;
+    while (i < n_nodes) {
+        const BVHNode nd = nodes[i];
+        float t_near;
+        bool hit_box = nd.aabb().intersect(inv_d, o_div, rec.t, t_near);
+        int cnt = float_as_int(nd.hi.w);
+        if (hit_box) {
+            if (cnt > 0) {  // leaf
+                int base = nd.prim_base();
+                for (int k = 0; k < cnt; ++k) {
+                    int pid = base + k;
+                    bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+                    float u, v;
+                    float t = intersect_prim(prims[pid], sph, ray, u, v);
+                    if (t > EPSILON && t < rec.t) {
+                        rec.t = t; rec.u = u; rec.v = v; rec.prim_idx = pid;
+                    }
+                }
+            }
+            ++i;                       // descend / next DFS node
+        } else {
+            i = cnt > 0 ? i + 1 : -cnt;  // leaf miss -> next; internal miss -> skip
+        }
+    }
+    if (rec.prim_idx < 0) rec.t = MAX_DIST;
+    return rec;
+}
+
+// Any-hit occlusion test: returns true if something blocks [EPSILON, tmax].
+HD bool occlusion_test_bvh(const BVHNode* nodes, int n_nodes,
+                           const Prim* prims, const uint32_t* prim_obj,
+                           const Ray& ray, float tmax) {
+    Vec3 inv_d = ray.d.rcp();
+    Vec3 o_div = ray.o * inv_d;
+    int i = 0;
+    while (i < n_nodes) {
+        const BVHNode nd = nodes[i];
+        float t_near;
+        bool hit_box = nd.aabb().intersect(inv_d, o_div, tmax, t_near);
+        int cnt = float_as_int(nd.hi.w);
+        if (hit_box) {
+            if (cnt > 0) {
+                int base = nd.prim_base();
+                for (int k = 0; k < cnt; ++k) {
+                    int pid = base + k;
+                    bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+                    float u, v;
+                    float t = intersect_prim(prims[pid], sph, ray, u, v);
+                    if (t > EPSILON && t < tmax) return true;
+                }
+            }
+            ++i;
+        } else {
+            i = cnt > 0 ? i + 1 : -cnt;
+        }
+    }
+    return false;
+}
+
+} // namespace hippt

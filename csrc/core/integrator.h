@@ -1,0 +1,178 @@
+// integrator.h — the unidirectional path-tracing core (surface scenes).
+// Single-source: this exact function body is the CPU reference renderer
+// (OpenMP over pixels) and the body of the GPU megakernel.
+//
+// Capability parity: reference src/pt_impl/megakernel_pt.cu:33-201
+// (render_pt_kernel: BVH hit -> MIS emitter-hit accumulation -> NEE with one
+// emitter sample + shadow ray -> BSDF sample -> per-lobe depth caps + RR with
+// threshold 0.1 after bounce 1) and the ToF time gating of megakernel_vpt.cu
+// (CONDITION_BLOCK time_in_range).
+#pragma once
+#include "scene_view.h"
+
+namespace hippt {
+
+struct PathStats {
+    int n_diffuse = 0, n_specular = 0, n_transmit = 0, n_volume = 0;
+};
+
+HD bool tof_in_range(const MaxDepthParams& md, float time) {
+    return !md.use_tof || (time >= md.min_time && time <= md.max_time);
+}
+
+// Pick an emitter uniformly; returns index and sets pdf.
+HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
+    if (sv.n_emitters <= 0) { pdf = 0.f; return -1; }
+    int i = (int)(sp.next1f() * sv.n_emitters);
+    i = i >= sv.n_emitters ? sv.n_emitters - 1 : i;
+    pdf = 1.f / sv.n_emitters;
+    return i;
+}
+
+// Full path trace for one camera ray. Returns radiance estimate.
+HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
+    Vec3 L(0.f), thp(1.f);
+    float prev_pdf = 0.f;       // BSDF pdf of the previous bounce (for MIS)
+    bool prev_delta = true;     // camera rays count as delta
+    Vec3 prev_n(0.f, 0.f, 1.f); // previous shading normal (envmap NEE pdf)
+    float path_time = 0.f;      // ToF accumulated distance
+    PathStats st;
+    const float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
+
+    int b = 0;
+    for (int iter = 0; iter < sv.md.max_depth * 2 + 8 && b < sv.md.max_depth; ++iter) {
+        HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+        if (hit.prim_idx < 0) {
+            // miss -> environment map with MIS against the cosine NEE pdf
+            if (sv.env_emitter >= 0) {
+                const EmitterParams& env = sv.emitters[sv.env_emitter];
+                Vec3 le = envmap_eval(env, ray.d, sv.textures);
+                float w = 1.f;
+                if (!prev_delta) {
+                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n) * sel_pdf;
+                    w = mis_weight(prev_pdf, light_pdf);
+                }
+                if (tof_in_range(sv.md, path_time + ENVMAP_DIST)) L += thp * le * w;
+            }
+            break;
+        }
+        Vec3 pos = ray.at(hit.t);
+        path_time += hit.t;
+        uint32_t po = sv.prim_obj[hit.prim_idx];
+        bool is_sphere = (po & PRIM_SPHERE_BIT) != 0;
+        const ObjInfo& obj = sv.objs[po & PRIM_OBJ_MASK];
+        const Prim prim = sv.prims[hit.prim_idx];
+        Interaction it = get_interaction(prim, sv.attrs[hit.prim_idx], is_sphere, pos, hit.u, hit.v);
+        const BsdfParams& bsdf = sv.bsdfs[obj.bsdf_id];
+        if (bsdf.tex[TEX_NORMAL] >= 0)
+            it.shading_n = apply_normal_map(sv.textures, bsdf.tex[TEX_NORMAL], it.uv, it.shading_n);
+
+        // ---- emitter hit accumulation with MIS (megakernel_pt.cu:91-152)
+        if (obj.emitter_id >= 0) {
+            const EmitterParams& em = sv.emitters[obj.emitter_id];
+            Vec3 le = emitter_eval_le(em, it.shading_n, -ray.d, it.uv, sv.textures);
+            if (!le.is_zero()) {
+                float w = 1.f;
+                if (!prev_delta) {
+                    float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n) * sel_pdf;
+                    w = mis_weight(prev_pdf, light_pdf);
+                }
+                if (tof_in_range(sv.md, path_time)) L += thp * le * w;
+            }
+        }
+
+        // ---- next-event estimation (one emitter sample + shadow ray)
+        if (!bsdf_is_delta(bsdf) && sv.n_emitters > 0) {
+            float epdf;
+            int ei = pick_emitter(sv, sp, epdf);
+            EmitterSampleRec er = emitter_sample(sv.emitters[ei], sv.emitter_geom(), pos, it.shading_n, sp);
+            if (er.pdf > 0.f && !er.radiance.is_zero()) {
+                Vec3 to_l = er.pos - pos;
+                float dist = to_l.length();
+                Vec3 wi = to_l * (1.f / fmaxf(dist, 1e-9f));
+                Vec3 f = bsdf_eval(bsdf, -ray.d, wi, it, sv.textures);
+                if (!f.is_zero()) {
+                    Ray sh_ray(fmadd(wi, EPSILON, pos), wi);
+                    float sh_max = (sv.emitters[ei].type == EM_ENVMAP ? ENVMAP_DIST : dist) - 2.f * EPSILON;
+                    if (!occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, sh_ray, sh_max)) {
+                        float light_pdf = er.pdf * epdf;
+                        float w = er.delta ? 1.f
+                                           : mis_weight(light_pdf, bsdf_pdf(bsdf, -ray.d, wi, it, sv.textures));
+                        if (tof_in_range(sv.md, path_time + dist))
+                            L += thp * f * er.radiance * (w / light_pdf);
+                    }
+                }
+            }
+        }
+
+        // ---- BSDF sampling
+        BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
+        if (bs.pdf <= 0.f || bs.weight.is_zero()) break;
+        thp *= bs.weight;
+        if (bs.weight.has_nan()) break;  // numeric scrub (reference numeric_err)
+
+        // per-lobe bounce caps (reference max_depth.h semantics)
+        if (!(bs.lobe & LOBE_NULL)) {
+            if (bs.lobe & LOBE_DIFFUSE)  { if (++st.n_diffuse  > sv.md.max_diffuse)  break; }
+            if (bs.lobe & LOBE_SPECULAR) { if (++st.n_specular > sv.md.max_specular) break; }
+            if (bs.lobe & LOBE_TRANSMIT) { if (++st.n_transmit > sv.md.max_transmit) break; }
+            ++b;
+        }
+        prev_delta = (bs.lobe & LOBE_DELTA) != 0;
+        prev_pdf = bs.pdf;
+        prev_n = it.shading_n;
+        ray = Ray(fmadd(bs.wi, EPSILON, pos), bs.wi);
+
+        // Russian roulette after bounce 1, threshold 0.1 (megakernel_pt.cu RR)
+        if (b > 1) {
+            float p = clampv(thp.max_elem(), 0.f, 1.f);
+            if (p < 0.1f) {
+                if (sp.next1f() >= p * 10.f) break;
+                thp *= (1.f / (p * 10.f));
+            }
+        }
+    }
+    if (L.has_nan()) return Vec3(0.f);
+    return L;
+}
+
+// Depth renderer: distance of the primary hit (reference pt_impl/depth.cu).
+HD float trace_depth(const SceneView& sv, const Ray& ray) {
+    HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+    return hit.prim_idx >= 0 ? hit.t : 0.f;
+}
+
+// BVH-cost visualizer: counts node visits (x) and primitive tests (y) for a
+// primary ray (reference pt_impl/bvh_cost.cu:38-101).
+HD Vec2 trace_bvh_cost(const SceneView& sv, const Ray& ray) {
+    Vec3 inv_d = ray.d.rcp();
+    Vec3 o_div = ray.o * inv_d;
+    float best_t = MAX_DIST;
+    int node_visits = 0, prim_tests = 0;
+    int i = 0;
+    while (i < sv.n_nodes) {
+        const BVHNode nd = sv.nodes[i];
+        ++node_visits;
+        float t_near;
+        bool hit_box = nd.aabb().intersect(inv_d, o_div, best_t, t_near);
+        int cnt = float_as_int(nd.hi.w);
+        if (hit_box) {
+            if (cnt > 0) {
+                int base = nd.prim_base();
+                for (int k = 0; k < cnt; ++k) {
+                    ++prim_tests;
+                    float u, v;
+                    bool sph = (sv.prim_obj[base + k] & PRIM_SPHERE_BIT) != 0;
+                    float t = intersect_prim(sv.prims[base + k], sph, ray, u, v);
+                    if (t > EPSILON && t < best_t) best_t = t;
+                }
+            }
+            ++i;
+        } else {
+            i = cnt > 0 ? i + 1 : -cnt;
+        }
+    }
+    return {(float)node_visits, (float)prim_tests};
+}
+
+} // namespace hippt

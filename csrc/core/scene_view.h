@@ -1,0 +1,68 @@
+// scene_view.h — the flat, pointer-based view of a scene that kernels and the
+// CPU reference renderer consume.  All arrays are built by the Python Scene
+// layer (hippt/scene/scene.py) as torch/numpy buffers and passed as raw
+// pointers; the struct goes to the GPU by value via kernarg.
+//
+// Capability parity: reference src/core/scene.cuh (Scene resource ownership,
+// export_prims), object.cuh (ObjInfo / CompactedObjInfo), max_depth.h
+// (MaxDepthParams incl. ToF min/max time).
+#pragma once
+#include "bvh.h"
+#include "bsdf.h"
+#include "emitter.h"
+#include "camera.h"
+#include "medium.h"
+
+namespace hippt {
+
+struct alignas(16) ObjInfo {
+    int32_t prim_base;
+    int32_t prim_cnt;
+    int32_t bsdf_id;
+    int32_t emitter_id;   // -1 = not an emitter
+    int32_t medium_in;    // interior medium id, -1 = vacuum
+    int32_t medium_out;   // exterior medium id, -1 = vacuum
+    uint32_t flags;       // bit0 = cullable (alpha-masked forward boundary)
+    float inv_area;       // 1 / total surface area
+};
+constexpr uint32_t OBJ_CULLABLE = 1u;
+
+struct MaxDepthParams {
+    int max_depth;     // total bounce cap
+    int max_diffuse;
+    int max_specular;
+    int max_transmit;
+    int max_volume;
+    float min_time;    // ToF gating window (SUPPORTS_TOF_RENDERING parity)
+    float max_time;
+    int use_tof;
+};
+
+struct SceneView {
+    // geometry + BVH
+    const BVHNode* nodes; int n_nodes;
+    const Prim* prims; const PrimAttr* attrs; const uint32_t* prim_obj; int n_prims;
+    const ObjInfo* objs; int n_objs;
+    // materials / emitters / textures
+    const BsdfParams* bsdfs; int n_bsdfs;
+    const EmitterParams* emitters; int n_emitters;
+    const int* emitter_prims;
+    const float* emitter_cdf;
+    const TexView* textures; int n_textures;
+    int env_emitter;   // emitter index of the EM_ENVMAP, or -1
+    // media
+    const MediumParams* media; int n_media;
+    const PhaseParams* phases;
+    int cam_medium;    // medium the camera sits in (-1 = vacuum)
+    // camera + depth caps
+    Camera cam;
+    MaxDepthParams md;
+
+    HD EmitterGeom emitter_geom() const {
+        return {prims, attrs, prim_obj, emitter_prims, emitter_cdf, textures};
+    }
+    HD uint32_t obj_of_prim(int pid) const { return prim_obj[pid] & PRIM_OBJ_MASK; }
+    HD bool prim_is_sphere(int pid) const { return (prim_obj[pid] & PRIM_SPHERE_BIT) != 0; }
+};
+
+} // namespace hippt

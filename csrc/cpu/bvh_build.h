@@ -1,0 +1,39 @@
+// bvh_build.h — host BVH construction API.
+//
+// Capability parity: reference src/impl/bvh.cu (16-bin centroid SAH with
+// overlap penalty, equal-count fallback, leaf threshold, post-build primitive
+// reordering) + src/core/bvh.cuh recursive_linearize.  SBVH (spatial splits)
+// lives in sbvh_build.cpp with the same output contract.
+#pragma once
+#include <vector>
+#include <cstdint>
+#include "../core/bvh.h"
+
+namespace hippt {
+
+struct BVHBuildResult {
+    std::vector<BVHNode> nodes;     // DFS-ordered skip-link nodes
+    std::vector<int> prim_order;    // prim_order[new_slot] = original prim index
+    int n_leaves = 0;
+    int max_depth = 0;
+    float sah_cost = 0.f;
+};
+
+struct BVHBuildConfig {
+    int max_leaf_prims = 4;
+    float overlap_w = 0.f;       // SAH overlap penalty weight (reference bvh_overlap_w)
+    bool use_sbvh = false;       // spatial splits (SBVH)
+    bool ref_unsplit = true;     // SBVH reference unsplitting
+    int n_threads = 8;
+};
+
+// prims/n: primitive array (triangles use v0/e1/e2; spheres per prim_obj bit31)
+BVHBuildResult build_bvh(const Prim* prims, const uint32_t* prim_obj, int n,
+                         const BVHBuildConfig& cfg);
+
+// SBVH: spatial-split BVH (Stich et al. style chopped binning); may duplicate
+// references, so prim_order can be longer than n.
+BVHBuildResult build_sbvh(const Prim* prims, const uint32_t* prim_obj, int n,
+                          const BVHBuildConfig& cfg);
+
+} // namespace hippt

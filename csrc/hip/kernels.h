@@ -1,0 +1,40 @@
+// kernels.h — launch API of the gfx950 HIP kernels (implemented in *.hip).
+#pragma once
+#include <cstdint>
+#include "../core/scene_view.h"
+
+namespace hippt {
+
+enum RendererKind : int {
+    R_MEGAKERNEL_PT = 0,
+    R_WAVEFRONT_PT = 1,
+    R_VOLUME_PT = 2,
+    R_LIGHT_TRACE = 3,
+    R_DEPTH = 4,
+    R_BVH_COST = 5,
+};
+
+// Accumulate nspp samples into accum (h*w*4: RGB sum + count) and var
+// (h*w*2: lum sum, lum^2 sum). sv must hold DEVICE pointers.
+// Returns hipError_t as int.
+int launch_render(const SceneView& sv, float* accum, float* var,
+                  int spp0, int nspp, uint32_t seed, int renderer,
+                  int spec_constraint, float caustic_scaling,
+                  void* stream);
+
+// Wavefront path tracer (SoA queues + compaction); state owned by WfState.
+struct WfState;
+WfState* wf_create(int width, int height);
+void wf_destroy(WfState* s);
+int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, float* var,
+                            int spp0, int nspp, uint32_t seed, int sort_mode, void* stream);
+
+// device helpers used by bind.cpp
+int dev_malloc(void** p, size_t n);
+int dev_free(void* p);
+int dev_upload(void* dst, const void* src, size_t n);
+int dev_memset(void* dst, int v, size_t n, void* stream);
+int dev_synchronize();
+int dev_set_device(int d);
+
+} // namespace hippt

