@@ -13,6 +13,15 @@ import os
 
 
 def _load_native():
+    # torch MUST be imported before the native extension: torch bundles its own
+    # ROCm runtime (torch/lib/libamdhip64.so, soname libamdhip64.so.7).  When
+    # torch loads first, _C's libamdhip64.so.7 dependency resolves to torch's
+    # already-loaded copy and both share ONE HIP/HSA runtime; the other order
+    # creates two HSA instances and hipSetDevice fails with hipErrorNoDevice.
+    try:
+        import torch  # noqa: F401
+    except ImportError:
+        pass
     try:
         from . import _C  # noqa: F401
         return _C
