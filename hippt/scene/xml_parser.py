@@ -46,7 +46,11 @@ def parse_rgb(value: str):
     v = value.strip()
     if v.startswith("#"):
         return tuple(int(v[i:i + 2], 16) / 255.0 for i in (1, 3, 5))
-    parts = [float(x) for x in v.replace(",", " ").split()]
+    try:
+        parts = [float(x) for x in v.replace(",", " ").split()]
+    except ValueError:
+        # reference quirk: <rgb name="type" value="Diamond"/> carries a string
+        return v
     if len(parts) == 1:
         return (parts[0],) * 3
     return tuple(parts[:3])
