@@ -73,10 +73,12 @@ struct SceneHolder {
 
     SceneView host_sv{};
     SceneView dev_sv{};
+    WfState* wf = nullptr;
 
     ~SceneHolder() { release(); }
 
     void release() {
+        if (wf) { wf_destroy(wf); wf = nullptr; }
         for (void* p : dev_bufs) dev_free(p);
         dev_bufs.clear();
         has_dev = false;
@@ -350,6 +352,16 @@ struct SceneHolder {
         dev_sv.cam = cam;   // camera / depth params may have changed (hot reload)
         dev_sv.md = md;
         dev_sv.cam_medium = cam_medium;
+        if (renderer == R_WAVEFRONT_PT) {
+            if (!wf) {
+                wf = wf_create(cam.w, cam.h);
+                if (!wf) throw std::runtime_error("wavefront state allocation failed");
+            }
+            py::gil_scoped_release rel;
+            HIP_OK(launch_render_wavefront(wf, dev_sv, (float*)accum_ptr, (float*)var_ptr,
+                                           spp0, nspp, seed, 0, (void*)stream));
+            return;
+        }
         HIP_OK(launch_render(dev_sv, (float*)accum_ptr, (float*)var_ptr, spp0, nspp, seed,
                              renderer, spec_constraint, caustic_scaling, (void*)stream));
     }

@@ -84,3 +84,24 @@ class TestGPUDeterminism:
         d2 = cornell_box(width=64, height=64, max_depth=4)
         b = hippt.PythonRenderer(d2, device_id=0).render(spp=8).cpu().numpy()
         np.testing.assert_array_equal(a, b)
+
+
+class TestWavefront:
+    def test_wfpt_matches_megakernel(self):
+        d = cornell_box(width=96, height=96, max_depth=5, renderer="wfpt")
+        wf = hippt.PythonRenderer(d, device_id=0).render(spp=24).cpu().numpy()
+        d2 = cornell_box(width=96, height=96, max_depth=5, renderer="pt")
+        mk = hippt.PythonRenderer(d2, device_id=0).render(spp=24).cpu().numpy()
+        assert np.isfinite(wf).all()
+        m1, m2 = wf[..., :3].mean(), mk[..., :3].mean()
+        assert abs(m1 - m2) / m2 < 0.03, (m1, m2)
+        # structural agreement (both estimate the same integral)
+        diff = np.abs(wf[..., :3] - mk[..., :3]).mean()
+        assert diff < 0.25 * m2 + 0.05, diff
+
+    def test_wfpt_kitchen(self):
+        d = kitchen(width=480, height=270, renderer="wfpt")
+        r = hippt.PythonRenderer(d, device_id=0)
+        img = r.render(spp=4).cpu().numpy()
+        assert np.isfinite(img).all()
+        assert img[..., :3].mean() > 0.01
