@@ -46,7 +46,7 @@ def main():
     import hippt
     from hippt.scene import procedural
 
-    renderer_kind = args.renderer or "wfpt"
+    renderer_kind = args.renderer or "pt"
     if args.scene == "kitchen":
         desc = procedural.kitchen(width=args.width, height=args.height,
                                   renderer=renderer_kind)

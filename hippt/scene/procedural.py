@@ -178,7 +178,7 @@ def kitchen(width=1920, height=1080, spp=64, renderer="wfpt", seed=5,
         x += w + 0.05
     objs.append(ObjectDesc(tris=np.concatenate(cab), bsdf=2))
     # countertop with displaced surface detail (the triangle-count driver)
-    n_grid = int(128 * math.sqrt(detail))
+    n_grid = int(172 * math.sqrt(detail))
     counter = displaced_grid_mesh(
         n_grid, n_grid, 2.4,
         lambda X, Z: 0.02 * np.sin(X * 21.0) * np.cos(Z * 17.0) +
@@ -186,6 +186,17 @@ def kitchen(width=1920, height=1080, spp=64, renderer="wfpt", seed=5,
         y0=0.92)
     counter = transform(counter, translate=(W / 2, 0, D - 1.5))
     objs.append(ObjectDesc(tris=counter, bsdf=9))
+    # tiled backsplash wall with relief (second detail surface)
+    n_bs = int(150 * math.sqrt(detail))
+    splash = displaced_grid_mesh(
+        n_bs, n_bs, 2.8,
+        lambda X, Z: 0.012 * np.sign(np.sin(X * 12.0) * np.sin(Z * 12.0)) +
+                     0.004 * np.sin(X * 40.0),
+        y0=0.0)
+    # rotate the heightfield sheet vertical against the back wall
+    splash = splash[:, :, [0, 2, 1]]   # swap y/z -> vertical panel
+    splash = transform(splash, translate=(W / 2, 1.9, D - 0.06))
+    objs.append(ObjectDesc(tris=splash, bsdf=2))
     # island
     objs.append(ObjectDesc(tris=box_mesh((W / 2 - 1.1, 0, 1.6), (W / 2 + 1.1, 0.95, 2.8)), bsdf=2))
     # appliances: fridge + oven (metal)
@@ -195,7 +206,7 @@ def kitchen(width=1920, height=1080, spp=64, renderer="wfpt", seed=5,
     # glassware + bowls on the counter/island (spheres & sphere meshes)
     glass_tris = []
     steel_tris = []
-    for i in range(int(14 * detail)):
+    for i in range(int(40 * detail)):
         cx = rng.uniform(W / 2 - 1.0, W / 2 + 1.0)
         cz = rng.uniform(1.7, 2.7)
         r = rng.uniform(0.04, 0.10)
