@@ -8,6 +8,7 @@
 // across 8 XCDs, and a per-kernel spp loop to amortize launch overhead.
 #include <hip/hip_runtime.h>
 #include "kernels.h"
+#include <cstdlib>
 #include "../core/integrator.h"
 #include "../core/integrator_vol.h"
 #include "../core/light_tracer.h"
@@ -15,8 +16,10 @@
 namespace hippt {
 
 // ------------------------------------------------------------- PT megakernel
-template <int RENDERER>
-__global__ __launch_bounds__(256)
+// MINWAVES = __launch_bounds__ waves/SIMD floor (occupancy vs register trade;
+// 0 keeps the compiler's choice).  Runtime-selectable for A/B via HIPPT_WAVES.
+template <int RENDERER, int MINWAVES = 0>
+__global__ __launch_bounds__(256, MINWAVES)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
               int spp0, int nspp, uint32_t seed) {
     const int px = blockIdx.x * 16 + threadIdx.x;
@@ -93,9 +96,19 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     case R_BVH_COST:
         hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         break;
-    default:
-        hipLaunchKernelGGL(k_render<R_MEGAKERNEL_PT>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+    default: {
+        static int waves = [] {
+            const char* e = getenv("HIPPT_WAVES");
+            return e ? atoi(e) : 0;
+        }();
+        if (waves >= 5)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        else if (waves == 4)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        else
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 0>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         break;
+    }
     }
     return (int)hipGetLastError();
 }
