@@ -463,5 +463,6 @@ PYBIND11_MODULE(_C, m) {
     m.attr("R_LIGHT_TRACE") = (int)R_LIGHT_TRACE;
     m.attr("R_DEPTH") = (int)R_DEPTH;
     m.attr("R_BVH_COST") = (int)R_BVH_COST;
+    m.attr("R_MEGAKERNEL_PT_DYN") = (int)R_MEGAKERNEL_PT_DYN;
     m.attr("WAVE_SIZE") = 64;
 }

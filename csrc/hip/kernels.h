@@ -12,6 +12,7 @@ enum RendererKind : int {
     R_LIGHT_TRACE = 3,
     R_DEPTH = 4,
     R_BVH_COST = 5,
+    R_MEGAKERNEL_PT_DYN = 6,
 };
 
 // Accumulate nspp samples into accum (h*w*4: RGB sum + count) and var

@@ -47,6 +47,47 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
     if (var) { var[pix * 2 + 0] += lum_s; var[pix * 2 + 1] += lum_s2; }
 }
 
+// ------------------------------------------------- persistent-tile variant
+// Capability parity: reference scheduler.cuh:75-96 PreemptivePersistentTileScheduler
+// (persistent blocks atomically grab the next tile id from a global counter).
+// On MI355X: grid = 256 CUs x blocks/CU, tile = 16x16 pixels, counter in
+// device memory zeroed per launch; removes the tail effect of uneven
+// per-tile path lengths on the 8-XCD chip.
+template <int RENDERER>
+__global__ __launch_bounds__(256, 4)
+void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
+                         int spp0, int nspp, uint32_t seed, uint32_t* work_counter,
+                         int tiles_x, int n_tiles) {
+    __shared__ uint32_t s_tile;
+    for (;;) {
+        if (threadIdx.x == 0 && threadIdx.y == 0)
+            s_tile = atomicAdd(work_counter, 1u);
+        __syncthreads();
+        uint32_t tile = s_tile;
+        __syncthreads();
+        if (tile >= (uint32_t)n_tiles) return;
+        int tx = (int)(tile % tiles_x), ty = (int)(tile / tiles_x);
+        int px = tx * 16 + threadIdx.x;
+        int py = ty * 16 + threadIdx.y;
+        if (px >= sv.cam.w || py >= sv.cam.h) continue;
+        const size_t pix = size_t(py) * sv.cam.w + px;
+        Vec3 Lsum(0.f);
+        float lum_s = 0.f, lum_s2 = 0.f;
+        for (int s = 0; s < nspp; ++s) {
+            Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
+            Ray ray = sv.cam.gen_ray(px, py, sp);
+            Vec3 L = trace_path(sv, ray, sp);
+            Lsum += L;
+            float lum = (L.x + L.y + L.z) * (1.f / 3.f);
+            lum_s += lum;
+            lum_s2 = fmaf(lum, lum, lum_s2);
+        }
+        float* a = accum + pix * 4;
+        a[0] += Lsum.x; a[1] += Lsum.y; a[2] += Lsum.z; a[3] += (float)nspp;
+        if (var) { var[pix * 2 + 0] += lum_s; var[pix * 2 + 1] += lum_s2; }
+    }
+}
+
 // ------------------------------------------------------------- light tracing
 __global__ __launch_bounds__(256)
 void k_render_lt(SceneView sv, float* __restrict__ accum,
@@ -96,6 +137,19 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     case R_BVH_COST:
         hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         break;
+    case R_MEGAKERNEL_PT_DYN: {
+        static uint32_t* counter = nullptr;
+        if (!counter) {
+            if (hipMalloc((void**)&counter, 4) != hipSuccess) return (int)hipGetLastError();
+        }
+        hipMemsetAsync(counter, 0, 4, st);
+        int tiles_x = (w + 15) / 16, tiles_y = (h + 15) / 16;
+        // 256 CUs x 4 blocks/CU (4 wave64 each at waves/SIMD>=4)
+        hipLaunchKernelGGL((k_render_persistent<R_MEGAKERNEL_PT>), dim3(256 * 4), block, 0, st,
+                           sv, accum, var, spp0, nspp, seed, counter, tiles_x,
+                           tiles_x * tiles_y);
+        break;
+    }
     default: {
         static int waves = [] {
             const char* e = getenv("HIPPT_WAVES");

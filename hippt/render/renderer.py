@@ -20,7 +20,7 @@ from ..utils.png import tonemap, write_png
 
 RENDERER_IDS = {
     "pt": C.R_MEGAKERNEL_PT,
-    "pt-dyn": C.R_MEGAKERNEL_PT,     # persistent-scheduler variant (GPU flag)
+    "pt-dyn": C.R_MEGAKERNEL_PT_DYN,  # persistent-tile scheduler (GPU)
     "wfpt": C.R_WAVEFRONT_PT,
     "vpt": C.R_VOLUME_PT,
     "lt": C.R_LIGHT_TRACE,
