@@ -40,6 +40,12 @@ struct WfState {
     uint32_t* hist = nullptr;    // nb_sort * 256
     int* live_dev = nullptr;
     int* live_host = nullptr;    // pinned
+    // shadow-ray queue: NEE visibility moved out of the fat shade kernel into
+    // a lean 8-wave-per-SIMD traversal kernel (MI355X occupancy design)
+    float4* sh_od = nullptr;     // origin xyz + tmax
+    float4* sh_dir = nullptr;    // dir xyz + payload idx (uint bits)
+    float4* sh_val = nullptr;    // premultiplied contribution rgb
+    int* sh_cnt = nullptr;       // device append counter
 };
 
 // lobe counters packed into prevn.w: 4 x 8-bit (diffuse, specular, transmit, total)
@@ -195,7 +201,7 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order, in
         }
     }
 
-    // NEE
+    // NEE: enqueue the shadow ray; visibility resolved by k_wf_shadow
     if (!bsdf_is_delta(bsdf) && sv.n_emitters > 0) {
         float epdf;
         int ei = pick_emitter(sv, sp, epdf);
@@ -207,13 +213,17 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order, in
             Vec3 wi = to_l * (1.f / fmaxf(dist, 1e-9f));
             Vec3 f = bsdf_eval(bsdf, -ray.d, wi, it, sv.textures);
             if (!f.is_zero()) {
-                Ray sh_ray(fmadd(wi, EPSILON, pos), wi);
                 float sh_max = (sv.emitters[ei].type == EM_ENVMAP ? ENVMAP_DIST : dist) - 2.f * EPSILON;
-                if (!occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, sh_ray, sh_max)) {
-                    float light_pdf = er.pdf * epdf;
-                    float w = er.delta ? 1.f
-                                       : mis_weight(light_pdf, bsdf_pdf(bsdf, -ray.d, wi, it, sv.textures));
-                    L += thp * f * er.radiance * (w / light_pdf);
+                float light_pdf = er.pdf * epdf;
+                float w = er.delta ? 1.f
+                                   : mis_weight(light_pdf, bsdf_pdf(bsdf, -ray.d, wi, it, sv.textures));
+                Vec3 val = thp * f * er.radiance * (w / light_pdf);
+                if (!val.is_zero() && !val.has_nan()) {
+                    int slot = atomicAdd(st.sh_cnt, 1);
+                    Vec3 so = fmadd(wi, EPSILON, pos);
+                    st.sh_od[slot] = make_float4(so.x, so.y, so.z, sh_max);
+                    st.sh_dir[slot] = make_float4(wi.x, wi.y, wi.z, uint_as_float((uint32_t)i));
+                    st.sh_val[slot] = make_float4(val.x, val.y, val.z, 0.f);
                 }
             }
         }
@@ -260,6 +270,23 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order, in
     // write status into the SAME slot of the order buffer (pre-sort input of
     // the next bounce); material id is refined by the trace kernel
     const_cast<uint32_t*>(order)[k] = (status << 24) | (uint32_t)i;
+}
+
+// ---------------------------------------------------- shadow-ray resolve
+// Lean traversal-only kernel (58 VGPR class -> 8 waves/SIMD): any-hit test,
+// then a race-free add into L (exactly one shadow ray per payload per bounce).
+__global__ __launch_bounds__(256)
+void k_wf_shadow(SceneView sv, WfState st) {
+    int k = blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= *st.sh_cnt) return;
+    float4 od = st.sh_od[k];
+    float4 dir = st.sh_dir[k];
+    Ray ray(Vec3(od.x, od.y, od.z), Vec3(dir.x, dir.y, dir.z));
+    if (occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray, od.w)) return;
+    int i = (int)float_as_uint(dir.w);
+    float4 v = st.sh_val[k];
+    float4 l4 = st.L[i];
+    st.L[i] = make_float4(l4.x + v.x, l4.y + v.y, l4.z + v.z, l4.w);
 }
 
 // ------------------------------------------------------- next closest hit
@@ -339,6 +366,10 @@ WfState* wf_create(int width, int height) {
     e |= wf_alloc(&s->idx[1], s->n);
     e |= wf_alloc(&s->hist, (size_t)s->nb_sort * 256);
     e |= wf_alloc(&s->live_dev, 1);
+    e |= wf_alloc(&s->sh_od, s->n);
+    e |= wf_alloc(&s->sh_dir, s->n);
+    e |= wf_alloc(&s->sh_val, s->n);
+    e |= wf_alloc(&s->sh_cnt, 1);
     e |= (int)hipHostMalloc((void**)&s->live_host, sizeof(int));
     if (e) { wf_destroy(s); return nullptr; }
     return s;
@@ -350,6 +381,7 @@ void wf_destroy(WfState* s) {
     hipFree(s->hit); hipFree(s->prevn); hipFree(s->rng);
     hipFree(s->idx[0]); hipFree(s->idx[1]); hipFree(s->hist);
     hipFree(s->live_dev);
+    hipFree(s->sh_od); hipFree(s->sh_dir); hipFree(s->sh_val); hipFree(s->sh_cnt);
     if (s->live_host) hipHostFree(s->live_host);
     delete s;
 }
@@ -381,8 +413,10 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             live = *st->live_host;
             if (live <= 0) break;
             dim3 grd_live((live + WF_BLOCK - 1) / WF_BLOCK);
+            hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, st->idx[cur], live,
                                bounce);
+            hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, 0, hs, sv, *st);
             hipLaunchKernelGGL(k_wf_trace, grd_live, blk, 0, hs, sv, *st, st->idx[cur], live);
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
