@@ -36,7 +36,8 @@ struct WfState {
     float4* hit = nullptr;       // t, u, v, prim_idx (int bits)
     float4* prevn = nullptr;     // previous shading normal + bounce counters packed in w
     unsigned long long* rng = nullptr;
-    uint32_t* idx[2] = {nullptr, nullptr};
+    uint32_t* status = nullptr;  // (status<<24)|payload, PIXEL order (trace input)
+    uint32_t* order = nullptr;   // compacted material-sorted view (shade input)
     uint32_t* hist = nullptr;    // nb_sort * 256
     int* live_dev = nullptr;
     int* live_host = nullptr;    // pinned
@@ -80,7 +81,7 @@ void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed) {
         uint32_t oi = sv.prim_obj[hit.prim_idx] & PRIM_OBJ_MASK;
         status = (uint32_t)(sv.objs[oi].bsdf_id & 0x3F);
     }
-    st.idx[0][i] = (status << 24) | (uint32_t)i;
+    st.status[i] = (status << 24) | (uint32_t)i;
 }
 
 // ------------------------------------------------------------ counting sort
@@ -267,9 +268,9 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order, in
     st.L[i] = make_float4(L.x, L.y, L.z, l4.w);
     st.prevn[i] = make_float4(prev_n.x, prev_n.y, prev_n.z, uint_as_float(counts));
     st.rng[i] = sp.state;
-    // write status into the SAME slot of the order buffer (pre-sort input of
-    // the next bounce); material id is refined by the trace kernel
-    const_cast<uint32_t*>(order)[k] = (status << 24) | (uint32_t)i;
+    // write status at the PAYLOAD index (status array stays in pixel order;
+    // the trace kernel scans it coherently); material id refined by trace
+    st.status[i] = (status << 24) | (uint32_t)i;
 }
 
 // ---------------------------------------------------- shadow-ray resolve
@@ -291,12 +292,14 @@ void k_wf_shadow(SceneView sv, WfState st) {
 
 // ------------------------------------------------------- next closest hit
 __global__ __launch_bounds__(256)
-void k_wf_trace(SceneView sv, WfState st, uint32_t* __restrict__ order, int n_live) {
-    int k = blockIdx.x * blockDim.x + threadIdx.x;
-    if (k >= n_live) return;
-    uint32_t entry = order[k];
-    if (entry >> 24 >= DEAD) return;  // terminated in shade
-    int i = (int)(entry & 0x00FFFFFFu);
+void k_wf_trace(SceneView sv, WfState st) {
+    // pixel-order scan: bounce rays of neighboring pixels traverse similar
+    // BVH subtrees, so keeping trace in payload order (NOT material-sorted
+    // order) preserves wave-level spatial coherence — measured 4.5ms -> see
+    // profiles/ for the sorted-order version this replaces.
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= st.n) return;
+    if (st.status[i] >> 24 >= DEAD) return;  // dead or terminated in shade
     float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i];
     Ray ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
     HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
@@ -325,7 +328,7 @@ void k_wf_trace(SceneView sv, WfState st, uint32_t* __restrict__ order, int n_li
         uint32_t oi = sv.prim_obj[hit.prim_idx] & PRIM_OBJ_MASK;
         status = (uint32_t)(sv.objs[oi].bsdf_id & 0x3F);
     }
-    order[k] = (status << 24) | (uint32_t)i;
+    st.status[i] = (status << 24) | (uint32_t)i;
 }
 
 // ----------------------------------------------------------------- splat
@@ -362,8 +365,8 @@ WfState* wf_create(int width, int height) {
     e |= wf_alloc(&s->hit, s->n);
     e |= wf_alloc(&s->prevn, s->n);
     e |= wf_alloc(&s->rng, s->n);
-    e |= wf_alloc(&s->idx[0], s->n);
-    e |= wf_alloc(&s->idx[1], s->n);
+    e |= wf_alloc(&s->status, s->n);
+    e |= wf_alloc(&s->order, s->n);
     e |= wf_alloc(&s->hist, (size_t)s->nb_sort * 256);
     e |= wf_alloc(&s->live_dev, 1);
     e |= wf_alloc(&s->sh_od, s->n);
@@ -379,7 +382,7 @@ void wf_destroy(WfState* s) {
     if (!s) return;
     hipFree(s->ray_o); hipFree(s->ray_d); hipFree(s->thp); hipFree(s->L);
     hipFree(s->hit); hipFree(s->prevn); hipFree(s->rng);
-    hipFree(s->idx[0]); hipFree(s->idx[1]); hipFree(s->hist);
+    hipFree(s->status); hipFree(s->order); hipFree(s->hist);
     hipFree(s->live_dev);
     hipFree(s->sh_od); hipFree(s->sh_dir); hipFree(s->sh_val); hipFree(s->sh_cnt);
     if (s->live_host) hipHostFree(s->live_host);
@@ -395,17 +398,16 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, 0, hs, sv, *st, spp0 + s, seed);
         int live = n;
-        int cur = 0;
+        const int nb = st->nb_sort;
         for (int bounce = 0; bounce < sv.md.max_depth + 1 && live > 0; ++bounce) {
-            // sort current entries by status byte (compacts dead to tail)
-            int nb = (live + SORT_BLOCK * SORT_ITEMS - 1) / (SORT_BLOCK * SORT_ITEMS);
+            // build the compacted material-sorted view over the FULL status
+            // array (pixel order preserved there for the trace kernel)
             hipLaunchKernelGGL(k_sort_hist, dim3(nb), dim3(SORT_BLOCK), 0, hs,
-                               st->idx[cur], live, st->hist);
+                               st->status, n, st->hist);
             hipLaunchKernelGGL(k_sort_scan, dim3(1), dim3(256), 0, hs,
-                               st->hist, nb, st->live_dev, live);
+                               st->hist, nb, st->live_dev, n);
             hipLaunchKernelGGL(k_sort_scatter, dim3(nb), dim3(SORT_BLOCK), 0, hs,
-                               st->idx[cur], live, st->hist, st->idx[cur ^ 1]);
-            cur ^= 1;
+                               st->status, n, st->hist, st->order);
             hipMemcpyAsync(st->live_host, st->live_dev, sizeof(int),
                            hipMemcpyDeviceToHost, hs);
             hipError_t se = hipStreamSynchronize(hs);
@@ -414,10 +416,10 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             if (live <= 0) break;
             dim3 grd_live((live + WF_BLOCK - 1) / WF_BLOCK);
             hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
-            hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, st->idx[cur], live,
+            hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, st->order, live,
                                bounce);
             hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, 0, hs, sv, *st);
-            hipLaunchKernelGGL(k_wf_trace, grd_live, blk, 0, hs, sv, *st, st->idx[cur], live);
+            hipLaunchKernelGGL(k_wf_trace, grd_n, blk, 0, hs, sv, *st);
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
     }

@@ -385,6 +385,80 @@ class Scene:
         self._set_camera_native()
         self.native.finalize()
 
+    # --------------------------------------------------- hot reload (GUI)
+    # parity: reference dynamic_bsdf.cu copy_to_gpu/create_on_gpu + Scene::
+    # update_emitters/materials/media; our tagged-union params make a type
+    # switch the same cheap struct overwrite as a parameter tweak.
+    def set_bsdf(self, i: int, b: BsdfDesc):
+        from .scene import BSDF_TYPES  # self-import safe
+        self.desc.bsdfs[i] = b
+        ty = BSDF_TYPES[b.type]
+        kd, ks, kg = list(b.kd), list(b.ks), list(b.kg)
+        ior, e0, e1 = b.ior, 0.0, 0.0
+        if ty == 5:
+            eta, kk = METALS.get(b.metal or "Au", METALS["Au"])
+            kd, ks = list(eta), list(kk)
+            if not any(kg):
+                kg = [1.0, 1.0, 1.0]
+            e0, e1 = b.roughness_x, b.roughness_y
+        elif ty in (3, 4):
+            e0, e1 = b.trans_scaler, b.thickness
+        elif ty == 6:
+            A, B = DISPERSION_PRESETS.get(b.preset or "diamond",
+                                          DISPERSION_PRESETS["diamond"])
+            e0, e1 = A, B
+        self.native.update_bsdf(i, ty, kd, ks, kg, ior, e0, e1, _tex_slots(b))
+        self.native.finalize()
+
+    def set_emitter(self, i: int, emission=None, scale=None, cos_max=None,
+                    azimuth=None, zenith=None, pos=None):
+        e = self.desc.emitters[i]
+        if emission is not None:
+            e.emission = tuple(emission)
+        if scale is not None:
+            e.scale = float(scale)
+        if cos_max is not None:
+            e.cos_max = float(cos_max)
+        if azimuth is not None:
+            e.azimuth = float(azimuth)
+        if zenith is not None:
+            e.zenith = float(zenith)
+        if pos is not None:
+            e.pos = tuple(pos)
+        aux = [0.0, 0.0, 0.0, 0.0]
+        if e.type == "point":
+            aux[:3] = list(e.pos)
+        elif e.type == "envmap":
+            aux[0], aux[1] = e.azimuth, e.zenith
+        elif e.type == "area-spot":
+            aux[3] = e.cos_max
+        self.native.update_emitter(i, list(e.emission), e.scale, aux)
+        self.native.finalize()
+
+    def set_medium(self, i: int, sigma_a=None, sigma_s=None, scale=None,
+                   emission_scale=None):
+        m = self.desc.media[i]
+        if sigma_a is not None:
+            m.sigma_a = tuple(sigma_a)
+        if sigma_s is not None:
+            m.sigma_s = tuple(sigma_s)
+        if scale is not None:
+            m.scale = float(scale)
+        if emission_scale is not None:
+            m.emission_scale = float(emission_scale)
+        self.native.update_medium(i, list(m.sigma_a), list(m.sigma_s), m.scale,
+                                  m.emission_scale)
+        self.native.finalize()
+
+    def set_depths(self, **kw):
+        cfg = self.desc.config
+        for k, v in kw.items():
+            setattr(cfg, k, v)
+        self.native.set_depths(cfg.max_depth, cfg.max_diffuse, cfg.max_specular,
+                               cfg.max_transmit, cfg.max_volume,
+                               cfg.min_time, cfg.max_time, int(cfg.use_tof))
+        self.native.finalize()
+
     @property
     def width(self):
         return self.desc.camera.width
