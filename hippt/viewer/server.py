@@ -247,13 +247,14 @@ class ViewerApp:
                 "resolution": [self.desc.camera.width, self.desc.camera.height]}
 
 
-def build_app(viewer: ViewerApp):
-    from fastapi import FastAPI, Response
-    from pydantic import BaseModel
 
-    app = FastAPI(title="hippt viewer")
+# Request models at module scope: with `from __future__ import annotations`
+# FastAPI resolves string annotations via module globals, so locally-scoped
+# models would silently degrade to query parameters.
+try:
+    from pydantic import BaseModel as _BaseModel
 
-    class BsdfReq(BaseModel):
+    class BsdfReq(_BaseModel):
         index: int
         type: Optional[str] = None
         kd: Optional[list] = None
@@ -264,30 +265,37 @@ def build_app(viewer: ViewerApp):
         roughness_y: Optional[float] = None
         metal: Optional[str] = None
 
-    class EmitterReq(BaseModel):
+    class EmitterReq(_BaseModel):
         index: int
         emission: Optional[list] = None
         scale: Optional[float] = None
 
-    class MediumReq(BaseModel):
+    class MediumReq(_BaseModel):
         index: int
         sigma_a: Optional[list] = None
         sigma_s: Optional[list] = None
         scale: Optional[float] = None
         emission_scale: Optional[float] = None
 
-    class DepthReq(BaseModel):
+    class DepthReq(_BaseModel):
         max_depth: Optional[int] = None
         max_diffuse: Optional[int] = None
         max_specular: Optional[int] = None
         max_transmit: Optional[int] = None
         max_volume: Optional[int] = None
 
-    class RendererReq(BaseModel):
+    class RendererReq(_BaseModel):
         kind: str
 
-    class MoveReq(BaseModel):
+    class MoveReq(_BaseModel):
         key: str
+except ImportError:  # viewer optional without fastapi/pydantic
+    pass
+
+def build_app(viewer: ViewerApp):
+    from fastapi import FastAPI, Response
+
+    app = FastAPI(title="hippt viewer")
 
     @app.get("/")
     def index():
