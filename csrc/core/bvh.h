@@ -33,6 +33,23 @@ struct alignas(16) BVHNode {
     HD AABB aabb() const { return AABB(lo.xyz(), hi.xyz()); }
 };
 
+// Safe direction reciprocal: a zero (or denormal) direction component makes
+// inv_d infinite and the slab test compute 0*inf = NaN; under -ffast-math
+// fmaxf(NaN, 0) = 0, so every AABB tests as HIT and one ray scans the whole
+// tree (measured: 90 ms dispatches vs 1 ms median on MI355X).  Clamping the
+// component magnitude keeps inv_d finite and the test conservative.
+HD float safe_rcp_1(float v) {
+    const float eps = 1e-12f;
+    // bit-level NaN/Inf test (survives -ffast-math where v != v folds away)
+    uint32_t bits = float_as_uint(v);
+    if ((bits & 0x7f800000u) == 0x7f800000u) v = eps;
+    if (fabsf(v) < eps) v = copysignf(eps, v);
+    return 1.f / v;
+}
+HD Vec3 safe_rcp_dir(const Vec3& d) {
+    return {safe_rcp_1(d.x), safe_rcp_1(d.y), safe_rcp_1(d.z)};
+}
+
 struct HitRecord {
     float t;
     float u, v;
@@ -47,7 +64,7 @@ HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
                                const Ray& ray, float tmax = MAX_DIST) {
     HitRecord rec;
     rec.t = tmax;
-    Vec3 inv_d = ray.d.rcp();
+    Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     int i = 0;
     while (i < n_nodes) {
@@ -81,7 +98,7 @@ HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
 HD bool occlusion_test_bvh(const BVHNode* nodes, int n_nodes,
                            const Prim* prims, const uint32_t* prim_obj,
                            const Ray& ray, float tmax) {
-    Vec3 inv_d = ray.d.rcp();
+    Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     int i = 0;
     while (i < n_nodes) {

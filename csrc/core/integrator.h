@@ -108,8 +108,8 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
         // ---- BSDF sampling
         BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
         if (bs.pdf <= 0.f || bs.weight.is_zero()) break;
+        if (bs.weight.has_nan() || bs.wi.has_nan()) break;  // numeric scrub
         thp *= bs.weight;
-        if (bs.weight.has_nan()) break;  // numeric scrub (reference numeric_err)
 
         // per-lobe bounce caps (reference max_depth.h semantics)
         if (!(bs.lobe & LOBE_NULL)) {
@@ -145,7 +145,7 @@ HD float trace_depth(const SceneView& sv, const Ray& ray) {
 // BVH-cost visualizer: counts node visits (x) and primitive tests (y) for a
 // primary ray (reference pt_impl/bvh_cost.cu:38-101).
 HD Vec2 trace_bvh_cost(const SceneView& sv, const Ray& ray) {
-    Vec3 inv_d = ray.d.rcp();
+    Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     float best_t = MAX_DIST;
     int node_visits = 0, prim_tests = 0;

@@ -191,7 +191,7 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp) {
             }
 
             BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
-            if (bs.pdf <= 0.f || bs.weight.is_zero() || bs.weight.has_nan()) break;
+            if (bs.pdf <= 0.f || bs.weight.is_zero() || bs.weight.has_nan() || bs.wi.has_nan()) break;
             thp *= bs.weight;
             if ((bs.lobe & LOBE_TRANSMIT) != 0) cross_boundary(stack, obj, bs.wi, geo_n);
             if (!(bs.lobe & LOBE_NULL)) {

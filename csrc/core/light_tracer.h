@@ -87,7 +87,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
         }
 
         BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
-        if (bs.pdf <= 0.f || bs.weight.is_zero() || bs.weight.has_nan()) break;
+        if (bs.pdf <= 0.f || bs.weight.is_zero() || bs.weight.has_nan() || bs.wi.has_nan()) break;
         thp *= bs.weight;
         if (bs.lobe & (LOBE_SPECULAR | LOBE_TRANSMIT)) ++n_spec;
         if (!(bs.lobe & LOBE_NULL)) {

@@ -233,7 +233,7 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order, in
     // BSDF sample
     uint32_t status = DEAD;
     BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
-    if (bs.pdf > 0.f && !bs.weight.is_zero() && !bs.weight.has_nan()) {
+    if (bs.pdf > 0.f && !bs.weight.is_zero() && !bs.weight.has_nan() && !bs.wi.has_nan()) {
         thp *= bs.weight;
         int nd = counts & 0xFF, ns = (counts >> 8) & 0xFF, nt = (counts >> 16) & 0xFF,
             nb = (counts >> 24) & 0xFF;
