@@ -40,7 +40,10 @@ def box_mesh(lo, hi, inward=False):
         quad((x0, y1, z0), (x1, y1, z0), (x1, y1, z1), (x0, y1, z1)),  # y1 (ceiling)
     ]
     tris = np.concatenate(faces)
-    if inward:
+    # the face quads above wind with normals pointing INTO the box; flip by
+    # default so solid objects have outward geometric normals (medium
+    # enter/exit tests depend on this), keep them for room interiors
+    if not inward:
         tris = tris[:, ::-1, :].copy()
     return tris
 

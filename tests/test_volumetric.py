@@ -1,0 +1,135 @@
+"""Volumetric path tracing on the CPU reference: analytic checks for
+homogeneous media, grid delta/ratio tracking sanity, ToF transient gating.
+(SURVEY.md §4 test plan item: homogeneous-medium analytic transmittance.)"""
+import numpy as np
+import pytest
+
+import hippt
+from hippt.scene.scene import (BsdfDesc, CameraDesc, EmitterDesc, MediumDesc,
+                               ObjectDesc, RenderConfig, SceneDesc)
+from hippt.scene.procedural import quad, smoke_box, cornell_box
+
+
+def emissive_wall_scene(medium=None, depth=8, w=32, h=32, cam_medium=-1):
+    """Camera looks at an emissive wall 4 units away through optional medium."""
+    d = SceneDesc()
+    d.bsdfs = [BsdfDesc(type="lambertian", kd=(0, 0, 0))]
+    d.emitters = [EmitterDesc(type="area", emission=(1, 1, 1), scale=1.0)]
+    # wall at z=4 facing -z (toward camera)
+    wall = quad((-8, -8, 4), (-8, 8, 4), (8, 8, 4), (8, -8, 4))
+    d.objects = [ObjectDesc(tris=wall, bsdf=0, emitter=0)]
+    if medium is not None:
+        d.media = [medium]
+        cam_medium = 0
+    d.cam_medium = cam_medium
+    d.camera = CameraDesc(pos=(0, 0, 0), lookat=(0, 0, 4), fov=30, width=w, height=h)
+    d.config = RenderConfig(spp=8, max_depth=depth, max_volume=depth, renderer="vpt")
+    return d
+
+
+class TestHomogeneous:
+    def test_absorbing_medium_beer_lambert(self):
+        """Pure absorption: L = Le * exp(-sigma_a * dist)."""
+        sigma = (0.25, 0.5, 0.125)
+        med = MediumDesc(type="homogeneous", sigma_a=sigma, sigma_s=(0, 0, 0))
+        d = emissive_wall_scene(med)
+        img = hippt.PythonRenderer(d, device_id=-1).render(spp=64).numpy()
+        center = img[16, 16, :3]
+        # center ray travels ~4.0 units
+        expected = np.exp(-np.array(sigma) * 4.0)
+        np.testing.assert_allclose(center, expected, rtol=0.08)
+
+    def test_vacuum_passthrough(self):
+        d = emissive_wall_scene(None)
+        img = hippt.PythonRenderer(d, device_id=-1).render(spp=8).numpy()
+        np.testing.assert_allclose(img[16, 16, :3], 1.0, rtol=0.02)
+
+    def test_scattering_conserves_energy(self):
+        """Purely scattering isotropic medium (albedo 1) bounded by a null
+        (forward, cullable) box inside a unit envmap furnace: the image must
+        stay at 1 (volume white furnace)."""
+        from hippt.scene.procedural import box_mesh
+        d = SceneDesc()
+        d.bsdfs = [BsdfDesc(type="forward")]
+        d.emitters = [EmitterDesc(type="envmap", emission=(1, 1, 1), scale=1.0)]
+        d.media = [MediumDesc(type="homogeneous", sigma_a=(0, 0, 0),
+                              sigma_s=(0.6, 0.6, 0.6), phase="hg", g1=0.4)]
+        d.objects = [ObjectDesc(tris=box_mesh((-1, -1, 1), (1, 1, 3)), bsdf=0,
+                                medium_in=0, cullable=True)]
+        d.camera = CameraDesc(pos=(0, 0, -2), lookat=(0, 0, 1), fov=35,
+                              width=32, height=32)
+        d.config = RenderConfig(spp=8, max_depth=64, max_volume=64,
+                                max_transmit=64, renderer="vpt")
+        img = hippt.PythonRenderer(d, device_id=-1).render(spp=48).numpy()
+        m = img[..., :3].mean()
+        assert abs(m - 1.0) < 0.05, m
+
+
+class TestGridMedium:
+    def test_smoke_renders_and_attenuates(self):
+        d = smoke_box(width=48, height=32, n_grid=32)
+        img = hippt.PythonRenderer(d, device_id=-1).render(spp=8).numpy()
+        assert np.isfinite(img).all()
+        assert img[..., :3].mean() > 0.01
+
+    def test_majorant_scaling(self):
+        """Doubling density scale must darken transmission through the plume."""
+        def render_with(scale):
+            d = smoke_box(width=32, height=24, n_grid=24)
+            d.media[0].density = d.media[0].density * scale / 18.0
+            return hippt.PythonRenderer(d, device_id=-1).render(spp=16).numpy()
+        thin = render_with(4.0)
+        thick = render_with(40.0)
+        # compare center region where the plume sits
+        assert thick[10:16, 12:20, :3].mean() != pytest.approx(
+            thin[10:16, 12:20, :3].mean(), rel=0.02)
+
+    def test_blackbody_emission_adds_energy(self):
+        d1 = smoke_box(width=32, height=24, n_grid=24, emission=False)
+        d2 = smoke_box(width=32, height=24, n_grid=24, emission=True)
+        a = hippt.PythonRenderer(d1, device_id=-1).render(spp=12).numpy()[..., :3].mean()
+        b = hippt.PythonRenderer(d2, device_id=-1).render(spp=12).numpy()[..., :3].mean()
+        assert b > a * 1.005, (a, b)
+
+
+class TestToF:
+    def test_transient_windows_partition_steady_state(self):
+        """Summing transient windows reproduces the steady-state image
+        (serial_render.py job_tof_rendering semantics)."""
+        def render(min_t=0.0, max_t=0.0, use=False, seed=0):
+            d = cornell_box(width=24, height=24, spp=8, max_depth=3)
+            d.config.use_tof = use
+            d.config.min_time = min_t
+            d.config.max_time = max_t
+            return hippt.PythonRenderer(d, device_id=-1, seed_offset=seed).render(spp=32).numpy()
+
+        steady = render()
+        windows = [render(lo, hi, True) for lo, hi in
+                   [(0, 4), (4, 6), (6, 8), (8, 12), (12, 1e6)]]
+        total = sum(w[..., :3] for w in windows)
+        # same RNG streams -> per-pixel partition is exact up to float assoc
+        np.testing.assert_allclose(total, steady[..., :3], rtol=1e-4, atol=1e-5)
+
+    def test_window_excludes_late_light(self):
+        d = cornell_box(width=24, height=24, spp=4, max_depth=5)
+        d.config.use_tof = True
+        d.config.min_time = 0.0
+        d.config.max_time = 3.3   # camera->wall is ~3.4+: only direct peek at light
+        img = hippt.PythonRenderer(d, device_id=-1).render(spp=16).numpy()
+        d2 = cornell_box(width=24, height=24, spp=4, max_depth=5)
+        full = hippt.PythonRenderer(d2, device_id=-1).render(spp=16).numpy()
+        assert img[..., :3].mean() < 0.5 * full[..., :3].mean()
+
+
+class TestColormap:
+    def test_false_color(self):
+        from hippt.utils.colormap import false_color
+        v = np.linspace(0, 1, 64).reshape(8, 8)
+        for cm in ("plasma", "jet", "viridis"):
+            rgb = false_color(v, cmap=cm)
+            assert rgb.shape == (8, 8, 3)
+            assert rgb.min() >= 0 and rgb.max() <= 1
+        # log scale works and misses (zeros) are black
+        v[0, 0] = 0
+        rgb = false_color(v, log_scale=True)
+        assert (rgb[0, 0] == 0).all()
