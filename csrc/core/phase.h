@@ -27,7 +27,9 @@ struct alignas(16) PhaseParams {
 
 // HG phase value p(cos_theta), normalized over the sphere.
 HD float hg_phase(float g, float cos_t) {
-    float denom = 1.f + g * g + 2.f * g * cos_t;
+    // p(cos) = (1-g^2) / (4pi (1 + g^2 - 2 g cos)^(3/2)); forward-peaked for
+    // g > 0, consistent with hg_sample_cos's inverse CDF
+    float denom = 1.f + g * g - 2.f * g * cos_t;
     return (1.f / (4.f * PI)) * (1.f - g * g) / fmaxf(denom * sqrtf(denom), 1e-8f);
 }
 
