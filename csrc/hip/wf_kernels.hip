@@ -85,6 +85,10 @@ void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed) {
 }
 
 // ------------------------------------------------------------ counting sort
+// Grouping sort (order within a bin is arrival order, which preserves
+// block-local pixel adjacency): one global 256-bin histogram, a 1-block
+// exclusive scan producing the live count on-device, and an atomic-cursor
+// scatter.  No host round trip; 3 memory-bound passes over 4B/entry.
 __global__ __launch_bounds__(SORT_BLOCK)
 void k_sort_hist(const uint32_t* __restrict__ in, int n, uint32_t* __restrict__ hist) {
     __shared__ uint32_t lh[256];
@@ -97,71 +101,63 @@ void k_sort_hist(const uint32_t* __restrict__ in, int n, uint32_t* __restrict__ 
     }
     __syncthreads();
     for (int t = threadIdx.x; t < 256; t += blockDim.x)
-        hist[blockIdx.x * 256 + t] = lh[t];
+        if (lh[t]) atomicAdd(&hist[t], lh[t]);
 }
 
-// single block: column sums -> exclusive bin offsets -> per-block bases
+// single block of 256: exclusive scan of the 256 bins -> cursors + live count
 __global__ __launch_bounds__(256)
-void k_sort_scan(uint32_t* __restrict__ hist, int nb, int* __restrict__ live_out, int n) {
-    __shared__ uint32_t bin_total[256];
-    __shared__ uint32_t bin_base[256];
-    int b = threadIdx.x;  // bin id, one thread per bin
-    uint32_t sum = 0;
-    for (int blk = 0; blk < nb; ++blk) {
-        uint32_t v = hist[blk * 256 + b];
-        hist[blk * 256 + b] = sum;  // prefix within bin (per-block start)
-        sum += v;
-    }
-    bin_total[b] = sum;
-    __syncthreads();
-    // exclusive scan over the 256 bins (simple Hillis-Steele in LDS)
-    uint32_t x = bin_total[b];
-    bin_base[b] = x;
+void k_sort_scan(uint32_t* __restrict__ hist, int* __restrict__ live_out) {
+    __shared__ uint32_t base[256];
+    int b = threadIdx.x;
+    uint32_t x = hist[b];
+    base[b] = x;
     __syncthreads();
     for (int off = 1; off < 256; off <<= 1) {
-        uint32_t v = (b >= off) ? bin_base[b - off] : 0;
+        uint32_t v = (b >= off) ? base[b - off] : 0;
         __syncthreads();
-        bin_base[b] += v;
+        base[b] += v;
         __syncthreads();
     }
-    uint32_t excl = bin_base[b] - bin_total[b];
-    __syncthreads();
-    bin_base[b] = excl;
-    __syncthreads();
-    if (b == 0 && live_out) {
-        // live = everything below the first dead bin (status < 0x80)
-        *live_out = (int)bin_base[DEAD];
-    }
-    // add bin base to each per-block prefix
-    for (int blk = 0; blk < nb; ++blk)
-        hist[blk * 256 + b] += bin_base[b];
+    uint32_t excl = base[b] - x;
+    hist[b] = excl;          // becomes the atomic cursor for scatter
+    if (b == (int)DEAD - 1 && live_out) *live_out = (int)base[b];
 }
 
 __global__ __launch_bounds__(SORT_BLOCK)
 void k_sort_scatter(const uint32_t* __restrict__ in, int n,
-                    const uint32_t* __restrict__ hist, uint32_t* __restrict__ out) {
-    __shared__ uint32_t cursor[256];
-    for (int t = threadIdx.x; t < 256; t += blockDim.x)
-        cursor[t] = hist[blockIdx.x * 256 + t];
+                    uint32_t* __restrict__ hist, uint32_t* __restrict__ out) {
+    __shared__ uint32_t lbase[256];
+    __shared__ uint32_t lcnt[256];
+    for (int t = threadIdx.x; t < 256; t += blockDim.x) lcnt[t] = 0;
     __syncthreads();
+    // block-local ranks first, then one global cursor bump per (block, bin)
     int base = blockIdx.x * SORT_BLOCK * SORT_ITEMS;
+    uint32_t rank[SORT_ITEMS];
+    uint32_t ent[SORT_ITEMS];
     for (int k = 0; k < SORT_ITEMS; ++k) {
         int i = base + k * SORT_BLOCK + threadIdx.x;
         if (i < n) {
-            uint32_t e = in[i];
-            uint32_t pos = atomicAdd(&cursor[e >> 24], 1u);
-            out[pos] = e;
+            ent[k] = in[i];
+            rank[k] = atomicAdd(&lcnt[ent[k] >> 24], 1u);
         }
+    }
+    __syncthreads();
+    for (int t = threadIdx.x; t < 256; t += blockDim.x)
+        lbase[t] = lcnt[t] ? atomicAdd(&hist[t], lcnt[t]) : 0;
+    __syncthreads();
+    for (int k = 0; k < SORT_ITEMS; ++k) {
+        int i = base + k * SORT_BLOCK + threadIdx.x;
+        if (i < n) out[lbase[ent[k] >> 24] + rank[k]] = ent[k];
     }
 }
 
 // ----------------------------------------------------------- bounce shade
 // NEE + emitter-hit MIS + BSDF sample for live rays (current hit record).
 __global__ __launch_bounds__(256)
-void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order, int n_live,
+void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
                 int bounce) {
     int k = blockIdx.x * blockDim.x + threadIdx.x;
-    if (k >= n_live) return;
+    if (k >= *st.live_dev) return;
     uint32_t entry = order[k];
     int i = (int)(entry & 0x00FFFFFFu);
     float4 h4 = st.hit[i];
@@ -397,28 +393,23 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     dim3 grd_n((n + WF_BLOCK - 1) / WF_BLOCK);
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, 0, hs, sv, *st, spp0 + s, seed);
-        int live = n;
         const int nb = st->nb_sort;
-        for (int bounce = 0; bounce < sv.md.max_depth + 1 && live > 0; ++bounce) {
+        for (int bounce = 0; bounce < sv.md.max_depth + 1; ++bounce) {
             // build the compacted material-sorted view over the FULL status
-            // array (pixel order preserved there for the trace kernel)
+            // array (pixel order preserved there for the trace kernel);
+            // the live count never leaves the device: shade/shadow/trace are
+            // launched full-width and exit by comparing against *live_dev.
+            hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
             hipLaunchKernelGGL(k_sort_hist, dim3(nb), dim3(SORT_BLOCK), 0, hs,
                                st->status, n, st->hist);
             hipLaunchKernelGGL(k_sort_scan, dim3(1), dim3(256), 0, hs,
-                               st->hist, nb, st->live_dev, n);
+                               st->hist, st->live_dev);
             hipLaunchKernelGGL(k_sort_scatter, dim3(nb), dim3(SORT_BLOCK), 0, hs,
                                st->status, n, st->hist, st->order);
-            hipMemcpyAsync(st->live_host, st->live_dev, sizeof(int),
-                           hipMemcpyDeviceToHost, hs);
-            hipError_t se = hipStreamSynchronize(hs);
-            if (se != hipSuccess) return (int)se;
-            live = *st->live_host;
-            if (live <= 0) break;
-            dim3 grd_live((live + WF_BLOCK - 1) / WF_BLOCK);
             hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
-            hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, st->order, live,
+            hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
-            hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, 0, hs, sv, *st);
+            hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, 0, hs, sv, *st);
             hipLaunchKernelGGL(k_wf_trace, grd_n, blk, 0, hs, sv, *st);
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
