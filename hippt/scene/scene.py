@@ -136,6 +136,7 @@ class CameraDesc:
     focal_dist: float = 1.0
     ortho: bool = False
     ortho_scale: float = 0.01
+    hflip: bool = False    # mirror image horizontally (reference sensor hflip)
 
 
 @dataclass
@@ -179,7 +180,9 @@ def camera_matrix(desc: CameraDesc) -> np.ndarray:
     u = np.asarray(desc.up, np.float64)
     r = np.cross(u, f)   # viewer's right (y-up, looking along f)
     r = r / np.linalg.norm(r)
-    tu = np.cross(f, r)
+    if desc.hflip:
+        r = -r
+    tu = np.cross(f, r if not desc.hflip else -r)
     # rows of R (camera->world, columns right/up/forward)
     R = np.stack([r, tu, f], axis=1)
     return R.astype(np.float32)

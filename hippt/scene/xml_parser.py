@@ -337,10 +337,5 @@ def parse_xml(path: str) -> SceneDesc:
     if ref and ref in medium_ids:
         d.cam_medium = medium_ids[ref]
 
-    if hflip:
-        # reference hflip: mirror the image horizontally = flip camera right axis
-        # implemented by flipping up vector handedness via negative fov? Use a
-        # dedicated flag: swap lookat handedness by negating the right axis.
-        d.camera.up = tuple(u for u in d.camera.up)
-        d._hflip = True
+    d.camera.hflip = hflip
     return d
