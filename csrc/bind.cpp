@@ -248,7 +248,7 @@ struct SceneHolder {
     }
 
     void fill_common(SceneView& sv) {
-        sv.n_nodes = (int)np_nodes.shape(0);
+        sv.n_nodes = (int)np_nodes.shape(0) - 1;  // last row = sentinel
         sv.n_prims = (int)np_prims.shape(0);
         sv.n_objs = np_objs.ndim() == 2 ? (int)np_objs.shape(0) : 0;
         sv.n_bsdfs = (int)bsdfs.size();

@@ -58,7 +58,13 @@ struct HitRecord {
 };
 
 // Closest-hit traversal.  `prim_obj[i]` carries PRIM_SPHERE_BIT.
-// Templated on node-fetch so kernels can interpose an LDS-cached top tree.
+//
+// MI355X latency design: the node array carries ONE zero sentinel row past
+// n_nodes, so each iteration can speculatively issue loads of BOTH possible
+// successors (DFS next = i+1, skip target) while the current node's slab
+// test and leaf processing execute.  The walk is latency-bound on L2/L3
+// (VALUBusy ~4% measured), so doubling memory-level parallelism on the
+// dependent chain buys more than the wasted second fetch costs.
 HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
                                const Prim* prims, const uint32_t* prim_obj,
                                const Ray& ray, float tmax = MAX_DIST) {
@@ -67,28 +73,31 @@ HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
     Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     int i = 0;
+    BVHNode nd = nodes[0];
     while (i < n_nodes) {
-        const BVHNode nd = nodes[i];
+        int cnt = float_as_int(nd.hi.w);
+        int i_hit = i + 1;
+        int i_miss = cnt > 0 ? i + 1 : -cnt;
+        // speculative fetches of both successors (sentinel row keeps the
+        // clamped indices in bounds)
+        BVHNode nd_hit = nodes[i_hit < n_nodes ? i_hit : n_nodes];
+        BVHNode nd_miss = nodes[i_miss < n_nodes ? i_miss : n_nodes];
         float t_near;
         bool hit_box = nd.aabb().intersect(inv_d, o_div, rec.t, t_near);
-        int cnt = float_as_int(nd.hi.w);
-        if (hit_box) {
-            if (cnt > 0) {  // leaf
-                int base = nd.prim_base();
-                for (int k = 0; k < cnt; ++k) {
-                    int pid = base + k;
-                    bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
-                    float u, v;
-                    float t = intersect_prim(prims[pid], sph, ray, u, v);
-                    if (t > EPSILON && t < rec.t) {
-                        rec.t = t; rec.u = u; rec.v = v; rec.prim_idx = pid;
-                    }
+        if (hit_box & (cnt > 0)) {
+            int base = float_as_int(nd.lo.w);
+            for (int k = 0; k < cnt; ++k) {
+                int pid = base + k;
+                bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+                float u, v;
+                float t = intersect_prim(prims[pid], sph, ray, u, v);
+                if (t > EPSILON && t < rec.t) {
+                    rec.t = t; rec.u = u; rec.v = v; rec.prim_idx = pid;
                 }
             }
-            ++i;                       // descend / next DFS node
-        } else {
-            i = cnt > 0 ? i + 1 : -cnt;  // leaf miss -> next; internal miss -> skip
         }
+        i = hit_box ? i_hit : i_miss;
+        nd = hit_box ? nd_hit : nd_miss;
     }
     if (rec.prim_idx < 0) rec.t = MAX_DIST;
     return rec;
@@ -101,26 +110,27 @@ HD bool occlusion_test_bvh(const BVHNode* nodes, int n_nodes,
     Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     int i = 0;
+    BVHNode nd = nodes[0];
     while (i < n_nodes) {
-        const BVHNode nd = nodes[i];
+        int cnt = float_as_int(nd.hi.w);
+        int i_hit = i + 1;
+        int i_miss = cnt > 0 ? i + 1 : -cnt;
+        BVHNode nd_hit = nodes[i_hit < n_nodes ? i_hit : n_nodes];
+        BVHNode nd_miss = nodes[i_miss < n_nodes ? i_miss : n_nodes];
         float t_near;
         bool hit_box = nd.aabb().intersect(inv_d, o_div, tmax, t_near);
-        int cnt = float_as_int(nd.hi.w);
-        if (hit_box) {
-            if (cnt > 0) {
-                int base = nd.prim_base();
-                for (int k = 0; k < cnt; ++k) {
-                    int pid = base + k;
-                    bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
-                    float u, v;
-                    float t = intersect_prim(prims[pid], sph, ray, u, v);
-                    if (t > EPSILON && t < tmax) return true;
-                }
+        if (hit_box & (cnt > 0)) {
+            int base = float_as_int(nd.lo.w);
+            for (int k = 0; k < cnt; ++k) {
+                int pid = base + k;
+                bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+                float u, v;
+                float t = intersect_prim(prims[pid], sph, ray, u, v);
+                if (t > EPSILON && t < tmax) return true;
             }
-            ++i;
-        } else {
-            i = cnt > 0 ? i + 1 : -cnt;
         }
+        i = hit_box ? i_hit : i_miss;
+        nd = hit_box ? nd_hit : nd_miss;
     }
     return false;
 }

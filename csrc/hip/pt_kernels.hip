@@ -155,17 +155,19 @@ int launch_render(const SceneView& sv, float* accum, float* var,
             const char* e = getenv("HIPPT_WAVES");
             return e ? atoi(e) : 0;
         }();
-        // measured on MI355X (kitchen 1080p): waves 5 > 4 > default(3); the
-        // kernel is memory-latency-bound (VALUBusy ~4%), so occupancy wins
-        // even at 39 VGPR spills.  HIPPT_WAVES overrides for A/B.
-        if (waves == 6)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        // measured on MI355X (kitchen 1080p, post NaN-fix): 129/154/168/180
+        // Msps at waves 3/4/5/6 — memory-latency-bound, occupancy wins even
+        // with VGPR spills.  HIPPT_WAVES overrides for A/B.
+        if (waves == 8)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        else if (waves == 5)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         else if (waves == 4)
             hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         else if (waves == 3)
             hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 0>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
         break;
     }
     }

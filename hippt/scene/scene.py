@@ -308,6 +308,9 @@ class Scene:
         nodes, order, stats = C.build_bvh(prims, prim_obj, cfg.max_leaf, cfg.overlap_w,
                                           cfg.use_sbvh, cfg.ref_unsplit)
         self.bvh_stats = stats
+        # one zero sentinel row past the tree: traversal speculatively fetches
+        # both successor nodes per step (csrc/core/bvh.h)
+        nodes = np.concatenate([nodes, np.zeros((1, 8), np.float32)])
         prims = np.ascontiguousarray(prims[order])
         attrs = np.ascontiguousarray(attrs[order])
         prim_obj = np.ascontiguousarray(prim_obj[order])
