@@ -59,12 +59,11 @@ struct HitRecord {
 
 // Closest-hit traversal.  `prim_obj[i]` carries PRIM_SPHERE_BIT.
 //
-// MI355X latency design: the node array carries ONE zero sentinel row past
-// n_nodes, so each iteration can speculatively issue loads of BOTH possible
-// successors (DFS next = i+1, skip target) while the current node's slab
-// test and leaf processing execute.  The walk is latency-bound on L2/L3
-// (VALUBusy ~4% measured), so doubling memory-level parallelism on the
-// dependent chain buys more than the wasted second fetch costs.
+// A speculative dual-fetch variant (load both successors per step through a
+// sentinel row) was measured 18% SLOWER on MI355X across megakernel and
+// wavefront (extra fetch traffic + register pressure beat the added MLP;
+// profiles/README.md), so the walk stays plain: one 32-byte node load per
+// step, branchless index advance.
 HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
                                const Prim* prims, const uint32_t* prim_obj,
                                const Ray& ray, float tmax = MAX_DIST) {
@@ -73,15 +72,9 @@ HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
     Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     int i = 0;
-    BVHNode nd = nodes[0];
     while (i < n_nodes) {
+        const BVHNode nd = nodes[i];
         int cnt = float_as_int(nd.hi.w);
-        int i_hit = i + 1;
-        int i_miss = cnt > 0 ? i + 1 : -cnt;
-        // speculative fetches of both successors (sentinel row keeps the
-        // clamped indices in bounds)
-        BVHNode nd_hit = nodes[i_hit < n_nodes ? i_hit : n_nodes];
-        BVHNode nd_miss = nodes[i_miss < n_nodes ? i_miss : n_nodes];
         float t_near;
         bool hit_box = nd.aabb().intersect(inv_d, o_div, rec.t, t_near);
         if (hit_box & (cnt > 0)) {
@@ -96,8 +89,7 @@ HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
                 }
             }
         }
-        i = hit_box ? i_hit : i_miss;
-        nd = hit_box ? nd_hit : nd_miss;
+        i = hit_box | (cnt > 0) ? i + 1 : -cnt;
     }
     if (rec.prim_idx < 0) rec.t = MAX_DIST;
     return rec;
@@ -110,13 +102,9 @@ HD bool occlusion_test_bvh(const BVHNode* nodes, int n_nodes,
     Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     int i = 0;
-    BVHNode nd = nodes[0];
     while (i < n_nodes) {
+        const BVHNode nd = nodes[i];
         int cnt = float_as_int(nd.hi.w);
-        int i_hit = i + 1;
-        int i_miss = cnt > 0 ? i + 1 : -cnt;
-        BVHNode nd_hit = nodes[i_hit < n_nodes ? i_hit : n_nodes];
-        BVHNode nd_miss = nodes[i_miss < n_nodes ? i_miss : n_nodes];
         float t_near;
         bool hit_box = nd.aabb().intersect(inv_d, o_div, tmax, t_near);
         if (hit_box & (cnt > 0)) {
@@ -129,8 +117,7 @@ HD bool occlusion_test_bvh(const BVHNode* nodes, int n_nodes,
                 if (t > EPSILON && t < tmax) return true;
             }
         }
-        i = hit_box ? i_hit : i_miss;
-        nd = hit_box ? nd_hit : nd_miss;
+        i = hit_box | (cnt > 0) ? i + 1 : -cnt;
     }
     return false;
 }
