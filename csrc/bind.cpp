@@ -43,7 +43,7 @@ Vec3 to_vec3(const std::vector<float>& v) { return {v[0], v[1], v[2]}; }
 
 struct SceneHolder {
     // ----- host data (kept alive / owned)
-    farr np_prims, np_attrs, np_nodes;
+    farr np_prims, np_attrs, np_nodes, np_nodes4;
     uarr np_prim_obj;
     iarr np_objs;
     std::vector<BsdfParams> bsdfs;
@@ -84,14 +84,17 @@ struct SceneHolder {
         has_dev = false;
     }
 
-    void set_geometry(farr prims, farr attrs, uarr prim_obj, farr nodes) {
+    void set_geometry(farr prims, farr attrs, uarr prim_obj, farr nodes, farr nodes4) {
         if (prims.ndim() != 2 || prims.shape(1) != 12) throw std::runtime_error("prims must be (n,12)");
         if (attrs.ndim() != 2 || attrs.shape(1) != 16) throw std::runtime_error("attrs must be (n,16)");
         if (nodes.ndim() != 2 || nodes.shape(1) != 8) throw std::runtime_error("nodes must be (m,8)");
+        if (nodes4.ndim() == 2 && nodes4.shape(0) > 0 && nodes4.shape(1) != 32)
+            throw std::runtime_error("nodes4 must be (m,32)");
         np_prims = std::move(prims);
         np_attrs = std::move(attrs);
         np_prim_obj = std::move(prim_obj);
         np_nodes = std::move(nodes);
+        np_nodes4 = std::move(nodes4);
     }
 
     void set_objects(iarr objs) {
@@ -249,6 +252,7 @@ struct SceneHolder {
 
     void fill_common(SceneView& sv) {
         sv.n_nodes = (int)np_nodes.shape(0) - 1;  // last row = sentinel
+        sv.n_nodes4 = np_nodes4.ndim() == 2 ? (int)np_nodes4.shape(0) : 0;
         sv.n_prims = (int)np_prims.shape(0);
         sv.n_objs = np_objs.ndim() == 2 ? (int)np_objs.shape(0) : 0;
         sv.n_bsdfs = (int)bsdfs.size();
@@ -264,6 +268,7 @@ struct SceneHolder {
     void finalize() {
         fill_common(host_sv);
         host_sv.nodes = (const BVHNode*)np_nodes.data();
+        host_sv.nodes4 = host_sv.n_nodes4 > 0 ? (const BVH4Node*)np_nodes4.data() : nullptr;
         host_sv.prims = (const Prim*)np_prims.data();
         host_sv.attrs = (const PrimAttr*)np_attrs.data();
         host_sv.prim_obj = np_prim_obj.data();
@@ -294,6 +299,8 @@ struct SceneHolder {
         finalize();
         fill_common(dev_sv);
         dev_sv.nodes = upload_vec((const BVHNode*)np_nodes.data(), np_nodes.shape(0));
+        dev_sv.nodes4 = dev_sv.n_nodes4 > 0
+            ? upload_vec((const BVH4Node*)np_nodes4.data(), np_nodes4.shape(0)) : nullptr;
         dev_sv.prims = upload_vec((const Prim*)np_prims.data(), np_prims.shape(0));
         dev_sv.attrs = upload_vec((const PrimAttr*)np_attrs.data(), np_attrs.shape(0));
         dev_sv.prim_obj = upload_vec(np_prim_obj.data(), np_prim_obj.size());
@@ -405,6 +412,52 @@ py::tuple py_build_bvh(farr prims, uarr prim_obj, int max_leaf, float overlap_w,
     return py::make_tuple(nodes, order, stats);
 }
 
+// Collapse a (m,8) skip-link node array into a (k,32) BVH4 array (bvh4.h).
+py::tuple py_collapse_bvh4(farr nodes) {
+    if (nodes.ndim() != 2 || nodes.shape(1) != 8) throw std::runtime_error("nodes must be (m,8)");
+    std::vector<BVHNode> bin((size_t)nodes.shape(0));
+    std::memcpy(bin.data(), nodes.data(), bin.size() * sizeof(BVHNode));
+    int depth4 = 0;
+    std::vector<BVH4Node> n4;
+    {
+        py::gil_scoped_release rel;
+        n4 = collapse_bvh4(bin, &depth4);
+    }
+    if (3 * depth4 > BVH4_STACK)
+        throw std::runtime_error("BVH4 depth exceeds traversal stack bound");
+    farr out({(py::ssize_t)n4.size(), (py::ssize_t)32});
+    std::memcpy(out.mutable_data(), n4.data(), n4.size() * sizeof(BVH4Node));
+    return py::make_tuple(out, depth4);
+}
+
+// Traversal self-test: compare binary skip-link vs 4-wide ordered traversal
+// (closest hit t/prim and occlusion verdict) on caller-supplied rays.
+// Returns the number of mismatching rays.
+int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
+                     farr ray_o, farr ray_d, float tmax) {
+    int n_nodes = (int)nodes.shape(0);
+    const BVHNode* bn = (const BVHNode*)nodes.data();
+    const BVH4Node* n4 = (const BVH4Node*)nodes4.data();
+    const Prim* pr = (const Prim*)prims.data();
+    const uint32_t* po = prim_obj.data();
+    int bad = 0;
+    int nr = (int)ray_o.shape(0);
+    for (int i = 0; i < nr; ++i) {
+        Ray r;
+        r.o = {ray_o.at(i, 0), ray_o.at(i, 1), ray_o.at(i, 2)};
+        r.d = {ray_d.at(i, 0), ray_d.at(i, 1), ray_d.at(i, 2)};
+        HitRecord a = ray_intersect_bvh(bn, n_nodes, pr, po, r, tmax);
+        HitRecord b = ray_intersect_bvh4(n4, pr, po, r, tmax);
+        bool occ_a = occlusion_test_bvh(bn, n_nodes, pr, po, r, tmax);
+        bool occ_b = occlusion_test_bvh4(n4, pr, po, r, tmax);
+        // prim index may differ only on exact t ties; compare t and object
+        bool hit_match = (a.prim_idx < 0) == (b.prim_idx < 0) &&
+                         (a.prim_idx < 0 || fabsf(a.t - b.t) <= 1e-5f * fmaxf(1.f, a.t));
+        if (!hit_match || occ_a != occ_b) ++bad;
+    }
+    return bad;
+}
+
 } // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -437,6 +490,8 @@ PYBIND11_MODULE(_C, m) {
     m.def("build_bvh", &py_build_bvh, py::arg("prims"), py::arg("prim_obj"),
           py::arg("max_leaf") = 4, py::arg("overlap_w") = 0.f,
           py::arg("use_sbvh") = false, py::arg("ref_unsplit") = true);
+    m.def("collapse_bvh4", &py_collapse_bvh4, py::arg("nodes"));
+    m.def("bvh4_selftest", &py_bvh4_selftest);
 
     m.def("dev_synchronize", [] { HIP_OK(dev_synchronize()); });
     m.def("dev_set_device", [](int d) { HIP_OK(dev_set_device(d)); });

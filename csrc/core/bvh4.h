@@ -1,0 +1,228 @@
+// bvh4.h — 4-wide BVH node layout + ordered short-stack traversal.
+//
+// Capability parity: same traversal semantics as the binary skip-link walk in
+// bvh.h (reference src/renderer/tracing_func.cuh:44-181), but the node is
+// re-designed for CDNA4 instead of copied: the binary walk is a serial
+// dependent-load chain (measured VALUBusy ~4% on MI355X — the megakernel is
+// latency-bound on L2/HBM, profiles/README.md), so the fix is to shorten the
+// chain, not to widen the fetch of a single step (speculative dual-fetch was
+// measured -18%).  A 4-wide node:
+//   * is one 128-byte record = 8 independent 16-byte loads per step — the
+//     memory parallelism lives INSIDE one step of the dependent chain,
+//   * halves tree depth (half as many dependent steps per ray),
+//   * tests 4 AABBs per step with pure VALU work, which is nearly free at 4%
+//     VALU utilization,
+//   * enables ordered (front-to-back) traversal with a short stack — the
+//     DFS skip-link layout cannot reorder children, a 4-wide stack walk can,
+//     and each stack entry carries t_near so stale far subtrees are culled
+//     on pop after the hit distance tightens.
+//
+// Box storage is SoA-within-node (lo_x[4], lo_y[4], ... ) so each of the 4
+// slab tests reads stride-16 floats from the same 2 cache lines.
+#pragma once
+#include "bvh.h"
+
+namespace hippt {
+
+// 128-byte 4-wide node.
+//   child[c] >= 0 : internal child, index of a BVH4Node
+//   child[c] <  0 : leaf child, prim_base = ~child[c], prim count = cnt[c]
+//   cnt[c] == 0 AND child[c] == 0: empty slot (box is inverted: never hits)
+struct alignas(16) BVH4Node {
+    float lo_x[4], lo_y[4], lo_z[4];
+    float hi_x[4], hi_y[4], hi_z[4];
+    int32_t child[4];
+    int32_t cnt[4];
+};
+static_assert(sizeof(BVH4Node) == 128, "BVH4Node must be 128 bytes");
+
+constexpr int BVH4_STACK = 64;  // >= 3 * max collapsed depth; checked at build
+
+// Intersect prims [base, base+cnt) and tighten rec.
+HD void bvh4_leaf_hit(const Prim* prims, const uint32_t* prim_obj,
+                      const Ray& ray, int base, int cnt, HitRecord& rec) {
+    for (int k = 0; k < cnt; ++k) {
+        int pid = base + k;
+        bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+        float u, v;
+        float t = intersect_prim(prims[pid], sph, ray, u, v);
+        if (t > EPSILON && t < rec.t) {
+            rec.t = t; rec.u = u; rec.v = v; rec.prim_idx = pid;
+        }
+    }
+}
+
+// Closest-hit ordered traversal.
+HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
+                                const Prim* prims, const uint32_t* prim_obj,
+                                const Ray& ray, float tmax = MAX_DIST) {
+    HitRecord rec;
+    rec.t = tmax;
+    const Vec3 inv_d = safe_rcp_dir(ray.d);
+    const Vec3 o_div = ray.o * inv_d;
+    // stack entry: (t_near bits << 32) | node index.  t_near >= 0 so the
+    // float bit pattern orders like the float; on pop, entries whose t_near
+    // is already beyond the current hit are skipped without a node load.
+    uint64_t stack[BVH4_STACK];
+    int sp = 0;
+    int cur = 0;
+    while (true) {
+        const BVH4Node nd = nodes[cur];
+        // 4 independent slab tests; keys sort hit children front-to-back.
+        // key = (t_near bits & ~3) | slot — low 2 bits carry the slot index.
+        uint32_t keys[4];
+        int nhit = 0;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            float t0x = fmaf(nd.lo_x[c], inv_d.x, -o_div.x);
+            float t1x = fmaf(nd.hi_x[c], inv_d.x, -o_div.x);
+            float t0y = fmaf(nd.lo_y[c], inv_d.y, -o_div.y);
+            float t1y = fmaf(nd.hi_y[c], inv_d.y, -o_div.y);
+            float t0z = fmaf(nd.lo_z[c], inv_d.z, -o_div.z);
+            float t1z = fmaf(nd.hi_z[c], inv_d.z, -o_div.z);
+            float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                fmaxf(fminf(t0z, t1z), 0.f));
+            float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                fminf(fmaxf(t0z, t1z), rec.t));
+            if (enter <= exit_) keys[nhit++] = (float_as_uint(enter) & ~3u) | (uint32_t)c;
+        }
+        // sort up to 4 keys ascending (nearest first): sorting network
+        if (nhit > 1) {
+            if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+            if (nhit > 2) {
+                if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                if (nhit > 3) {
+                    if (keys[2] > keys[3]) { uint32_t t = keys[2]; keys[2] = keys[3]; keys[3] = t; }
+                    if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                    if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                }
+            }
+        }
+        // near->far: leaves intersect inline (tightening rec.t before far
+        // subtrees are entered); internals: nearest continues, rest pushed.
+        int next = -1;
+        for (int k = 0; k < nhit; ++k) {
+            int c = (int)(keys[k] & 3u);
+            int ch = nd.child[c];
+            int pc = nd.cnt[c];
+            if (ch < 0) {
+                bvh4_leaf_hit(prims, prim_obj, ray, ~ch, pc, rec);
+            } else if (next < 0) {
+                next = ch;
+            } else {
+                stack[sp++] = ((uint64_t)(keys[k] & ~3u) << 32) | (uint32_t)ch;
+            }
+        }
+        if (next >= 0) { cur = next; continue; }
+        // pop, skipping subtrees now beyond the hit distance
+        for (;;) {
+            if (sp == 0) {
+                if (rec.prim_idx < 0) rec.t = MAX_DIST;
+                return rec;
+            }
+            uint64_t e = stack[--sp];
+            if (uint_as_float((uint32_t)(e >> 32)) < rec.t) { cur = (int)(uint32_t)e; break; }
+        }
+    }
+}
+
+// Any-hit occlusion test: returns true if something blocks [EPSILON, tmax].
+// No ordering (any hit ends the walk) — children are pushed unordered.
+HD bool occlusion_test_bvh4(const BVH4Node* nodes,
+                            const Prim* prims, const uint32_t* prim_obj,
+                            const Ray& ray, float tmax) {
+    const Vec3 inv_d = safe_rcp_dir(ray.d);
+    const Vec3 o_div = ray.o * inv_d;
+    int stack[BVH4_STACK];
+    int sp = 0;
+    int cur = 0;
+    while (true) {
+        const BVH4Node nd = nodes[cur];
+        int next = -1;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            float t0x = fmaf(nd.lo_x[c], inv_d.x, -o_div.x);
+            float t1x = fmaf(nd.hi_x[c], inv_d.x, -o_div.x);
+            float t0y = fmaf(nd.lo_y[c], inv_d.y, -o_div.y);
+            float t1y = fmaf(nd.hi_y[c], inv_d.y, -o_div.y);
+            float t0z = fmaf(nd.lo_z[c], inv_d.z, -o_div.z);
+            float t1z = fmaf(nd.hi_z[c], inv_d.z, -o_div.z);
+            float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                fmaxf(fminf(t0z, t1z), 0.f));
+            float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                fminf(fmaxf(t0z, t1z), tmax));
+            if (enter > exit_) continue;
+            int ch = nd.child[c];
+            if (ch < 0) {
+                int base = ~ch, pc = nd.cnt[c];
+                for (int k = 0; k < pc; ++k) {
+                    int pid = base + k;
+                    bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+                    float u, v;
+                    float t = intersect_prim(prims[pid], sph, ray, u, v);
+                    if (t > EPSILON && t < tmax) return true;
+                }
+            } else if (next < 0) {
+                next = ch;
+            } else {
+                stack[sp++] = ch;
+            }
+        }
+        if (next >= 0) { cur = next; continue; }
+        if (sp == 0) return false;
+        cur = stack[--sp];
+    }
+}
+
+// Node-visit / prim-test counting walk for the BVH-cost visualizer
+// (reference pt_impl/bvh_cost.cu:38-101; counts reflect the traversal that
+// actually runs, i.e. the 4-wide one).
+HD Vec2 bvh4_cost(const BVH4Node* nodes, const Prim* prims,
+                  const uint32_t* prim_obj, const Ray& ray) {
+    const Vec3 inv_d = safe_rcp_dir(ray.d);
+    const Vec3 o_div = ray.o * inv_d;
+    float best_t = MAX_DIST;
+    int node_visits = 0, prim_tests = 0;
+    int stack[BVH4_STACK];
+    int sp = 0;
+    int cur = 0;
+    while (true) {
+        const BVH4Node nd = nodes[cur];
+        ++node_visits;
+        int next = -1;
+        for (int c = 0; c < 4; ++c) {
+            float t0x = fmaf(nd.lo_x[c], inv_d.x, -o_div.x);
+            float t1x = fmaf(nd.hi_x[c], inv_d.x, -o_div.x);
+            float t0y = fmaf(nd.lo_y[c], inv_d.y, -o_div.y);
+            float t1y = fmaf(nd.hi_y[c], inv_d.y, -o_div.y);
+            float t0z = fmaf(nd.lo_z[c], inv_d.z, -o_div.z);
+            float t1z = fmaf(nd.hi_z[c], inv_d.z, -o_div.z);
+            float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                fmaxf(fminf(t0z, t1z), 0.f));
+            float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                fminf(fmaxf(t0z, t1z), best_t));
+            if (enter > exit_) continue;
+            int ch = nd.child[c];
+            if (ch < 0) {
+                int base = ~ch, pc = nd.cnt[c];
+                for (int k = 0; k < pc; ++k) {
+                    ++prim_tests;
+                    float u, v;
+                    bool sph = (prim_obj[base + k] & PRIM_SPHERE_BIT) != 0;
+                    float t = intersect_prim(prims[base + k], sph, ray, u, v);
+                    if (t > EPSILON && t < best_t) best_t = t;
+                }
+            } else if (next < 0) {
+                next = ch;
+            } else {
+                stack[sp++] = ch;
+            }
+        }
+        if (next >= 0) { cur = next; continue; }
+        if (sp == 0) return {(float)node_visits, (float)prim_tests};
+        cur = stack[--sp];
+    }
+}
+
+} // namespace hippt

@@ -41,7 +41,7 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
 
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 2 + 8 && b < sv.md.max_depth; ++iter) {
-        HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+        HitRecord hit = scene_intersect(sv, ray);
         if (hit.prim_idx < 0) {
             // miss -> environment map with MIS against the cosine NEE pdf
             if (sv.env_emitter >= 0) {
@@ -94,7 +94,7 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
                 if (!f.is_zero()) {
                     Ray sh_ray(fmadd(wi, EPSILON, pos), wi);
                     float sh_max = (sv.emitters[ei].type == EM_ENVMAP ? ENVMAP_DIST : dist) - 2.f * EPSILON;
-                    if (!occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, sh_ray, sh_max)) {
+                    if (!scene_occluded(sv, sh_ray, sh_max)) {
                         float light_pdf = er.pdf * epdf;
                         float w = er.delta ? 1.f
                                            : mis_weight(light_pdf, bsdf_pdf(bsdf, -ray.d, wi, it, sv.textures));
@@ -138,13 +138,14 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
 
 // Depth renderer: distance of the primary hit (reference pt_impl/depth.cu).
 HD float trace_depth(const SceneView& sv, const Ray& ray) {
-    HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+    HitRecord hit = scene_intersect(sv, ray);
     return hit.prim_idx >= 0 ? hit.t : 0.f;
 }
 
 // BVH-cost visualizer: counts node visits (x) and primitive tests (y) for a
 // primary ray (reference pt_impl/bvh_cost.cu:38-101).
 HD Vec2 trace_bvh_cost(const SceneView& sv, const Ray& ray) {
+    if (sv.nodes4) return bvh4_cost(sv.nodes4, sv.prims, sv.prim_obj, ray);
     Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     float best_t = MAX_DIST;

@@ -46,7 +46,7 @@ HD Vec3 transmittance_estimate(const SceneView& sv, Vec3 from, const Vec3& wi, f
         Ray r(fmadd(wi, EPSILON, from), wi);
         remaining -= EPSILON;
         if (remaining <= EPSILON) break;
-        HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, r, remaining);
+        HitRecord hit = scene_intersect(sv, r, remaining);
         float seg = hit.prim_idx >= 0 ? hit.t : remaining;
         int med = stack.current();
         if (med >= 0) {
@@ -79,7 +79,7 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp) {
 
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 3 + 16 && b < sv.md.max_depth; ++iter) {
-        HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+        HitRecord hit = scene_intersect(sv, ray);
         float t_surf = hit.prim_idx >= 0 ? hit.t : MAX_DIST;
 
         // ---- medium flight

@@ -43,7 +43,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
         Vec3 wc = to_cam * (1.f / dist);
         float cos_c = fmaxf(1e-6f, (-wc).dot(cam_fwd));
         Ray sh(fmadd(wc, EPSILON, pos), wc);
-        if (occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, sh, dist - 2.f * EPSILON))
+        if (scene_occluded(sv, sh, dist - 2.f * EPSILON))
             return;
         float W = sv.cam.focal * sv.cam.focal / (cos_c * cos_c * cos_c);
         Vec3 val = f_times_cos * (W / d2) * inv_npix * caustic_scaling;
@@ -69,7 +69,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
     PathStats st;
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 2 + 8 && b < sv.md.max_depth; ++iter) {
-        HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+        HitRecord hit = scene_intersect(sv, ray);
         if (hit.prim_idx < 0) break;
         Vec3 pos = ray.at(hit.t);
         uint32_t po = sv.prim_obj[hit.prim_idx];

@@ -8,6 +8,7 @@
 // (MaxDepthParams incl. ToF min/max time).
 #pragma once
 #include "bvh.h"
+#include "bvh4.h"
 #include "bsdf.h"
 #include "emitter.h"
 #include "camera.h"
@@ -39,8 +40,10 @@ struct MaxDepthParams {
 };
 
 struct SceneView {
-    // geometry + BVH
+    // geometry + BVH (binary skip-link kept for the cost visualizer and as
+    // the collapse source; traversal runs on the 4-wide tree when present)
     const BVHNode* nodes; int n_nodes;
+    const BVH4Node* nodes4; int n_nodes4;
     const Prim* prims; const PrimAttr* attrs; const uint32_t* prim_obj; int n_prims;
     const ObjInfo* objs; int n_objs;
     // materials / emitters / textures
@@ -64,5 +67,19 @@ struct SceneView {
     HD uint32_t obj_of_prim(int pid) const { return prim_obj[pid] & PRIM_OBJ_MASK; }
     HD bool prim_is_sphere(int pid) const { return (prim_obj[pid] & PRIM_SPHERE_BIT) != 0; }
 };
+
+// Traversal entry points used by every integrator/kernel: 4-wide ordered
+// walk when the collapsed tree is present, binary skip-link walk otherwise.
+HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
+                             float tmax = MAX_DIST) {
+    if (sv.nodes4)
+        return ray_intersect_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
+    return ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray, tmax);
+}
+HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax) {
+    if (sv.nodes4)
+        return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
+    return occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray, tmax);
+}
 
 } // namespace hippt

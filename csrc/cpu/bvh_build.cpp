@@ -221,3 +221,94 @@ BVHBuildResult build_bvh(const Prim* prims, const uint32_t* prim_obj, int n,
 }
 
 } // namespace hippt
+
+// ---------------------------------------------------------------- BVH4 ----
+namespace hippt {
+namespace {
+
+// Recover the two children of internal skip-link node i:
+// left = i+1; right = skip(left) for internal left, left+1 for leaf left.
+inline int right_child(const std::vector<BVHNode>& bin, int left) {
+    return bin[left].is_leaf() ? left + 1 : bin[left].skip();
+}
+
+struct Collapser {
+    const std::vector<BVHNode>& bin;
+    std::vector<BVH4Node> out;
+    int max_depth = 0;
+
+    float area(int i) const {
+        Vec3 e = bin[i].hi.xyz() - bin[i].lo.xyz();
+        return 2.f * (e.x * e.y + e.y * e.z + e.z * e.x);
+    }
+
+    // emit the BVH4 node for binary node bi (must be internal unless root leaf)
+    int emit(int bi, int depth) {
+        max_depth = std::max(max_depth, depth);
+        int my = (int)out.size();
+        out.emplace_back();
+        int slots[4];
+        int n_slots = 0;
+        if (bin[bi].is_leaf()) {
+            slots[n_slots++] = bi;           // degenerate: whole tree is 1 leaf
+        } else {
+            int l = bi + 1, r = right_child(bin, l);
+            slots[n_slots++] = l;
+            slots[n_slots++] = r;
+            // expand the largest-area internal slot until 4 slots
+            while (n_slots < 4) {
+                int pick = -1;
+                float best = -1.f;
+                for (int s = 0; s < n_slots; ++s)
+                    if (!bin[slots[s]].is_leaf() && area(slots[s]) > best) {
+                        best = area(slots[s]);
+                        pick = s;
+                    }
+                if (pick < 0) break;
+                int p = slots[pick];
+                int pl = p + 1, pr = right_child(bin, pl);
+                slots[pick] = pl;
+                slots[n_slots++] = pr;
+            }
+        }
+        // fill node (children recursively, in slot order = DFS)
+        BVH4Node tmp{};
+        for (int c = 0; c < 4; ++c) {
+            if (c < n_slots) {
+                int s = slots[c];
+                tmp.lo_x[c] = bin[s].lo.x; tmp.lo_y[c] = bin[s].lo.y; tmp.lo_z[c] = bin[s].lo.z;
+                tmp.hi_x[c] = bin[s].hi.x; tmp.hi_y[c] = bin[s].hi.y; tmp.hi_z[c] = bin[s].hi.z;
+                if (bin[s].is_leaf()) {
+                    tmp.child[c] = ~bin[s].prim_base();
+                    tmp.cnt[c] = bin[s].prim_cnt();
+                } else {
+                    tmp.child[c] = emit(s, depth + 1);
+                    tmp.cnt[c] = 0;
+                }
+            } else {
+                // empty slot: degenerate far-away point box (a min/max slab
+                // test un-inverts an inverted box, so inversion can't be the
+                // never-hit encoding) + 0-prim leaf so a hit is still a no-op
+                tmp.lo_x[c] = tmp.lo_y[c] = tmp.lo_z[c] = 3.0e38f;
+                tmp.hi_x[c] = tmp.hi_y[c] = tmp.hi_z[c] = 3.0e38f;
+                tmp.child[c] = ~0;
+                tmp.cnt[c] = 0;
+            }
+        }
+        out[my] = tmp;
+        return my;
+    }
+};
+
+} // namespace
+
+std::vector<BVH4Node> collapse_bvh4(const std::vector<BVHNode>& bin, int* max_depth4) {
+    Collapser col{bin};
+    if (bin.empty()) return {};
+    col.out.reserve(bin.size() / 2 + 1);
+    col.emit(0, 1);
+    if (max_depth4) *max_depth4 = col.max_depth;
+    return col.out;
+}
+
+} // namespace hippt

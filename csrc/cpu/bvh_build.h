@@ -8,6 +8,7 @@
 #include <vector>
 #include <cstdint>
 #include "../core/bvh.h"
+#include "../core/bvh4.h"
 
 namespace hippt {
 
@@ -30,6 +31,14 @@ struct BVHBuildConfig {
 // prims/n: primitive array (triangles use v0/e1/e2; spheres per prim_obj bit31)
 BVHBuildResult build_bvh(const Prim* prims, const uint32_t* prim_obj, int n,
                          const BVHBuildConfig& cfg);
+
+// Collapse a DFS skip-link binary BVH (build_bvh/build_sbvh output) into a
+// 4-wide BVH (bvh4.h).  Children of each 4-wide node are found by repeatedly
+// expanding the largest-area internal slot of the binary pair until 4 slots
+// exist.  Returns nodes in DFS order (root = 0); max_depth4 reports the
+// collapsed tree depth (traversal stack bound = 3 * depth).
+std::vector<BVH4Node> collapse_bvh4(const std::vector<BVHNode>& bin,
+                                    int* max_depth4 = nullptr);
 
 // SBVH: spatial-split BVH (Stich et al. style chopped binning); may duplicate
 // references, so prim_order can be longer than n.

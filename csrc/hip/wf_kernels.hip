@@ -62,7 +62,7 @@ void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed) {
     int px = i % st.w, py = i / st.w;
     Sampler sp(uint32_t(i), uint32_t(spp_idx) * SEED_SCALER + seed);
     Ray ray = sv.cam.gen_ray(px, py, sp);
-    HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+    HitRecord hit = scene_intersect(sv, ray);
     st.ray_o[i] = make_float4(ray.o.x, ray.o.y, ray.o.z, 0.f);
     st.ray_d[i] = make_float4(ray.d.x, ray.d.y, ray.d.z, 0.f);
     st.thp[i] = make_float4(1.f, 1.f, 1.f, uint_as_float(1u));  // prev_delta=1
@@ -279,7 +279,7 @@ void k_wf_shadow(SceneView sv, WfState st) {
     float4 od = st.sh_od[k];
     float4 dir = st.sh_dir[k];
     Ray ray(Vec3(od.x, od.y, od.z), Vec3(dir.x, dir.y, dir.z));
-    if (occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray, od.w)) return;
+    if (scene_occluded(sv, ray, od.w)) return;
     int i = (int)float_as_uint(dir.w);
     float4 v = st.sh_val[k];
     float4 l4 = st.L[i];
@@ -298,7 +298,7 @@ void k_wf_trace(SceneView sv, WfState st) {
     if (st.status[i] >> 24 >= DEAD) return;  // dead or terminated in shade
     float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i];
     Ray ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
-    HitRecord hit = ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray);
+    HitRecord hit = scene_intersect(sv, ray);
     st.hit[i] = make_float4(hit.t, hit.u, hit.v, int_as_float(hit.prim_idx));
     uint32_t status;
     if (hit.prim_idx < 0) {

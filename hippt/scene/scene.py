@@ -308,6 +308,14 @@ class Scene:
         nodes, order, stats = C.build_bvh(prims, prim_obj, cfg.max_leaf, cfg.overlap_w,
                                           cfg.use_sbvh, cfg.ref_unsplit)
         self.bvh_stats = stats
+        # 4-wide collapse of the binary tree: the traversal that actually runs
+        # (ordered short-stack walk over 128-byte nodes, csrc/core/bvh4.h)
+        import os
+        if os.environ.get("HIPPT_NO_BVH4"):   # A/B hook: binary skip-link walk
+            nodes4, depth4 = np.zeros((0, 32), np.float32), 0
+        else:
+            nodes4, depth4 = C.collapse_bvh4(nodes)
+        self.bvh_stats = dict(stats, n_nodes4=int(nodes4.shape[0]), depth4=int(depth4))
         # one zero sentinel row past the tree: traversal speculatively fetches
         # both successor nodes per step (csrc/core/bvh.h)
         nodes = np.concatenate([nodes, np.zeros((1, 8), np.float32)])
@@ -318,7 +326,7 @@ class Scene:
         new_of_old = np.empty(len(order), np.int64)
         new_of_old[order] = np.arange(len(order))
 
-        self.native.set_geometry(prims, attrs, prim_obj, nodes)
+        self.native.set_geometry(prims, attrs, prim_obj, nodes, nodes4)
         self.native.set_objects(objs)
         self._np = dict(prims=prims, attrs=attrs, prim_obj=prim_obj, nodes=nodes, objs=objs)
 

@@ -116,3 +116,77 @@ class TestRNG:
         assert not np.allclose(a, b)
         # but the means must agree (same scene)
         assert abs(a[..., :3].mean() - b[..., :3].mean()) < 0.05 * max(a[..., :3].mean(), 1e-9)
+
+
+class TestBVH4:
+    """4-wide collapsed BVH (csrc/core/bvh4.h) vs binary skip-link walk."""
+
+    def _build(self, n, seed, use_sbvh=False, spheres=False):
+        tris = _random_tris(n, seed=seed, spread=3.0)
+        prims = _to_prims(tris)
+        pobj = np.zeros(n, np.uint32)
+        if spheres:
+            # last quarter become spheres: v0 = center, v0.w = radius
+            k = n // 4
+            prims[-k:, :] = 0
+            rng = np.random.default_rng(seed + 7)
+            prims[-k:, 0:3] = rng.uniform(-3, 3, (k, 3)).astype(np.float32)
+            prims[-k:, 3] = rng.uniform(0.05, 0.4, k).astype(np.float32)
+            pobj[-k:] = 1 << 31
+        nodes, order, stats = C.build_bvh(prims, pobj, 4, 0.0, use_sbvh, True)
+        prims = np.ascontiguousarray(prims[order % n])
+        pobj = np.ascontiguousarray(pobj[order % n])
+        nodes4, depth4 = C.collapse_bvh4(nodes)
+        return prims, pobj, nodes, nodes4, depth4
+
+    def test_collapse_invariants(self):
+        prims, pobj, nodes, nodes4, depth4 = self._build(800, seed=11)
+        meta = nodes4.reshape(-1, 32)
+        child = meta[:, 24:28].view(np.int32)
+        cnt = meta[:, 28:32].view(np.int32)
+        # internal children point forward (DFS order), inside the array
+        internal = child > 0
+        assert (child[internal] < len(nodes4)).all()
+        # leaf prim ranges cover [0, n) exactly once
+        covered = np.zeros(len(prims), bool)
+        for i in range(len(nodes4)):
+            for c in range(4):
+                if child[i, c] < 0 and cnt[i, c] > 0:
+                    b, k = ~child[i, c], cnt[i, c]
+                    assert not covered[b:b + k].any()
+                    covered[b:b + k] = True
+        assert covered.all()
+        # stack bound honored
+        assert 3 * depth4 <= 64
+
+    def _rays(self, m, seed):
+        rng = np.random.default_rng(seed)
+        o = rng.uniform(-6, 6, (m, 3)).astype(np.float32)
+        d = rng.normal(size=(m, 3)).astype(np.float32)
+        d /= np.linalg.norm(d, axis=1, keepdims=True)
+        # include axis-aligned rays (zero direction components: the NaN-safe
+        # reciprocal path) and rays from far outside
+        d[:16] = np.eye(3, dtype=np.float32)[rng.integers(0, 3, 16)] * \
+            rng.choice([-1.0, 1.0], 16)[:, None]
+        return o, d
+
+    def test_traversal_agreement_bvh(self):
+        prims, pobj, nodes, nodes4, _ = self._build(800, seed=12)
+        o, d = self._rays(4000, seed=13)
+        assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7) == 0
+
+    def test_traversal_agreement_sbvh(self):
+        prims, pobj, nodes, nodes4, _ = self._build(600, seed=14, use_sbvh=True)
+        o, d = self._rays(4000, seed=15)
+        assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7) == 0
+
+    def test_traversal_agreement_spheres(self):
+        prims, pobj, nodes, nodes4, _ = self._build(400, seed=16, spheres=True)
+        o, d = self._rays(4000, seed=17)
+        assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7) == 0
+
+    def test_tiny_scene(self):
+        # single-leaf degenerate tree
+        prims, pobj, nodes, nodes4, depth4 = self._build(2, seed=18)
+        o, d = self._rays(500, seed=19)
+        assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7) == 0
