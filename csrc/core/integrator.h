@@ -145,7 +145,9 @@ HD float trace_depth(const SceneView& sv, const Ray& ray) {
 // BVH-cost visualizer: counts node visits (x) and primitive tests (y) for a
 // primary ray (reference pt_impl/bvh_cost.cu:38-101).
 HD Vec2 trace_bvh_cost(const SceneView& sv, const Ray& ray) {
-    if (sv.nodes4) return bvh4_cost(sv.nodes4, sv.prims, sv.prim_obj, ray);
+    return bvh4_cost(sv.nodes4, sv.prims, sv.prim_obj, ray);
+}
+HD Vec2 trace_bvh_cost_binary(const SceneView& sv, const Ray& ray) {
     Vec3 inv_d = safe_rcp_dir(ray.d);
     Vec3 o_div = ray.o * inv_d;
     float best_t = MAX_DIST;

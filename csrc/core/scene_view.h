@@ -68,18 +68,17 @@ struct SceneView {
     HD bool prim_is_sphere(int pid) const { return (prim_obj[pid] & PRIM_SPHERE_BIT) != 0; }
 };
 
-// Traversal entry points used by every integrator/kernel: 4-wide ordered
-// walk when the collapsed tree is present, binary skip-link walk otherwise.
+// Traversal entry points used by every integrator/kernel.  The 4-wide
+// ordered walk is the ONLY device path — keeping the binary walk as a
+// runtime fallback was measured to cost the megakernel ~11% (dead code +
+// its register pressure under __launch_bounds__); the binary tree stays as
+// the collapse source and for host-side self-tests (tests/test_core.py).
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST) {
-    if (sv.nodes4)
-        return ray_intersect_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
-    return ray_intersect_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray, tmax);
+    return ray_intersect_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
 }
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax) {
-    if (sv.nodes4)
-        return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
-    return occlusion_test_bvh(sv.nodes, sv.n_nodes, sv.prims, sv.prim_obj, ray, tmax);
+    return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
 }
 
 } // namespace hippt

@@ -18,12 +18,26 @@ namespace hippt {
 // ------------------------------------------------------------- PT megakernel
 // MINWAVES = __launch_bounds__ waves/SIMD floor (occupancy vs register trade;
 // 0 keeps the compiler's choice).  Runtime-selectable for A/B via HIPPT_WAVES.
+// XCD-aware tile swizzle: MI355X dispatches consecutive workgroups
+// round-robin over the 8 XCDs (each with its own L2).  Remapping the flat
+// block id so blocks congruent mod 8 cover one contiguous screen band gives
+// each XCD a compact working set (one band's BVH leaves + prims) instead of
+// a scatter of the whole frame.  Bijection on [0, n_tiles):
+//   xcd = flat % 8, id = xcd*(n/8) + min(xcd, n%8) + flat/8.
+__device__ inline int xcd_swizzle(int flat, int n_tiles) {
+    int xcd = flat & 7, local = flat >> 3;
+    int per = n_tiles >> 3, rem = n_tiles & 7;
+    return xcd * per + (xcd < rem ? xcd : rem) + local;
+}
+
 template <int RENDERER, int MINWAVES = 0>
 __global__ __launch_bounds__(256, MINWAVES)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed) {
-    const int px = blockIdx.x * 16 + threadIdx.x;
-    const int py = blockIdx.y * 16 + threadIdx.y;
+              int spp0, int nspp, uint32_t seed, int swiz) {
+    int tile = blockIdx.y * gridDim.x + blockIdx.x;
+    if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
+    const int px = (tile % gridDim.x) * 16 + threadIdx.x;
+    const int py = (tile / gridDim.x) * 16 + threadIdx.y;
     if (px >= sv.cam.w || py >= sv.cam.h) return;
     const size_t pix = size_t(py) * sv.cam.w + px;
 
@@ -114,6 +128,12 @@ int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling, void* stream) {
     hipStream_t st = (hipStream_t)stream;
+    // HIPPT_SWIZZLE=0 disables the XCD band swizzle (A/B hook; default on)
+    static int swiz_v = [] {
+        const char* e = getenv("HIPPT_SWIZZLE");
+        return e ? atoi(e) : 1;
+    }();
+    auto swiz = [] { return swiz_v; };
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
     dim3 grid((w + 15) / 16, (h + 15) / 16);
@@ -129,13 +149,13 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         break;
     }
     case R_VOLUME_PT:
-        hipLaunchKernelGGL(k_render<R_VOLUME_PT>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        hipLaunchKernelGGL(k_render<R_VOLUME_PT>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL(k_render<R_DEPTH>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        hipLaunchKernelGGL(k_render<R_DEPTH>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed, 0);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+        hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed, 0);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -159,15 +179,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         // Msps at waves 3/4/5/6 — memory-latency-bound, occupancy wins even
         // with VGPR spills.  HIPPT_WAVES overrides for A/B.
         if (waves == 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
         else if (waves == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
         else if (waves == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
         else if (waves == 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 0>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 0>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
         break;
     }
     }

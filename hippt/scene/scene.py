@@ -310,11 +310,7 @@ class Scene:
         self.bvh_stats = stats
         # 4-wide collapse of the binary tree: the traversal that actually runs
         # (ordered short-stack walk over 128-byte nodes, csrc/core/bvh4.h)
-        import os
-        if os.environ.get("HIPPT_NO_BVH4"):   # A/B hook: binary skip-link walk
-            nodes4, depth4 = np.zeros((0, 32), np.float32), 0
-        else:
-            nodes4, depth4 = C.collapse_bvh4(nodes)
+        nodes4, depth4 = C.collapse_bvh4(nodes)
         self.bvh_stats = dict(stats, n_nodes4=int(nodes4.shape[0]), depth4=int(depth4))
         # one zero sentinel row past the tree: traversal speculatively fetches
         # both successor nodes per step (csrc/core/bvh.h)
