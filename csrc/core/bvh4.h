@@ -52,10 +52,22 @@ HD void bvh4_leaf_hit(const Prim* prims, const uint32_t* prim_obj,
     }
 }
 
+// Traversal stack: bottom lds_n entries live in LDS (device kernels pass a
+// per-thread slot pointer into a __shared__ array, entry d at slot[d*256]
+// for 256-thread blocks), the rest spill to a private (scratch) overflow
+// array.  Scratch-backed stacks were measured to be the BVH4 bottleneck on
+// MI355X (the walk is latency-bound; per-step 8-byte scratch push/pops
+// thrash the vector caches), and the LDS allocation doubles as the
+// occupancy governor: LDSN entries/thread x 8 B x 256 threads of LDS per
+// block caps blocks/CU exactly where the scratch walk wants to run.
+// On the host (lds_n = 0) everything goes to the plain array.
+constexpr int BVH4_LDS_STRIDE = 256;  // all render blocks are 256 threads
+
 // Closest-hit ordered traversal.
 HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
                                 const Prim* prims, const uint32_t* prim_obj,
-                                const Ray& ray, float tmax = MAX_DIST) {
+                                const Ray& ray, float tmax = MAX_DIST,
+                                uint64_t* lds_slot = nullptr, int lds_n = 0) {
     HitRecord rec;
     rec.t = tmax;
     const Vec3 inv_d = safe_rcp_dir(ray.d);
@@ -111,7 +123,10 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
             } else if (next < 0) {
                 next = ch;
             } else {
-                stack[sp++] = ((uint64_t)(keys[k] & ~3u) << 32) | (uint32_t)ch;
+                uint64_t e = ((uint64_t)(keys[k] & ~3u) << 32) | (uint32_t)ch;
+                if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = e;
+                else stack[sp - lds_n] = e;
+                ++sp;
             }
         }
         if (next >= 0) { cur = next; continue; }
@@ -121,7 +136,8 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
                 if (rec.prim_idx < 0) rec.t = MAX_DIST;
                 return rec;
             }
-            uint64_t e = stack[--sp];
+            --sp;
+            uint64_t e = sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE] : stack[sp - lds_n];
             if (uint_as_float((uint32_t)(e >> 32)) < rec.t) { cur = (int)(uint32_t)e; break; }
         }
     }
@@ -131,7 +147,8 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
 // No ordering (any hit ends the walk) — children are pushed unordered.
 HD bool occlusion_test_bvh4(const BVH4Node* nodes,
                             const Prim* prims, const uint32_t* prim_obj,
-                            const Ray& ray, float tmax) {
+                            const Ray& ray, float tmax,
+                            uint64_t* lds_slot = nullptr, int lds_n = 0) {
     const Vec3 inv_d = safe_rcp_dir(ray.d);
     const Vec3 o_div = ray.o * inv_d;
     int stack[BVH4_STACK];
@@ -166,12 +183,15 @@ HD bool occlusion_test_bvh4(const BVH4Node* nodes,
             } else if (next < 0) {
                 next = ch;
             } else {
-                stack[sp++] = ch;
+                if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = (uint64_t)(uint32_t)ch;
+                else stack[sp - lds_n] = ch;
+                ++sp;
             }
         }
         if (next >= 0) { cur = next; continue; }
         if (sp == 0) return false;
-        cur = stack[--sp];
+        --sp;
+        cur = sp < lds_n ? (int)(uint32_t)lds_slot[sp * BVH4_LDS_STRIDE] : stack[sp - lds_n];
     }
 }
 

@@ -30,7 +30,7 @@ HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
 }
 
 // Full path trace for one camera ray. Returns radiance estimate.
-HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
+HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
     Vec3 L(0.f), thp(1.f);
     float prev_pdf = 0.f;       // BSDF pdf of the previous bounce (for MIS)
     bool prev_delta = true;     // camera rays count as delta
@@ -41,7 +41,7 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
 
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 2 + 8 && b < sv.md.max_depth; ++iter) {
-        HitRecord hit = scene_intersect(sv, ray);
+        HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
         if (hit.prim_idx < 0) {
             // miss -> environment map with MIS against the cosine NEE pdf
             if (sv.env_emitter >= 0) {
@@ -94,7 +94,7 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
                 if (!f.is_zero()) {
                     Ray sh_ray(fmadd(wi, EPSILON, pos), wi);
                     float sh_max = (sv.emitters[ei].type == EM_ENVMAP ? ENVMAP_DIST : dist) - 2.f * EPSILON;
-                    if (!scene_occluded(sv, sh_ray, sh_max)) {
+                    if (!scene_occluded(sv, sh_ray, sh_max, tc)) {
                         float light_pdf = er.pdf * epdf;
                         float w = er.delta ? 1.f
                                            : mis_weight(light_pdf, bsdf_pdf(bsdf, -ray.d, wi, it, sv.textures));
@@ -137,8 +137,8 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp) {
 }
 
 // Depth renderer: distance of the primary hit (reference pt_impl/depth.cu).
-HD float trace_depth(const SceneView& sv, const Ray& ray) {
-    HitRecord hit = scene_intersect(sv, ray);
+HD float trace_depth(const SceneView& sv, const Ray& ray, TravCtx tc = {}) {
+    HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
     return hit.prim_idx >= 0 ? hit.t : 0.f;
 }
 

@@ -39,6 +39,7 @@ HD void cross_boundary(VolStack& st, const ObjInfo& obj, const Vec3& d, const Ve
 // accumulating per-segment medium transmittance; reference
 // occlusion_transmittance_estimate, megakernel_vpt.cu:104-201).
 HD Vec3 transmittance_estimate(const SceneView& sv, Vec3 from, const Vec3& wi, float dist,
+                               TravCtx tc,
                                VolStack stack, Sampler& sp) {
     Vec3 tr(1.f);
     float remaining = dist;
@@ -46,7 +47,7 @@ HD Vec3 transmittance_estimate(const SceneView& sv, Vec3 from, const Vec3& wi, f
         Ray r(fmadd(wi, EPSILON, from), wi);
         remaining -= EPSILON;
         if (remaining <= EPSILON) break;
-        HitRecord hit = scene_intersect(sv, r, remaining);
+        HitRecord hit = scene_intersect(sv, r, remaining, tc);
         float seg = hit.prim_idx >= 0 ? hit.t : remaining;
         int med = stack.current();
         if (med >= 0) {
@@ -66,7 +67,7 @@ HD Vec3 transmittance_estimate(const SceneView& sv, Vec3 from, const Vec3& wi, f
     return tr;
 }
 
-HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp) {
+HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
     Vec3 L(0.f), thp(1.f);
     float prev_pdf = 0.f;
     bool prev_delta = true;
@@ -79,7 +80,7 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp) {
 
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 3 + 16 && b < sv.md.max_depth; ++iter) {
-        HitRecord hit = scene_intersect(sv, ray);
+        HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
         float t_surf = hit.prim_idx >= 0 ? hit.t : MAX_DIST;
 
         // ---- medium flight
@@ -113,7 +114,7 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp) {
                     float dist = to_l.length();
                     Vec3 wi = to_l * (1.f / fmaxf(dist, 1e-9f));
                     float fp = phase_eval(ph, ray.d.dot(wi));
-                    Vec3 tr = transmittance_estimate(sv, pos, wi, dist - EPSILON, stack, sp);
+                    Vec3 tr = transmittance_estimate(sv, pos, wi, dist - EPSILON, tc, stack, sp);
                     if (!tr.is_zero()) {
                         float light_pdf = er.pdf * epdf;
                         float w = er.delta ? 1.f : mis_weight(light_pdf, fp);
@@ -178,7 +179,7 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp) {
                     Vec3 f = bsdf_eval(bsdf, -ray.d, wi, it, sv.textures);
                     if (!f.is_zero()) {
                         float sh_max = (sv.emitters[ei].type == EM_ENVMAP ? ENVMAP_DIST : dist) - 2.f * EPSILON;
-                        Vec3 tr = transmittance_estimate(sv, pos, wi, sh_max, stack, sp);
+                        Vec3 tr = transmittance_estimate(sv, pos, wi, sh_max, tc, stack, sp);
                         if (!tr.is_zero()) {
                             float light_pdf = er.pdf * epdf;
                             float w = er.delta ? 1.f

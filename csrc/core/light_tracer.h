@@ -19,7 +19,7 @@ namespace hippt {
 
 // SPLAT_FN: void(int pix, Vec3 value) — device uses atomics, host plain adds.
 template <typename SPLAT_FN>
-HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat,
+HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat, TravCtx tc,
                               int spec_constraint, float caustic_scaling) {
     if (sv.n_emitters <= 0) return;
     float epdf;
@@ -43,7 +43,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
         Vec3 wc = to_cam * (1.f / dist);
         float cos_c = fmaxf(1e-6f, (-wc).dot(cam_fwd));
         Ray sh(fmadd(wc, EPSILON, pos), wc);
-        if (scene_occluded(sv, sh, dist - 2.f * EPSILON))
+        if (scene_occluded(sv, sh, dist - 2.f * EPSILON, tc))
             return;
         float W = sv.cam.focal * sv.cam.focal / (cos_c * cos_c * cos_c);
         Vec3 val = f_times_cos * (W / d2) * inv_npix * caustic_scaling;
@@ -69,7 +69,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
     PathStats st;
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 2 + 8 && b < sv.md.max_depth; ++iter) {
-        HitRecord hit = scene_intersect(sv, ray);
+        HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
         if (hit.prim_idx < 0) break;
         Vec3 pos = ray.at(hit.t);
         uint32_t po = sv.prim_obj[hit.prim_idx];
@@ -114,7 +114,7 @@ inline void trace_light_path(const SceneView& sv, Sampler& sp, float* img, int w
         img[pix * 4 + 0] += v.x;
         img[pix * 4 + 1] += v.y;
         img[pix * 4 + 2] += v.z;
-    }, spec_constraint, caustic_scaling);
+    }, TravCtx{}, spec_constraint, caustic_scaling);
 }
 
 } // namespace hippt

@@ -73,12 +73,23 @@ struct SceneView {
 // runtime fallback was measured to cost the megakernel ~11% (dead code +
 // its register pressure under __launch_bounds__); the binary tree stays as
 // the collapse source and for host-side self-tests (tests/test_core.py).
+// Per-thread traversal-stack context: device kernels point lds_slot into a
+// __shared__ array (entry d at lds_slot[d*256]); host passes the default
+// (pure private stack).  See bvh4.h for why the stack lives in LDS.
+struct TravCtx {
+    uint64_t* lds_slot = nullptr;
+    int lds_n = 0;
+};
+
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
-                             float tmax = MAX_DIST) {
-    return ray_intersect_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
+                             float tmax = MAX_DIST, TravCtx tc = {}) {
+    return ray_intersect_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
+                              tc.lds_slot, tc.lds_n);
 }
-HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax) {
-    return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax);
+HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
+                       TravCtx tc = {}) {
+    return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
+                               tc.lds_slot, tc.lds_n);
 }
 
 } // namespace hippt

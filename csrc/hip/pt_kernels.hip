@@ -30,10 +30,15 @@ __device__ inline int xcd_swizzle(int flat, int n_tiles) {
     return xcd * per + (xcd < rem ? xcd : rem) + local;
 }
 
-template <int RENDERER, int MINWAVES = 0>
-__global__ __launch_bounds__(256, MINWAVES)
+// lds_n = traversal-stack entries per thread held in (dynamic) LDS, bvh4.h.
+// The LDS block allocation (lds_n x 8 B x 256 threads) is also the occupancy
+// governor: 12/16/20/26/40 entries -> 6/5/4/3/2 waves per SIMD.
+template <int RENDERER>
+__global__ __launch_bounds__(256)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed, int swiz) {
+              int spp0, int nspp, uint32_t seed, int swiz, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n};
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
     if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
     const int px = (tile % gridDim.x) * 16 + threadIdx.x;
@@ -47,10 +52,10 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
         Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
         Ray ray = sv.cam.gen_ray(px, py, sp);
         Vec3 L(0.f);
-        if constexpr (RENDERER == R_VOLUME_PT) L = trace_path_volumetric(sv, ray, sp);
-        else if constexpr (RENDERER == R_DEPTH) L = Vec3(trace_depth(sv, ray));
+        if constexpr (RENDERER == R_VOLUME_PT) L = trace_path_volumetric(sv, ray, sp, tc);
+        else if constexpr (RENDERER == R_DEPTH) L = Vec3(trace_depth(sv, ray, tc));
         else if constexpr (RENDERER == R_BVH_COST) { Vec2 c = trace_bvh_cost(sv, ray); L = Vec3(c.x, c.y, 0.f); }
-        else L = trace_path(sv, ray, sp);
+        else L = trace_path(sv, ray, sp, tc);
         Lsum += L;
         float lum = (L.x + L.y + L.z) * (1.f / 3.f);
         lum_s += lum;
@@ -68,10 +73,12 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
 // device memory zeroed per launch; removes the tail effect of uneven
 // per-tile path lengths on the 8-XCD chip.
 template <int RENDERER>
-__global__ __launch_bounds__(256, 4)
+__global__ __launch_bounds__(256)
 void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
                          int spp0, int nspp, uint32_t seed, uint32_t* work_counter,
-                         int tiles_x, int n_tiles) {
+                         int tiles_x, int n_tiles, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n};
     __shared__ uint32_t s_tile;
     for (;;) {
         if (threadIdx.x == 0 && threadIdx.y == 0)
@@ -90,7 +97,7 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
         for (int s = 0; s < nspp; ++s) {
             Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
             Ray ray = sv.cam.gen_ray(px, py, sp);
-            Vec3 L = trace_path(sv, ray, sp);
+            Vec3 L = trace_path(sv, ray, sp, tc);
             Lsum += L;
             float lum = (L.x + L.y + L.z) * (1.f / 3.f);
             lum_s += lum;
@@ -106,7 +113,9 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
 __global__ __launch_bounds__(256)
 void k_render_lt(SceneView sv, float* __restrict__ accum,
                  long long n_paths, int spp0, int nspp, uint32_t seed,
-                 int spec_constraint, float caustic_scaling) {
+                 int spec_constraint, float caustic_scaling, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     const long long stride = (long long)gridDim.x * blockDim.x;
     for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n_paths; i += stride) {
         Sampler sp(uint32_t(i & 0xffffffff), uint32_t(spp0) * SEED_SCALER + seed + uint32_t(i >> 32));
@@ -114,7 +123,7 @@ void k_render_lt(SceneView sv, float* __restrict__ accum,
             atomicAdd(accum + pix * 4 + 0, v.x);
             atomicAdd(accum + pix * 4 + 1, v.y);
             atomicAdd(accum + pix * 4 + 2, v.z);
-        }, spec_constraint, caustic_scaling);
+        }, tc, spec_constraint, caustic_scaling);
     }
 }
 
@@ -134,6 +143,22 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         return e ? atoi(e) : 1;
     }();
     auto swiz = [] { return swiz_v; };
+    // HIPPT_OCC = waves/SIMD cap (2..6), enforced by the dynamic-LDS
+    // traversal-stack size: lds_n {40,26,20,16,12} -> 2..6 waves/SIMD.
+    static int lds_n = [] {
+        const char* e = getenv("HIPPT_OCC");
+        int occ = e ? atoi(e) : 4;
+        int m[7] = {20, 20, 40, 26, 20, 16, 12};
+        int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
+        if (v * 256 * 8 > 65536) {
+            (void)hipFuncSetAttribute((const void*)&k_render<R_MEGAKERNEL_PT>,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * 256 * 8);
+            (void)hipFuncSetAttribute((const void*)&k_render<R_VOLUME_PT>,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * 256 * 8);
+        }
+        return v;
+    }();
+    const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
     dim3 grid((w + 15) / 16, (h + 15) / 16);
@@ -141,21 +166,21 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     case R_LIGHT_TRACE: {
         long long n_paths = (long long)w * h * nspp;
         int nblk = 256 * 8 * 4;  // 256 CUs x enough blocks to fill + stride
-        hipLaunchKernelGGL(k_render_lt, dim3(nblk), dim3(256), 0, st,
-                           sv, accum, n_paths, spp0, nspp, seed, spec_constraint, caustic_scaling);
+        hipLaunchKernelGGL(k_render_lt, dim3(nblk), dim3(256), shmem, st,
+                           sv, accum, n_paths, spp0, nspp, seed, spec_constraint, caustic_scaling, lds_n);
         size_t npix = (size_t)w * h;
         hipLaunchKernelGGL(k_add_count, dim3((npix + 255) / 256), dim3(256), 0, st,
                            accum, npix, (float)nspp);
         break;
     }
     case R_VOLUME_PT:
-        hipLaunchKernelGGL(k_render<R_VOLUME_PT>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
+        hipLaunchKernelGGL(k_render<R_VOLUME_PT>, grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL(k_render<R_DEPTH>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed, 0);
+        hipLaunchKernelGGL(k_render<R_DEPTH>, grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, 0, st, sv, accum, var, spp0, nspp, seed, 0);
+        hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -164,30 +189,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         }
         hipMemsetAsync(counter, 0, 4, st);
         int tiles_x = (w + 15) / 16, tiles_y = (h + 15) / 16;
-        // 256 CUs x 4 blocks/CU (4 wave64 each at waves/SIMD>=4)
-        hipLaunchKernelGGL((k_render_persistent<R_MEGAKERNEL_PT>), dim3(256 * 4), block, 0, st,
+        // 256 CUs x 4 blocks/CU (lds_n=20 caps residency at 4 blocks/CU)
+        hipLaunchKernelGGL(k_render_persistent<R_MEGAKERNEL_PT>, dim3(256 * 4), block, shmem, st,
                            sv, accum, var, spp0, nspp, seed, counter, tiles_x,
-                           tiles_x * tiles_y);
+                           tiles_x * tiles_y, lds_n);
         break;
     }
     default: {
-        static int waves = [] {
-            const char* e = getenv("HIPPT_WAVES");
-            return e ? atoi(e) : 0;
-        }();
-        // measured on MI355X (kitchen 1080p, post NaN-fix): 129/154/168/180
-        // Msps at waves 3/4/5/6 — memory-latency-bound, occupancy wins even
-        // with VGPR spills.  HIPPT_WAVES overrides for A/B.
-        if (waves == 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
-        else if (waves == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
-        else if (waves == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
-        else if (waves == 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 0>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
-        else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, 0, st, sv, accum, var, spp0, nspp, seed, swiz());
+        hipLaunchKernelGGL(k_render<R_MEGAKERNEL_PT>, grid, block, shmem, st,
+                           sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     }
     }

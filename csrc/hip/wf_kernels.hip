@@ -17,6 +17,7 @@
 // read back once per bounce (4 bytes) instead of a device-wide binary search.
 #include <hip/hip_runtime.h>
 #include "kernels.h"
+#include <cstdlib>
 #include "../core/integrator.h"
 
 namespace hippt {
@@ -56,13 +57,15 @@ __device__ __forceinline__ uint32_t pack_counts(int d, int s, int t, int b) {
 
 // ----------------------------------------------------------------- raygen
 __global__ __launch_bounds__(256)
-void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed) {
+void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= st.n) return;
     int px = i % st.w, py = i / st.w;
     Sampler sp(uint32_t(i), uint32_t(spp_idx) * SEED_SCALER + seed);
     Ray ray = sv.cam.gen_ray(px, py, sp);
-    HitRecord hit = scene_intersect(sv, ray);
+    HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
     st.ray_o[i] = make_float4(ray.o.x, ray.o.y, ray.o.z, 0.f);
     st.ray_d[i] = make_float4(ray.d.x, ray.d.y, ray.d.z, 0.f);
     st.thp[i] = make_float4(1.f, 1.f, 1.f, uint_as_float(1u));  // prev_delta=1
@@ -273,13 +276,15 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
 // Lean traversal-only kernel (58 VGPR class -> 8 waves/SIMD): any-hit test,
 // then a race-free add into L (exactly one shadow ray per payload per bounce).
 __global__ __launch_bounds__(256)
-void k_wf_shadow(SceneView sv, WfState st) {
+void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     int k = blockIdx.x * blockDim.x + threadIdx.x;
     if (k >= *st.sh_cnt) return;
     float4 od = st.sh_od[k];
     float4 dir = st.sh_dir[k];
     Ray ray(Vec3(od.x, od.y, od.z), Vec3(dir.x, dir.y, dir.z));
-    if (scene_occluded(sv, ray, od.w)) return;
+    if (scene_occluded(sv, ray, od.w, tc)) return;
     int i = (int)float_as_uint(dir.w);
     float4 v = st.sh_val[k];
     float4 l4 = st.L[i];
@@ -288,7 +293,9 @@ void k_wf_shadow(SceneView sv, WfState st) {
 
 // ------------------------------------------------------- next closest hit
 __global__ __launch_bounds__(256)
-void k_wf_trace(SceneView sv, WfState st) {
+void k_wf_trace(SceneView sv, WfState st, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     // pixel-order scan: bounce rays of neighboring pixels traverse similar
     // BVH subtrees, so keeping trace in payload order (NOT material-sorted
     // order) preserves wave-level spatial coherence — measured 4.5ms -> see
@@ -298,7 +305,7 @@ void k_wf_trace(SceneView sv, WfState st) {
     if (st.status[i] >> 24 >= DEAD) return;  // dead or terminated in shade
     float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i];
     Ray ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
-    HitRecord hit = scene_intersect(sv, ray);
+    HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
     st.hit[i] = make_float4(hit.t, hit.u, hit.v, int_as_float(hit.prim_idx));
     uint32_t status;
     if (hit.prim_idx < 0) {
@@ -391,8 +398,28 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     const int n = st->n;
     dim3 blk(WF_BLOCK);
     dim3 grd_n((n + WF_BLOCK - 1) / WF_BLOCK);
+    // HIPPT_WF_OCC = waves/SIMD cap for the traversal kernels, enforced by
+    // the dynamic-LDS stack size (lds_n entries x 8 B x 256 threads/block):
+    // {2,3,4,5,6} waves -> lds_n {40,26,20,16,12}.  See bvh4.h for why the
+    // stack lives in LDS.
+    static int lds_n = [] {
+        const char* e = getenv("HIPPT_WF_OCC");
+        int occ = e ? atoi(e) : 4;
+        int m[7] = {20, 20, 40, 26, 20, 16, 12};
+        int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
+        if (v * WF_BLOCK * 8 > 65536) {
+            (void)hipFuncSetAttribute((const void*)&k_wf_raygen,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
+            (void)hipFuncSetAttribute((const void*)&k_wf_trace,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
+            (void)hipFuncSetAttribute((const void*)&k_wf_shadow,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
+        }
+        return v;
+    }();
+    const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
     for (int s = 0; s < nspp; ++s) {
-        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, 0, hs, sv, *st, spp0 + s, seed);
+        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
         const int nb = st->nb_sort;
         for (int bounce = 0; bounce < sv.md.max_depth + 1; ++bounce) {
             // build the compacted material-sorted view over the FULL status
@@ -409,8 +436,8 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
-            hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, 0, hs, sv, *st);
-            hipLaunchKernelGGL(k_wf_trace, grd_n, blk, 0, hs, sv, *st);
+            hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n);
+            hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
     }
