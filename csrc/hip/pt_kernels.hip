@@ -9,6 +9,7 @@
 #include <hip/hip_runtime.h>
 #include "kernels.h"
 #include <cstdlib>
+#include <cstring>
 #include "../core/integrator.h"
 #include "../core/integrator_vol.h"
 #include "../core/light_tracer.h"
@@ -143,11 +144,16 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         return e ? atoi(e) : 1;
     }();
     auto swiz = [] { return swiz_v; };
-    // HIPPT_OCC = waves/SIMD cap (2..6), enforced by the dynamic-LDS
-    // traversal-stack size: lds_n {40,26,20,16,12} -> 2..6 waves/SIMD.
-    static int lds_n = [] {
+    // Two decoupled knobs (defaults from same-box A/B on MI355X, see
+    // profiles/README.md):
+    //   HIPPT_OCC   = waves/SIMD cap 2..6, enforced by reserving dynamic LDS
+    //                 ({40,26,20,16,12} x 2 KB) whether or not the stack
+    //                 uses it.  Default 6 (= no effective cap).
+    //   HIPPT_STACK = lds | scratch: where the BVH4 traversal stack lives.
+    //                 Megakernel default: scratch (LDS was measured -28%).
+    static int occ_res = [] {
         const char* e = getenv("HIPPT_OCC");
-        int occ = e ? atoi(e) : 4;
+        int occ = e ? atoi(e) : 6;
         int m[7] = {20, 20, 40, 26, 20, 16, 12};
         int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
         if (v * 256 * 8 > 65536) {
@@ -158,7 +164,11 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         }
         return v;
     }();
-    const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
+    static int lds_n = [] {
+        const char* e = getenv("HIPPT_STACK");
+        return (e && strcmp(e, "lds") == 0) ? occ_res : 0;
+    }();
+    const uint32_t shmem = (uint32_t)occ_res * 256 * 8;
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
     dim3 grid((w + 15) / 16, (h + 15) / 16);

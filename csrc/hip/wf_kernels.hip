@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 #include "kernels.h"
 #include <cstdlib>
+#include <cstring>
 #include "../core/integrator.h"
 
 namespace hippt {
@@ -398,13 +399,12 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     const int n = st->n;
     dim3 blk(WF_BLOCK);
     dim3 grd_n((n + WF_BLOCK - 1) / WF_BLOCK);
-    // HIPPT_WF_OCC = waves/SIMD cap for the traversal kernels, enforced by
-    // the dynamic-LDS stack size (lds_n entries x 8 B x 256 threads/block):
-    // {2,3,4,5,6} waves -> lds_n {40,26,20,16,12}.  See bvh4.h for why the
-    // stack lives in LDS.
-    static int lds_n = [] {
+    // HIPPT_WF_OCC = waves/SIMD cap (reserved LDS) and HIPPT_WF_STACK =
+    // lds | scratch for the traversal kernels (defaults: occ 6, lds —
+    // measured best for the lean trace/shadow kernels, profiles/README.md).
+    static int occ_res = [] {
         const char* e = getenv("HIPPT_WF_OCC");
-        int occ = e ? atoi(e) : 4;
+        int occ = e ? atoi(e) : 6;
         int m[7] = {20, 20, 40, 26, 20, 16, 12};
         int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
         if (v * WF_BLOCK * 8 > 65536) {
@@ -417,7 +417,11 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         }
         return v;
     }();
-    const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
+    static int lds_n = [] {
+        const char* e = getenv("HIPPT_WF_STACK");
+        return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res;
+    }();
+    const uint32_t shmem = (uint32_t)occ_res * WF_BLOCK * 8;
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
         const int nb = st->nb_sort;
