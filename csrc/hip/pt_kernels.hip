@@ -34,8 +34,8 @@ __device__ inline int xcd_swizzle(int flat, int n_tiles) {
 // lds_n = traversal-stack entries per thread held in (dynamic) LDS, bvh4.h.
 // The LDS block allocation (lds_n x 8 B x 256 threads) is also the occupancy
 // governor: 12/16/20/26/40 entries -> 6/5/4/3/2 waves per SIMD.
-template <int RENDERER>
-__global__ __launch_bounds__(256)
+template <int RENDERER, int MINW = 6>
+__global__ __launch_bounds__(256, MINW)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
               int spp0, int nspp, uint32_t seed, int swiz, int lds_n) {
     extern __shared__ uint64_t s_stk[];
@@ -73,8 +73,8 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
 // On MI355X: grid = 256 CUs x blocks/CU, tile = 16x16 pixels, counter in
 // device memory zeroed per launch; removes the tail effect of uneven
 // per-tile path lengths on the 8-XCD chip.
-template <int RENDERER>
-__global__ __launch_bounds__(256)
+template <int RENDERER, int MINW = 6>
+__global__ __launch_bounds__(256, MINW)
 void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
                          int spp0, int nspp, uint32_t seed, uint32_t* work_counter,
                          int tiles_x, int n_tiles, int lds_n) {
@@ -111,7 +111,7 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
 }
 
 // ------------------------------------------------------------- light tracing
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(256, 6)
 void k_render_lt(SceneView sv, float* __restrict__ accum,
                  long long n_paths, int spp0, int nspp, uint32_t seed,
                  int spec_constraint, float caustic_scaling, int lds_n) {
@@ -146,29 +146,21 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     auto swiz = [] { return swiz_v; };
     // Two decoupled knobs (defaults from same-box A/B on MI355X, see
     // profiles/README.md):
-    //   HIPPT_OCC   = waves/SIMD cap 2..6, enforced by reserving dynamic LDS
-    //                 ({40,26,20,16,12} x 2 KB) whether or not the stack
-    //                 uses it.  Default 6 (= no effective cap).
-    //   HIPPT_STACK = lds | scratch: where the BVH4 traversal stack lives.
-    //                 Megakernel default: scratch (LDS was measured -28%).
-    static int occ_res = [] {
+    //   HIPPT_OCC   = __launch_bounds__ min waves/SIMD {3,4,5,6,8}.  The
+    //   register CAP is what the walk wants (waves 6 spills VGPRs yet was
+    //   measured ~1.5x over the compiler's natural allocation).  Default 6.
+    //   HIPPT_STACK = lds | scratch: where the BVH4 traversal stack lives
+    //   (LDS entries sized {26,20,16,12,12} for occ {3,4,5,6,8}).
+    static int occ_v = [] {
         const char* e = getenv("HIPPT_OCC");
-        int occ = e ? atoi(e) : 6;
-        int m[7] = {20, 20, 40, 26, 20, 16, 12};
-        int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
-        if (v * 256 * 8 > 65536) {
-            (void)hipFuncSetAttribute((const void*)&k_render<R_MEGAKERNEL_PT>,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * 256 * 8);
-            (void)hipFuncSetAttribute((const void*)&k_render<R_VOLUME_PT>,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * 256 * 8);
-        }
-        return v;
+        return e ? atoi(e) : 6;
     }();
     static int lds_n = [] {
         const char* e = getenv("HIPPT_STACK");
-        return (e && strcmp(e, "lds") == 0) ? occ_res : 0;
+        if (!(e && strcmp(e, "lds") == 0)) return 0;
+        return occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12;
     }();
-    const uint32_t shmem = (uint32_t)occ_res * 256 * 8;
+    const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
     dim3 grid((w + 15) / 16, (h + 15) / 16);
@@ -184,13 +176,13 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         break;
     }
     case R_VOLUME_PT:
-        hipLaunchKernelGGL(k_render<R_VOLUME_PT>, grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL(k_render<R_DEPTH>, grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL(k_render<R_BVH_COST>, grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -200,14 +192,22 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         hipMemsetAsync(counter, 0, 4, st);
         int tiles_x = (w + 15) / 16, tiles_y = (h + 15) / 16;
         // 256 CUs x 4 blocks/CU (lds_n=20 caps residency at 4 blocks/CU)
-        hipLaunchKernelGGL(k_render_persistent<R_MEGAKERNEL_PT>, dim3(256 * 4), block, shmem, st,
+        hipLaunchKernelGGL((k_render_persistent<R_MEGAKERNEL_PT, 6>), dim3(256 * 4), block, shmem, st,
                            sv, accum, var, spp0, nspp, seed, counter, tiles_x,
                            tiles_x * tiles_y, lds_n);
         break;
     }
     default: {
-        hipLaunchKernelGGL(k_render<R_MEGAKERNEL_PT>, grid, block, shmem, st,
-                           sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        if (occ_v <= 3)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        else if (occ_v == 4)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        else if (occ_v == 5)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        else if (occ_v >= 8)
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        else
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     }
     }

@@ -421,7 +421,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         const char* e = getenv("HIPPT_WF_STACK");
         return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res;
     }();
-    const uint32_t shmem = (uint32_t)occ_res * WF_BLOCK * 8;
+    const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
         const int nb = st->nb_sort;
