@@ -138,10 +138,12 @@ int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling, void* stream) {
     hipStream_t st = (hipStream_t)stream;
-    // HIPPT_SWIZZLE=0 disables the XCD band swizzle (A/B hook; default on)
+    // HIPPT_SWIZZLE=1 enables the XCD band swizzle (measured -14% on the
+    // kitchen megakernel — the round-robin XCD dispatch already spreads
+    // neighbor tiles well; default off, kept as an A/B hook)
     static int swiz_v = [] {
         const char* e = getenv("HIPPT_SWIZZLE");
-        return e ? atoi(e) : 1;
+        return e ? atoi(e) : 0;
     }();
     auto swiz = [] { return swiz_v; };
     // Two decoupled knobs (defaults from same-box A/B on MI355X, see
@@ -157,7 +159,7 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }();
     static int lds_n = [] {
         const char* e = getenv("HIPPT_STACK");
-        if (!(e && strcmp(e, "lds") == 0)) return 0;
+        if (e && strcmp(e, "scratch") == 0) return 0;   // default: lds
         return occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12;
     }();
     const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
