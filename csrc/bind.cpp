@@ -448,6 +448,9 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
         r.d = {ray_d.at(i, 0), ray_d.at(i, 1), ray_d.at(i, 2)};
         HitRecord a = ray_intersect_bvh(bn, n_nodes, pr, po, r, tmax);
         HitRecord b = ray_intersect_bvh4(n4, pr, po, r, tmax);
+        HitRecord w = ray_intersect_bvh4_ww(n4, pr, po, r, tmax);
+        if ((w.prim_idx < 0) != (b.prim_idx < 0) ||
+            (b.prim_idx >= 0 && fabsf(w.t - b.t) > 1e-5f * fmaxf(1.f, b.t))) { ++bad; continue; }
         bool occ_a = occlusion_test_bvh(bn, n_nodes, pr, po, r, tmax);
         bool occ_b = occlusion_test_bvh4(n4, pr, po, r, tmax);
         // prim index may differ only on exact t ties; compare t and object

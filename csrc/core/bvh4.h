@@ -143,6 +143,106 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
     }
 }
 
+// While-while variant of the closest-hit walk (Aila & Laine style phase
+// batching, re-derived for wave64): lanes alternate between a node-walk
+// phase and a leaf-test phase at the OUTER loop level, so a wave tends to
+// execute runs of same-kind work instead of interleaving a node step on
+// some lanes with prim tests on others.  Leaf children are POSTPONED onto
+// the stack (tagged entries) instead of intersected inline.  Measured A/B
+// against the inline walk via HIPPT_TRAV=ww (divergence: VALUUtilization
+// 15% on the inline walk).
+//
+// Stack/cur entry low word: bit31 = leaf, [30:23] = prim count (<=255),
+// [22:0] = prim base (scene must have < 8.4M prims; launch code falls back
+// to the inline walk otherwise).
+HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
+                                   const Prim* prims, const uint32_t* prim_obj,
+                                   const Ray& ray, float tmax,
+                                   uint64_t* lds_slot = nullptr, int lds_n = 0) {
+    HitRecord rec;
+    rec.t = tmax;
+    const Vec3 inv_d = safe_rcp_dir(ray.d);
+    const Vec3 o_div = ray.o * inv_d;
+    uint64_t stack[BVH4_STACK];
+    int sp = 0;
+    constexpr uint32_t DONE = 0x7fffffffu;
+    uint32_t cur = 0;
+    for (;;) {
+        // ---- node phase: walk internal nodes until a leaf surfaces
+        while (cur < 0x80000000u && cur != DONE) {
+            const BVH4Node nd = nodes[cur];
+            uint32_t keys[4];
+            int nhit = 0;
+#pragma unroll
+            for (int c = 0; c < 4; ++c) {
+                float t0x = fmaf(nd.lo_x[c], inv_d.x, -o_div.x);
+                float t1x = fmaf(nd.hi_x[c], inv_d.x, -o_div.x);
+                float t0y = fmaf(nd.lo_y[c], inv_d.y, -o_div.y);
+                float t1y = fmaf(nd.hi_y[c], inv_d.y, -o_div.y);
+                float t0z = fmaf(nd.lo_z[c], inv_d.z, -o_div.z);
+                float t1z = fmaf(nd.hi_z[c], inv_d.z, -o_div.z);
+                float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                    fmaxf(fminf(t0z, t1z), 0.f));
+                float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                    fminf(fmaxf(t0z, t1z), rec.t));
+                if (enter <= exit_) keys[nhit++] = (float_as_uint(enter) & ~3u) | (uint32_t)c;
+            }
+            if (nhit > 1) {
+                if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                if (nhit > 2) {
+                    if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                    if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                    if (nhit > 3) {
+                        if (keys[2] > keys[3]) { uint32_t t = keys[2]; keys[2] = keys[3]; keys[3] = t; }
+                        if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                        if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                    }
+                }
+            }
+            uint32_t next = DONE;
+            for (int k = nhit - 1; k >= 0; --k) {  // far -> near so near pops first
+                int c = (int)(keys[k] & 3u);
+                int ch = nd.child[c];
+                int pc = nd.cnt[c];
+                if (ch < 0 && pc == 0) continue;   // empty slot
+                uint32_t lo = ch < 0
+                    ? (0x80000000u | ((uint32_t)pc << 23) | (uint32_t)(~ch))
+                    : (uint32_t)ch;
+                if (k == 0) {
+                    next = lo;
+                } else {
+                    uint64_t e = ((uint64_t)(keys[k] & ~3u) << 32) | lo;
+                    if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = e;
+                    else stack[sp - lds_n] = e;
+                    ++sp;
+                }
+            }
+            if (next != DONE) { cur = next; continue; }
+            for (;;) {  // pop, culling stale subtrees
+                if (sp == 0) { cur = DONE; break; }
+                --sp;
+                uint64_t e = sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE] : stack[sp - lds_n];
+                if (uint_as_float((uint32_t)(e >> 32)) < rec.t) { cur = (uint32_t)e; break; }
+            }
+        }
+        if (cur == DONE) break;
+        // ---- leaf phase: drain consecutive leaf entries
+        while (cur >= 0x80000000u) {
+            bvh4_leaf_hit(prims, prim_obj, ray, (int)(cur & 0x007fffffu),
+                          (int)((cur >> 23) & 0xffu), rec);
+            for (;;) {
+                if (sp == 0) { cur = DONE; break; }
+                --sp;
+                uint64_t e = sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE] : stack[sp - lds_n];
+                if (uint_as_float((uint32_t)(e >> 32)) < rec.t) { cur = (uint32_t)e; break; }
+            }
+        }
+        if (cur == DONE) break;
+    }
+    if (rec.prim_idx < 0) rec.t = MAX_DIST;
+    return rec;
+}
+
 // Any-hit occlusion test: returns true if something blocks [EPSILON, tmax].
 // No ordering (any hit ends the walk) — children are pushed unordered.
 HD bool occlusion_test_bvh4(const BVH4Node* nodes,
