@@ -152,9 +152,8 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
 // against the inline walk via HIPPT_TRAV=ww (divergence: VALUUtilization
 // 15% on the inline walk).
 //
-// Stack/cur entry low word: bit31 = leaf, [30:23] = prim count (<=255),
-// [22:0] = prim base (scene must have < 8.4M prims; launch code falls back
-// to the inline walk otherwise).
+// Stack/cur entry low word: bit31 = leaf, [30:27] = prim count (builders
+// cap leaves at 15 prims), [26:0] = prim base (up to 134M prims).
 HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
                                    const Prim* prims, const uint32_t* prim_obj,
                                    const Ray& ray, float tmax,
@@ -206,7 +205,7 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
                 int pc = nd.cnt[c];
                 if (ch < 0 && pc == 0) continue;   // empty slot
                 uint32_t lo = ch < 0
-                    ? (0x80000000u | ((uint32_t)pc << 23) | (uint32_t)(~ch))
+                    ? (0x80000000u | ((uint32_t)pc << 27) | (uint32_t)(~ch))
                     : (uint32_t)ch;
                 if (k == 0) {
                     next = lo;
@@ -228,8 +227,8 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
         if (cur == DONE) break;
         // ---- leaf phase: drain consecutive leaf entries
         while (cur >= 0x80000000u) {
-            bvh4_leaf_hit(prims, prim_obj, ray, (int)(cur & 0x007fffffu),
-                          (int)((cur >> 23) & 0xffu), rec);
+            bvh4_leaf_hit(prims, prim_obj, ray, (int)(cur & 0x07ffffffu),
+                          (int)((cur >> 27) & 0xfu), rec);
             for (;;) {
                 if (sp == 0) { cur = DONE; break; }
                 --sp;

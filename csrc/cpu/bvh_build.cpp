@@ -118,8 +118,9 @@ struct Builder {
         int axis;
         int m = find_split(lo, hi, box, axis);
         if (m < 0) {
-            if (n > 255) {
-                // forced equal-count split (leaf prim_cnt must fit traversal budget)
+            if (n > 15) {
+                // forced equal-count split: leaves are capped at 15 prims so
+                // the while-while walk can pack (cnt,base) into 31 bits
                 m = lo + n / 2;
                 std::nth_element(bp.begin() + lo, bp.begin() + m, bp.begin() + hi,
                                  [&](const BuildPrim& a, const BuildPrim& b_) {
@@ -191,7 +192,7 @@ BVHBuildResult build_bvh(const Prim* prims, const uint32_t* prim_obj, int n,
     BVHBuildResult res;
     if (n <= 0) return res;
     Builder b;
-    b.max_leaf = std::max(1, cfg.max_leaf_prims);
+    b.max_leaf = std::max(1, std::min(cfg.max_leaf_prims, 15));
     b.overlap_w = cfg.overlap_w;
     b.bp.resize(n);
     for (int i = 0; i < n; ++i) {

@@ -216,7 +216,7 @@ void build_rec(const Ctx& ctx, std::vector<Refr>&& refs, int depth, BuildOut& ou
 
     bool use_spatial = best_sp_axis >= 0 && best_sp_cost < best_obj_cost;
     if (!use_spatial && best_obj_axis < 0) {
-        if (n <= std::max(ctx.max_leaf, 8) * 16) { make_leaf(); return; }
+        if (n <= 15) { make_leaf(); return; }  // leaf cap: ww walk packs cnt in 4 bits
         // degenerate: median split
         std::vector<Refr> l(refs.begin(), refs.begin() + n / 2);
         std::vector<Refr> r(refs.begin() + n / 2, refs.end());
@@ -343,7 +343,7 @@ BVHBuildResult build_sbvh(const Prim* prims, const uint32_t* prim_obj, int n,
         refs[i] = {box, i};
         root_box.grow(box);
     }
-    Ctx ctx{prims, prim_obj, std::max(1, cfg.max_leaf_prims), cfg.ref_unsplit,
+    Ctx ctx{prims, prim_obj, std::max(1, std::min(cfg.max_leaf_prims, 15)), cfg.ref_unsplit,
             root_box.area() > 0.f ? 1.f / root_box.area() : 0.f};
     BuildOut out;
     out.nodes.reserve(2 * n);

@@ -37,9 +37,9 @@ __device__ inline int xcd_swizzle(int flat, int n_tiles) {
 template <int RENDERER, int MINW = 6>
 __global__ __launch_bounds__(256, MINW)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int ww) {
+              int spp0, int nspp, uint32_t seed, int swiz, int lds_n) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n, ww};
+    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n};
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
     if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
     const int px = (tile % gridDim.x) * 16 + threadIdx.x;
@@ -162,12 +162,6 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         if (e && strcmp(e, "scratch") == 0) return 0;   // default: lds
         return occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12;
     }();
-    // HIPPT_TRAV=ww selects the phase-batched walk (A/B; needs base<2^23)
-    static int ww_env = [] {
-        const char* e = getenv("HIPPT_TRAV");
-        return (e && strcmp(e, "ww") == 0) ? 1 : 0;
-    }();
-    const int ww_v = (ww_env && sv.n_prims < (1 << 23)) ? 1 : 0;
     const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
@@ -184,13 +178,13 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         break;
     }
     case R_VOLUME_PT:
-        hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, ww_v);
+        hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, ww_v);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, ww_v);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -207,15 +201,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     }
     }

@@ -58,9 +58,9 @@ __device__ __forceinline__ uint32_t pack_counts(int d, int s, int t, int b) {
 
 // ----------------------------------------------------------------- raygen
 __global__ __launch_bounds__(256)
-void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n, int ww) {
+void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n, ww};
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= st.n) return;
     int px = i % st.w, py = i / st.w;
@@ -294,9 +294,9 @@ void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
 
 // ------------------------------------------------------- next closest hit
 __global__ __launch_bounds__(256)
-void k_wf_trace(SceneView sv, WfState st, int lds_n, int ww) {
+void k_wf_trace(SceneView sv, WfState st, int lds_n) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n, ww};
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     // pixel-order scan: bounce rays of neighboring pixels traverse similar
     // BVH subtrees, so keeping trace in payload order (NOT material-sorted
     // order) preserves wave-level spatial coherence — measured 4.5ms -> see
@@ -422,13 +422,8 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res;
     }();
     const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
-    static int ww_env = [] {
-        const char* e = getenv("HIPPT_TRAV");
-        return (e && strcmp(e, "ww") == 0) ? 1 : 0;
-    }();
-    const int ww_v = (ww_env && sv.n_prims < (1 << 23)) ? 1 : 0;
     for (int s = 0; s < nspp; ++s) {
-        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n, ww_v);
+        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
         const int nb = st->nb_sort;
         for (int bounce = 0; bounce < sv.md.max_depth + 1; ++bounce) {
             // build the compacted material-sorted view over the FULL status
@@ -446,7 +441,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
             hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n);
-            hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n, ww_v);
+            hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
     }
