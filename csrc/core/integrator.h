@@ -29,18 +29,49 @@ HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
     return i;
 }
 
-// Full path trace for one camera ray. Returns radiance estimate.
-HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
-    Vec3 L(0.f), thp(1.f);
-    float prev_pdf = 0.f;       // BSDF pdf of the previous bounce (for MIS)
-    bool prev_delta = true;     // camera rays count as delta
-    Vec3 prev_n(0.f, 0.f, 1.f); // previous shading normal (envmap NEE pdf)
-    float path_time = 0.f;      // ToF accumulated distance
+// Resumable per-path state: one `path_step` = one bounce of the
+// unidirectional path loop.  Shared by trace_path (per-sample form) and the
+// megakernel's path-regeneration loop (k_render starts the next sample the
+// moment a lane's path dies instead of idling until the wave's longest path
+// finishes — wave64 tail divergence was the measured bottleneck,
+// VALUUtilization ~15-20%, profiles/README.md).
+struct PathState {
+    Ray ray;
+    Vec3 L, thp;
+    float prev_pdf;
+    bool prev_delta;
+    Vec3 prev_n;
+    float path_time;
     PathStats st;
-    const float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
+    int b, iter;
 
-    int b = 0;
-    for (int iter = 0; iter < sv.md.max_depth * 2 + 8 && b < sv.md.max_depth; ++iter) {
+    HD void reset(const Ray& r) {
+        ray = r;
+        L = Vec3(0.f); thp = Vec3(1.f);
+        prev_pdf = 0.f;
+        prev_delta = true;
+        prev_n = Vec3(0.f, 0.f, 1.f);
+        path_time = 0.f;
+        st = PathStats();
+        b = 0; iter = 0;
+    }
+};
+
+// One bounce; returns true when the path is finished (L is final).
+HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
+    if (ps.iter >= sv.md.max_depth * 2 + 8 || ps.b >= sv.md.max_depth) return true;
+    ++ps.iter;
+    Ray& ray = ps.ray;
+    Vec3& L = ps.L;
+    Vec3& thp = ps.thp;
+    float& prev_pdf = ps.prev_pdf;
+    bool& prev_delta = ps.prev_delta;
+    Vec3& prev_n = ps.prev_n;
+    float& path_time = ps.path_time;
+    PathStats& st = ps.st;
+    int& b = ps.b;
+    const float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
+    {
         HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
         if (hit.prim_idx < 0) {
             // miss -> environment map with MIS against the cosine NEE pdf
@@ -54,7 +85,7 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
                 }
                 if (tof_in_range(sv.md, path_time + ENVMAP_DIST)) L += thp * le * w;
             }
-            break;
+            return true;
         }
         Vec3 pos = ray.at(hit.t);
         path_time += hit.t;
@@ -107,15 +138,15 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
 
         // ---- BSDF sampling
         BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
-        if (bs.pdf <= 0.f || bs.weight.is_zero()) break;
-        if (bs.weight.has_nan() || bs.wi.has_nan()) break;  // numeric scrub
+        if (bs.pdf <= 0.f || bs.weight.is_zero()) return true;
+        if (bs.weight.has_nan() || bs.wi.has_nan()) return true;  // numeric scrub
         thp *= bs.weight;
 
         // per-lobe bounce caps (reference max_depth.h semantics)
         if (!(bs.lobe & LOBE_NULL)) {
-            if (bs.lobe & LOBE_DIFFUSE)  { if (++st.n_diffuse  > sv.md.max_diffuse)  break; }
-            if (bs.lobe & LOBE_SPECULAR) { if (++st.n_specular > sv.md.max_specular) break; }
-            if (bs.lobe & LOBE_TRANSMIT) { if (++st.n_transmit > sv.md.max_transmit) break; }
+            if (bs.lobe & LOBE_DIFFUSE)  { if (++st.n_diffuse  > sv.md.max_diffuse)  return true; }
+            if (bs.lobe & LOBE_SPECULAR) { if (++st.n_specular > sv.md.max_specular) return true; }
+            if (bs.lobe & LOBE_TRANSMIT) { if (++st.n_transmit > sv.md.max_transmit) return true; }
             ++b;
         }
         prev_delta = (bs.lobe & LOBE_DELTA) != 0;
@@ -127,13 +158,21 @@ HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
         if (b > 1) {
             float p = clampv(thp.max_elem(), 0.f, 1.f);
             if (p < 0.1f) {
-                if (sp.next1f() >= p * 10.f) break;
+                if (sp.next1f() >= p * 10.f) return true;
                 thp *= (1.f / (p * 10.f));
             }
         }
     }
-    if (L.has_nan()) return Vec3(0.f);
-    return L;
+    return false;
+}
+
+// Full path trace for one camera ray. Returns radiance estimate.
+HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
+    PathState ps;
+    ps.reset(ray);
+    while (!path_step(sv, ps, sp, tc)) {}
+    if (ps.L.has_nan()) return Vec3(0.f);
+    return ps.L;
 }
 
 // Depth renderer: distance of the primary hit (reference pt_impl/depth.cu).

@@ -49,6 +49,28 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
 
     Vec3 Lsum(0.f);
     float lum_s = 0.f, lum_s2 = 0.f;
+    if constexpr (RENDERER == R_MEGAKERNEL_PT) {
+        // Path regeneration: the lane starts its next sample the moment its
+        // path dies, instead of idling until the wave's longest path ends
+        // (wave64 tail divergence was the measured bottleneck).  Sampler
+        // streams and results are identical to the per-sample loop.
+        int s = 0;
+        Sampler sp(uint32_t(pix), uint32_t(spp0) * SEED_SCALER + seed);
+        PathState ps;
+        ps.reset(sv.cam.gen_ray(px, py, sp));
+        for (;;) {
+            if (path_step(sv, ps, sp, tc)) {
+                Vec3 L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
+                Lsum += L;
+                float lum = (L.x + L.y + L.z) * (1.f / 3.f);
+                lum_s += lum;
+                lum_s2 = fmaf(lum, lum, lum_s2);
+                if (++s >= nspp) break;
+                sp = Sampler(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
+                ps.reset(sv.cam.gen_ray(px, py, sp));
+            }
+        }
+    } else
     for (int s = 0; s < nspp; ++s) {
         Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
         Ray ray = sv.cam.gen_ray(px, py, sp);
@@ -95,14 +117,21 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
         const size_t pix = size_t(py) * sv.cam.w + px;
         Vec3 Lsum(0.f);
         float lum_s = 0.f, lum_s2 = 0.f;
-        for (int s = 0; s < nspp; ++s) {
-            Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
-            Ray ray = sv.cam.gen_ray(px, py, sp);
-            Vec3 L = trace_path(sv, ray, sp, tc);
-            Lsum += L;
-            float lum = (L.x + L.y + L.z) * (1.f / 3.f);
-            lum_s += lum;
-            lum_s2 = fmaf(lum, lum, lum_s2);
+        int s = 0;
+        Sampler sp(uint32_t(pix), uint32_t(spp0) * SEED_SCALER + seed);
+        PathState ps;
+        ps.reset(sv.cam.gen_ray(px, py, sp));
+        for (;;) {
+            if (path_step(sv, ps, sp, tc)) {
+                Vec3 L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
+                Lsum += L;
+                float lum = (L.x + L.y + L.z) * (1.f / 3.f);
+                lum_s += lum;
+                lum_s2 = fmaf(lum, lum, lum_s2);
+                if (++s >= nspp) break;
+                sp = Sampler(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
+                ps.reset(sv.cam.gen_ray(px, py, sp));
+            }
         }
         float* a = accum + pix * 4;
         a[0] += Lsum.x; a[1] += Lsum.y; a[2] += Lsum.z; a[3] += (float)nspp;
@@ -155,7 +184,7 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     //   (LDS entries sized {26,20,16,12,12} for occ {3,4,5,6,8}).
     static int occ_v = [] {
         const char* e = getenv("HIPPT_OCC");
-        return e ? atoi(e) : 6;
+        return e ? atoi(e) : 4;   // ww walk: occ4 measured 140.6 vs 128.9 @6
     }();
     static int lds_n = [] {
         const char* e = getenv("HIPPT_STACK");
