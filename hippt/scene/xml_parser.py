@@ -67,14 +67,18 @@ def _props(elem) -> Dict[str, object]:
             out[name] = int(float(ch.get("value")))
         elif tag == "float":
             out[name] = float(ch.get("value"))
-        elif tag == "bool":
+        elif tag in ("bool", "boolean"):
             out[name] = ch.get("value", "false").lower() in ("true", "1")
         elif tag == "string":
             out[name] = ch.get("value")
         elif tag == "rgb":
             out[name] = parse_rgb(ch.get("value"))
         elif tag == "point":
-            out[name] = (float(ch.get("x", 0)), float(ch.get("y", 0)), float(ch.get("z", 0)))
+            # both attribute styles: x/y/z or value="x, y, z"
+            if ch.get("value") is not None:
+                out[name] = parse_rgb(ch.get("value"))
+            else:
+                out[name] = (float(ch.get("x", 0)), float(ch.get("y", 0)), float(ch.get("z", 0)))
         elif tag == "ref":
             refs[ch.get("type", "material")] = ch.get("id")
         elif tag == "transform":

@@ -125,10 +125,31 @@ BALLS_XML = """<?xml version='1.0' encoding='utf-8'?>
 """
 
 
+def sky_texture():
+    """Procedural lat-long sky for env-balls.xml (no HDR assets ship)."""
+    from hippt.utils.png import write_png
+    h, w = 128, 256
+    y = np.linspace(1, -1, h)[:, None]
+    x = np.linspace(0, 2 * np.pi, w)[None, :]
+    sky = np.zeros((h, w, 3), np.float32)
+    sky[..., 2] = 0.55 + 0.4 * np.clip(y, 0, 1)                       # blue up
+    sky[..., 1] = 0.45 + 0.25 * np.clip(y, 0, 1) + 0.2 * np.clip(-y, 0, 1)
+    sky[..., 0] = 0.4 + 0.5 * np.clip(-y, 0, 1)                       # warm down
+    # sun disk
+    sun = np.exp(-(((x - 1.1) ** 2) / 0.01 + ((y - 0.45) ** 2) / 0.004))
+    sky += 8.0 * sun[..., None] * np.array([1.0, 0.95, 0.85], np.float32)
+    img = (np.clip(sky / (1 + sky), 0, 1) * 255).astype(np.uint8)
+    tdir = os.path.join(ROOT, "scenes", "textures")
+    os.makedirs(tdir, exist_ok=True)
+    write_png(os.path.join(tdir, "sky.png"), img)
+
+
 def main():
     cbox_meshes()
     save_obj(f"{MESH}/floor_big.obj", quad((-6, 0, -6), (-6, 0, 6), (6, 0, 6), (6, 0, -6)))
     save_obj(f"{MESH}/lamp.obj", quad((1.5, 4, -1.5), (1.5, 4, 1.5), (-1.5, 4, 1.5), (-1.5, 4, -1.5)))
+    save_obj(f"{MESH}/smoke_bound.obj", box_mesh((-0.6, 0.05, 0.4), (0.6, 1.6, 1.6)))
+    sky_texture()
     with open(os.path.join(ROOT, "scenes", "cornell-box.xml"), "w") as f:
         f.write(CORNELL_XML)
     with open(os.path.join(ROOT, "scenes", "balls.xml"), "w") as f:
