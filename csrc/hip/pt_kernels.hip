@@ -35,7 +35,11 @@ __device__ inline int xcd_swizzle(int flat, int n_tiles) {
 // The LDS block allocation (lds_n x 8 B x 256 threads) is also the occupancy
 // governor: 12/16/20/26/40 entries -> 6/5/4/3/2 waves per SIMD.
 template <int RENDERER, int MINW = 6>
-__global__ __launch_bounds__(256, MINW)
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
+// stack caps blocks/CU at MINW waves/SIMD, so the allocator may spend the
+// full 512/MINW VGPRs instead of hoarding occupancy it can't get (measured:
+// plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
               int spp0, int nspp, uint32_t seed, int swiz, int lds_n) {
     extern __shared__ uint64_t s_stk[];
@@ -96,7 +100,8 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
 // device memory zeroed per launch; removes the tail effect of uneven
 // per-tile path lengths on the 8-XCD chip.
 template <int RENDERER, int MINW = 6>
-__global__ __launch_bounds__(256, MINW)
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))
 void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
                          int spp0, int nspp, uint32_t seed, uint32_t* work_counter,
                          int tiles_x, int n_tiles, int lds_n) {
@@ -135,7 +140,8 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
 }
 
 // ------------------------------------------------------------- light tracing
-__global__ __launch_bounds__(256, 6)
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(6, 6)))
 void k_render_lt(SceneView sv, float* __restrict__ accum,
                  long long n_paths, int spp0, int nspp, uint32_t seed,
                  int spec_constraint, float caustic_scaling, int lds_n) {

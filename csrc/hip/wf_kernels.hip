@@ -58,6 +58,7 @@ __device__ __forceinline__ uint32_t pack_counts(int d, int s, int t, int b) {
 
 // ----------------------------------------------------------------- raygen
 __global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
 void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc{&s_stk[threadIdx.x], lds_n};
@@ -277,6 +278,7 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
 // Lean traversal-only kernel (58 VGPR class -> 8 waves/SIMD): any-hit test,
 // then a race-free add into L (exactly one shadow ray per payload per bounce).
 __global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
 void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc{&s_stk[threadIdx.x], lds_n};
@@ -294,6 +296,7 @@ void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
 
 // ------------------------------------------------------- next closest hit
 __global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
 void k_wf_trace(SceneView sv, WfState st, int lds_n) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc{&s_stk[threadIdx.x], lds_n};
