@@ -30,6 +30,7 @@ def main():
     ap.add_argument("--reduce-interval", type=int, default=1,
                     help="all-reduce the accumulator every this many steps")
     ap.add_argument("--cpu", action="store_true", help="CPU reference path (debug)")
+    ap.add_argument("--sbvh", action="store_true", help="spatial-split BVH for the scene")
     args = ap.parse_args()
 
     import torch
@@ -61,6 +62,8 @@ def main():
         renderer_kind = "vpt"
     else:
         raise SystemExit(f"unknown scene {args.scene}")
+    if args.sbvh or os.environ.get("HIPPT_SBVH"):
+        desc.config.use_sbvh = True
 
     device = None if args.cpu else local_rank
     if not args.cpu:

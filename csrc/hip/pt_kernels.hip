@@ -208,7 +208,10 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         break;
     }
     case R_VOLUME_PT:
-        hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        if (occ_v == 4)
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+        else
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     case R_DEPTH:
         hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
