@@ -43,7 +43,7 @@ Vec3 to_vec3(const std::vector<float>& v) { return {v[0], v[1], v[2]}; }
 
 struct SceneHolder {
     // ----- host data (kept alive / owned)
-    farr np_prims, np_attrs, np_nodes, np_nodes4;
+    farr np_prims, np_attrs, np_nodes, np_nodes4, np_nodes8;
     uarr np_prim_obj;
     iarr np_objs;
     std::vector<BsdfParams> bsdfs;
@@ -84,7 +84,8 @@ struct SceneHolder {
         has_dev = false;
     }
 
-    void set_geometry(farr prims, farr attrs, uarr prim_obj, farr nodes, farr nodes4) {
+    void set_geometry(farr prims, farr attrs, uarr prim_obj, farr nodes, farr nodes4,
+                      farr nodes8) {
         if (prims.ndim() != 2 || prims.shape(1) != 12) throw std::runtime_error("prims must be (n,12)");
         if (attrs.ndim() != 2 || attrs.shape(1) != 16) throw std::runtime_error("attrs must be (n,16)");
         if (nodes.ndim() != 2 || nodes.shape(1) != 8) throw std::runtime_error("nodes must be (m,8)");
@@ -95,6 +96,9 @@ struct SceneHolder {
         np_prim_obj = std::move(prim_obj);
         np_nodes = std::move(nodes);
         np_nodes4 = std::move(nodes4);
+        if (nodes8.ndim() == 2 && nodes8.shape(0) > 0 && nodes8.shape(1) != 64)
+            throw std::runtime_error("nodes8 must be (m,64)");
+        np_nodes8 = std::move(nodes8);
     }
 
     void set_objects(iarr objs) {
@@ -253,6 +257,7 @@ struct SceneHolder {
     void fill_common(SceneView& sv) {
         sv.n_nodes = (int)np_nodes.shape(0) - 1;  // last row = sentinel
         sv.n_nodes4 = np_nodes4.ndim() == 2 ? (int)np_nodes4.shape(0) : 0;
+        sv.n_nodes8 = np_nodes8.ndim() == 2 ? (int)np_nodes8.shape(0) : 0;
         sv.n_prims = (int)np_prims.shape(0);
         sv.n_objs = np_objs.ndim() == 2 ? (int)np_objs.shape(0) : 0;
         sv.n_bsdfs = (int)bsdfs.size();
@@ -269,6 +274,7 @@ struct SceneHolder {
         fill_common(host_sv);
         host_sv.nodes = (const BVHNode*)np_nodes.data();
         host_sv.nodes4 = host_sv.n_nodes4 > 0 ? (const BVH4Node*)np_nodes4.data() : nullptr;
+        host_sv.nodes8 = host_sv.n_nodes8 > 0 ? (const BVH8Node*)np_nodes8.data() : nullptr;
         host_sv.prims = (const Prim*)np_prims.data();
         host_sv.attrs = (const PrimAttr*)np_attrs.data();
         host_sv.prim_obj = np_prim_obj.data();
@@ -301,6 +307,8 @@ struct SceneHolder {
         dev_sv.nodes = upload_vec((const BVHNode*)np_nodes.data(), np_nodes.shape(0));
         dev_sv.nodes4 = dev_sv.n_nodes4 > 0
             ? upload_vec((const BVH4Node*)np_nodes4.data(), np_nodes4.shape(0)) : nullptr;
+        dev_sv.nodes8 = dev_sv.n_nodes8 > 0
+            ? upload_vec((const BVH8Node*)np_nodes8.data(), np_nodes8.shape(0)) : nullptr;
         dev_sv.prims = upload_vec((const Prim*)np_prims.data(), np_prims.shape(0));
         dev_sv.attrs = upload_vec((const PrimAttr*)np_attrs.data(), np_attrs.shape(0));
         dev_sv.prim_obj = upload_vec(np_prim_obj.data(), np_prim_obj.size());
@@ -412,6 +420,23 @@ py::tuple py_build_bvh(farr prims, uarr prim_obj, int max_leaf, float overlap_w,
     return py::make_tuple(nodes, order, stats);
 }
 
+py::tuple py_collapse_bvh8(farr nodes) {
+    if (nodes.ndim() != 2 || nodes.shape(1) != 8) throw std::runtime_error("nodes must be (m,8)");
+    std::vector<BVHNode> bin((size_t)nodes.shape(0));
+    std::memcpy(bin.data(), nodes.data(), bin.size() * sizeof(BVHNode));
+    int depth8 = 0;
+    std::vector<BVH8Node> n8;
+    {
+        py::gil_scoped_release rel;
+        n8 = collapse_bvh8(bin, &depth8);
+    }
+    if (7 * depth8 > BVH4_STACK)
+        throw std::runtime_error("BVH8 depth exceeds traversal stack bound");
+    farr out({(py::ssize_t)n8.size(), (py::ssize_t)64});
+    std::memcpy(out.mutable_data(), n8.data(), n8.size() * sizeof(BVH8Node));
+    return py::make_tuple(out, depth8);
+}
+
 // Collapse a (m,8) skip-link node array into a (k,32) BVH4 array (bvh4.h).
 py::tuple py_collapse_bvh4(farr nodes) {
     if (nodes.ndim() != 2 || nodes.shape(1) != 8) throw std::runtime_error("nodes must be (m,8)");
@@ -434,7 +459,9 @@ py::tuple py_collapse_bvh4(farr nodes) {
 // (closest hit t/prim and occlusion verdict) on caller-supplied rays.
 // Returns the number of mismatching rays.
 int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
-                     farr ray_o, farr ray_d, float tmax) {
+                     farr ray_o, farr ray_d, float tmax, farr nodes8) {
+    const BVH8Node* n8p = (nodes8.ndim() == 2 && nodes8.shape(0) > 0)
+        ? (const BVH8Node*)nodes8.data() : nullptr;
     int n_nodes = (int)nodes.shape(0);
     const BVHNode* bn = (const BVHNode*)nodes.data();
     const BVH4Node* n4 = (const BVH4Node*)nodes4.data();
@@ -451,6 +478,14 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
         HitRecord w = ray_intersect_bvh4_ww(n4, pr, po, r, tmax);
         if ((w.prim_idx < 0) != (b.prim_idx < 0) ||
             (b.prim_idx >= 0 && fabsf(w.t - b.t) > 1e-5f * fmaxf(1.f, b.t))) { ++bad; continue; }
+        if (n8p) {
+            HitRecord w8 = ray_intersect_bvh8_ww(n8p, pr, po, r, tmax);
+            bool occ8 = occlusion_test_bvh8(n8p, pr, po, r, tmax);
+            bool occ_ref = occlusion_test_bvh4(n4, pr, po, r, tmax);
+            if ((w8.prim_idx < 0) != (b.prim_idx < 0) ||
+                (b.prim_idx >= 0 && fabsf(w8.t - b.t) > 1e-5f * fmaxf(1.f, b.t)) ||
+                occ8 != occ_ref) { ++bad; continue; }
+        }
         bool occ_a = occlusion_test_bvh(bn, n_nodes, pr, po, r, tmax);
         bool occ_b = occlusion_test_bvh4(n4, pr, po, r, tmax);
         // prim index may differ only on exact t ties; compare t and object
@@ -494,7 +529,11 @@ PYBIND11_MODULE(_C, m) {
           py::arg("max_leaf") = 4, py::arg("overlap_w") = 0.f,
           py::arg("use_sbvh") = false, py::arg("ref_unsplit") = true);
     m.def("collapse_bvh4", &py_collapse_bvh4, py::arg("nodes"));
-    m.def("bvh4_selftest", &py_bvh4_selftest);
+    m.def("collapse_bvh8", &py_collapse_bvh8, py::arg("nodes"));
+    m.def("bvh4_selftest", &py_bvh4_selftest,
+          py::arg("prims"), py::arg("prim_obj"), py::arg("nodes"), py::arg("nodes4"),
+          py::arg("ray_o"), py::arg("ray_d"), py::arg("tmax"),
+          py::arg("nodes8") = farr());
 
     m.def("dev_synchronize", [] { HIP_OK(dev_synchronize()); });
     m.def("dev_set_device", [](int d) { HIP_OK(dev_set_device(d)); });

@@ -9,6 +9,7 @@
 #pragma once
 #include "bvh.h"
 #include "bvh4.h"
+#include "bvh8.h"
 #include "bsdf.h"
 #include "emitter.h"
 #include "camera.h"
@@ -44,6 +45,7 @@ struct SceneView {
     // the collapse source; traversal runs on the 4-wide tree when present)
     const BVHNode* nodes; int n_nodes;
     const BVH4Node* nodes4; int n_nodes4;
+    const BVH8Node* nodes8; int n_nodes8;   // A/B: used when non-null
     const Prim* prims; const PrimAttr* attrs; const uint32_t* prim_obj; int n_prims;
     const ObjInfo* objs; int n_objs;
     // materials / emitters / textures
@@ -86,11 +88,17 @@ struct TravCtx {
 // the inline walk stays for host-side self-tests.
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST, TravCtx tc = {}) {
+    if (sv.nodes8)
+        return ray_intersect_bvh8_ww(sv.nodes8, sv.prims, sv.prim_obj, ray, tmax,
+                                     tc.lds_slot, tc.lds_n);
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                  tc.lds_slot, tc.lds_n);
 }
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                        TravCtx tc = {}) {
+    if (sv.nodes8)
+        return occlusion_test_bvh8(sv.nodes8, sv.prims, sv.prim_obj, ray, tmax,
+                                   tc.lds_slot, tc.lds_n);
     return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                tc.lds_slot, tc.lds_n);
 }

@@ -312,4 +312,80 @@ std::vector<BVH4Node> collapse_bvh4(const std::vector<BVHNode>& bin, int* max_de
     return col.out;
 }
 
+namespace {
+
+// 8-wide collapse (same expansion rule, width 8; bvh8.h traversal).
+struct Collapser8 {
+    const std::vector<BVHNode>& bin;
+    std::vector<BVH8Node> out;
+    int max_depth = 0;
+
+    float area(int i) const {
+        Vec3 e = bin[i].hi.xyz() - bin[i].lo.xyz();
+        return 2.f * (e.x * e.y + e.y * e.z + e.z * e.x);
+    }
+
+    int emit(int bi, int depth) {
+        max_depth = std::max(max_depth, depth);
+        int my = (int)out.size();
+        out.emplace_back();
+        int slots[8];
+        int n_slots = 0;
+        if (bin[bi].is_leaf()) {
+            slots[n_slots++] = bi;
+        } else {
+            int l = bi + 1, r = right_child(bin, l);
+            slots[n_slots++] = l;
+            slots[n_slots++] = r;
+            while (n_slots < 8) {
+                int pick = -1;
+                float best = -1.f;
+                for (int s = 0; s < n_slots; ++s)
+                    if (!bin[slots[s]].is_leaf() && area(slots[s]) > best) {
+                        best = area(slots[s]);
+                        pick = s;
+                    }
+                if (pick < 0) break;
+                int p = slots[pick];
+                int pl = p + 1, pr = right_child(bin, pl);
+                slots[pick] = pl;
+                slots[n_slots++] = pr;
+            }
+        }
+        BVH8Node tmp{};
+        for (int c = 0; c < 8; ++c) {
+            if (c < n_slots) {
+                int s = slots[c];
+                tmp.lo_x[c] = bin[s].lo.x; tmp.lo_y[c] = bin[s].lo.y; tmp.lo_z[c] = bin[s].lo.z;
+                tmp.hi_x[c] = bin[s].hi.x; tmp.hi_y[c] = bin[s].hi.y; tmp.hi_z[c] = bin[s].hi.z;
+                if (bin[s].is_leaf()) {
+                    tmp.child[c] = ~bin[s].prim_base();
+                    tmp.cnt[c] = bin[s].prim_cnt();
+                } else {
+                    tmp.child[c] = emit(s, depth + 1);
+                    tmp.cnt[c] = 0;
+                }
+            } else {
+                tmp.lo_x[c] = tmp.lo_y[c] = tmp.lo_z[c] = 3.0e38f;
+                tmp.hi_x[c] = tmp.hi_y[c] = tmp.hi_z[c] = 3.0e38f;
+                tmp.child[c] = ~0;
+                tmp.cnt[c] = 0;
+            }
+        }
+        out[my] = tmp;
+        return my;
+    }
+};
+
+} // namespace
+
+std::vector<BVH8Node> collapse_bvh8(const std::vector<BVHNode>& bin, int* max_depth8) {
+    Collapser8 col{bin};
+    if (bin.empty()) return {};
+    col.out.reserve(bin.size() / 4 + 1);
+    col.emit(0, 1);
+    if (max_depth8) *max_depth8 = col.max_depth;
+    return col.out;
+}
+
 } // namespace hippt

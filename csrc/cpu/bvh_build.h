@@ -9,6 +9,7 @@
 #include <cstdint>
 #include "../core/bvh.h"
 #include "../core/bvh4.h"
+#include "../core/bvh8.h"
 
 namespace hippt {
 
@@ -39,6 +40,10 @@ BVHBuildResult build_bvh(const Prim* prims, const uint32_t* prim_obj, int n,
 // collapsed tree depth (traversal stack bound = 3 * depth).
 std::vector<BVH4Node> collapse_bvh4(const std::vector<BVHNode>& bin,
                                     int* max_depth4 = nullptr);
+
+// 8-wide collapse (256-byte nodes, bvh8.h).
+std::vector<BVH8Node> collapse_bvh8(const std::vector<BVHNode>& bin,
+                                    int* max_depth8 = nullptr);
 
 // SBVH: spatial-split BVH (Stich et al. style chopped binning); may duplicate
 // references, so prim_order can be longer than n.

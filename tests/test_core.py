@@ -190,3 +190,16 @@ class TestBVH4:
         prims, pobj, nodes, nodes4, depth4 = self._build(2, seed=18)
         o, d = self._rays(500, seed=19)
         assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7) == 0
+
+    def test_bvh8_agreement(self):
+        prims, pobj, nodes, nodes4, _ = self._build(800, seed=21)
+        nodes8, depth8 = C.collapse_bvh8(nodes)
+        assert 7 * depth8 <= 64
+        o, d = self._rays(4000, seed=22)
+        assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7, nodes8) == 0
+
+    def test_bvh8_agreement_spheres(self):
+        prims, pobj, nodes, nodes4, _ = self._build(400, seed=23, spheres=True)
+        nodes8, depth8 = C.collapse_bvh8(nodes)
+        o, d = self._rays(4000, seed=24)
+        assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7, nodes8) == 0
