@@ -86,19 +86,17 @@ struct TravCtx {
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
 // / +19% wavefront over the inline-leaf ordered walk, profiles/README.md);
 // the inline walk stays for host-side self-tests.
+// BVH8 (bvh8.h) was measured 2.4x SLOWER than BVH4 here (256-byte node
+// copy = 64 VGPRs of temporaries + SAH dilution at width 8), so the 8-wide
+// walk is NOT wired into the kernels — even a dead `if (sv.nodes8)` branch
+// cost ~3% of kernel throughput.  bvh8.h stays host-tested for the record.
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST, TravCtx tc = {}) {
-    if (sv.nodes8)
-        return ray_intersect_bvh8_ww(sv.nodes8, sv.prims, sv.prim_obj, ray, tmax,
-                                     tc.lds_slot, tc.lds_n);
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                  tc.lds_slot, tc.lds_n);
 }
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                        TravCtx tc = {}) {
-    if (sv.nodes8)
-        return occlusion_test_bvh8(sv.nodes8, sv.prims, sv.prim_obj, ray, tmax,
-                                   tc.lds_slot, tc.lds_n);
     return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                tc.lds_slot, tc.lds_n);
 }
