@@ -305,7 +305,10 @@ class Scene:
         objs = np.stack(obj_rows, axis=0) if obj_rows else np.zeros((0, 8), np.int32)
 
         # ---- BVH (native builder, multithreaded)
-        nodes, order, stats = C.build_bvh(prims, prim_obj, cfg.max_leaf, cfg.overlap_w,
+        import os as _os
+        max_leaf = int(_os.environ.get("HIPPT_MAX_LEAF", cfg.max_leaf))
+        overlap_w = float(_os.environ.get("HIPPT_OVERLAP_W", cfg.overlap_w))
+        nodes, order, stats = C.build_bvh(prims, prim_obj, max_leaf, overlap_w,
                                           cfg.use_sbvh, cfg.ref_unsplit)
         self.bvh_stats = stats
         # 4-wide collapse of the binary tree: the traversal that actually runs
