@@ -152,7 +152,7 @@ class RenderConfig:
     use_tof: bool = False
     renderer: str = "pt"               # pt | wfpt | vpt | lt | depth | bvh-cost | pt-dyn
     max_leaf: int = 4
-    overlap_w: float = 0.0
+    overlap_w: float = 0.6   # SAH overlap penalty (reference accelerator default)
     use_sbvh: bool = False
     ref_unsplit: bool = True
     cache_level: int = 6
