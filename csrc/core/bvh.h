@@ -89,7 +89,7 @@ HD HitRecord ray_intersect_bvh(const BVHNode* nodes, int n_nodes,
                 }
             }
         }
-        i = hit_box | (cnt > 0) ? i + 1 : -cnt;
+        i = (hit_box | (cnt > 0)) ? i + 1 : -cnt;
     }
     if (rec.prim_idx < 0) rec.t = MAX_DIST;
     return rec;
@@ -117,7 +117,7 @@ HD bool occlusion_test_bvh(const BVHNode* nodes, int n_nodes,
                 if (t > EPSILON && t < tmax) return true;
             }
         }
-        i = hit_box | (cnt > 0) ? i + 1 : -cnt;
+        i = (hit_box | (cnt > 0)) ? i + 1 : -cnt;
     }
     return false;
 }

@@ -224,7 +224,7 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         if (!counter) {
             if (hipMalloc((void**)&counter, 4) != hipSuccess) return (int)hipGetLastError();
         }
-        hipMemsetAsync(counter, 0, 4, st);
+        (void)hipMemsetAsync(counter, 0, 4, st);
         int tiles_x = (w + 15) / 16, tiles_y = (h + 15) / 16;
         // 256 CUs x 4 blocks/CU (lds_n=20 caps residency at 4 blocks/CU)
         hipLaunchKernelGGL((k_render_persistent<R_MEGAKERNEL_PT, 6>), dim3(256 * 4), block, shmem, st,

@@ -387,12 +387,12 @@ WfState* wf_create(int width, int height) {
 
 void wf_destroy(WfState* s) {
     if (!s) return;
-    hipFree(s->ray_o); hipFree(s->ray_d); hipFree(s->thp); hipFree(s->L);
-    hipFree(s->hit); hipFree(s->prevn); hipFree(s->rng);
-    hipFree(s->status); hipFree(s->order); hipFree(s->hist);
-    hipFree(s->live_dev);
-    hipFree(s->sh_od); hipFree(s->sh_dir); hipFree(s->sh_val); hipFree(s->sh_cnt);
-    if (s->live_host) hipHostFree(s->live_host);
+    (void)hipFree(s->ray_o); (void)hipFree(s->ray_d); (void)hipFree(s->thp); (void)hipFree(s->L);
+    (void)hipFree(s->hit); (void)hipFree(s->prevn); (void)hipFree(s->rng);
+    (void)hipFree(s->status); (void)hipFree(s->order); (void)hipFree(s->hist);
+    (void)hipFree(s->live_dev);
+    (void)hipFree(s->sh_od); (void)hipFree(s->sh_dir); (void)hipFree(s->sh_val); (void)hipFree(s->sh_cnt);
+    if (s->live_host) (void)hipHostFree(s->live_host);
     delete s;
 }
 
@@ -433,14 +433,14 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             // array (pixel order preserved there for the trace kernel);
             // the live count never leaves the device: shade/shadow/trace are
             // launched full-width and exit by comparing against *live_dev.
-            hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
+            (void)hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
             hipLaunchKernelGGL(k_sort_hist, dim3(nb), dim3(SORT_BLOCK), 0, hs,
                                st->status, n, st->hist);
             hipLaunchKernelGGL(k_sort_scan, dim3(1), dim3(256), 0, hs,
                                st->hist, st->live_dev);
             hipLaunchKernelGGL(k_sort_scatter, dim3(nb), dim3(SORT_BLOCK), 0, hs,
                                st->status, n, st->hist, st->order);
-            hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
+            (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
             hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n);
