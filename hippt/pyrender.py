@@ -84,6 +84,38 @@ class PythonRenderer:
     def save(self, path: str, gamma: float = 2.1):
         self.renderer.save(path, gamma)
 
+    def save_state(self, path: str):
+        """Checkpoint the warm accumulation state (radiance sums, variance
+        sums, sample counter) so a long accumulation can resume after a
+        restart.  The reference keeps this state implicit and unpersisted
+        (tracer_base.cuh:135-158 accum buffer + accum_cnt); here it is a
+        first-class .npz snapshot."""
+        self._check()
+        import numpy as np
+        r = self.renderer
+        accum = r.accum.cpu().numpy() if r.device is not None else r.accum
+        var = r.var.cpu().numpy() if r.device is not None else r.var
+        np.savez_compressed(path, accum=accum, var=var,
+                            accum_cnt=np.int64(r.accum_cnt))
+
+    def load_state(self, path: str):
+        """Resume from a save_state() snapshot (shape-checked)."""
+        self._check()
+        import numpy as np
+        z = np.load(path)
+        r = self.renderer
+        if tuple(z["accum"].shape) != tuple(r.accum.shape):
+            raise ValueError(f"checkpoint shape {z['accum'].shape} != "
+                             f"framebuffer {tuple(r.accum.shape)}")
+        if r.device is not None:
+            import torch
+            r.accum.copy_(torch.from_numpy(z["accum"]).to(r.accum.device))
+            r.var.copy_(torch.from_numpy(z["var"]).to(r.var.device))
+        else:
+            r.accum[:] = z["accum"]
+            r.var[:] = z["var"]
+        r.accum_cnt = int(z["accum_cnt"])
+
     def _check(self):
         if self._released:
             raise RuntimeError("renderer already released")

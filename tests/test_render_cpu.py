@@ -119,3 +119,33 @@ class TestDebugRenderers:
         img = render_desc(d, spp=2)
         assert img[..., 0].mean() > 1.0   # node visits
         assert img[..., 1].mean() > 0.1   # prim tests
+
+
+def test_checkpoint_resume(tmp_path):
+    """save_state/load_state round-trips the warm accumulator: N spp, save,
+    fresh renderer, load, +M spp == continuous N+M spp render."""
+    from hippt.scene.procedural import cornell_box
+    d = cornell_box(width=32, height=32, spp=1, max_depth=3)
+    r1 = hippt.PythonRenderer(d, device_id=-1)
+    r1.render(spp=4)
+    ckpt = str(tmp_path / "state.npz")
+    r1.save_state(ckpt)
+    d2 = cornell_box(width=32, height=32, spp=1, max_depth=3)
+    r2 = hippt.PythonRenderer(d2, device_id=-1)
+    r2.load_state(ckpt)
+    assert r2.counter() == r1.counter()
+    b = r2.render(spp=4).numpy()
+    d3 = cornell_box(width=32, height=32, spp=1, max_depth=3)
+    r3 = hippt.PythonRenderer(d3, device_id=-1)
+    r3.render(spp=4)
+    c = r3.render(spp=4).numpy()
+    np.testing.assert_allclose(b, c, rtol=1e-5, atol=1e-6)
+
+    # shape mismatch rejected
+    d4 = cornell_box(width=16, height=16, spp=1, max_depth=3)
+    r4 = hippt.PythonRenderer(d4, device_id=-1)
+    try:
+        r4.load_state(ckpt)
+        assert False, "expected ValueError"
+    except ValueError:
+        pass
