@@ -23,7 +23,7 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--width", type=int, default=1920)
     ap.add_argument("--height", type=int, default=1080)
-    ap.add_argument("--spp-per-step", type=int, default=16)
+    ap.add_argument("--spp-per-step", type=int, default=32)
     ap.add_argument("--renderer", type=str, default=None,
                     help="pt | wfpt | pt-dyn (default: fastest available)")
     ap.add_argument("--scene", type=str, default="kitchen")
