@@ -407,7 +407,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     // measured best for the lean trace/shadow kernels, profiles/README.md).
     static int occ_res = [] {
         const char* e = getenv("HIPPT_WF_OCC");
-        int occ = e ? atoi(e) : 6;
+        int occ = e ? atoi(e) : 5;   // ww walk: occ4/5 ~100 vs occ6 98.5 Msps
         int m[7] = {20, 20, 40, 26, 20, 16, 12};
         int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
         if (v * WF_BLOCK * 8 > 65536) {
