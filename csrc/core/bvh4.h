@@ -294,6 +294,79 @@ HD bool occlusion_test_bvh4(const BVH4Node* nodes,
     }
 }
 
+// Phase-batched any-hit occlusion walk: same while-while shaping as the
+// closest-hit walk (leaf children postponed onto the stack so a wave runs
+// node steps and prim tests in separate phases), but unordered and with an
+// immediate `true` return on the first blocking hit.  A/B hook
+// HIPPT_OCC_WW selects this against the inline-leaf occlusion walk.
+HD bool occlusion_test_bvh4_ww(const BVH4Node* nodes,
+                               const Prim* prims, const uint32_t* prim_obj,
+                               const Ray& ray, float tmax,
+                               uint64_t* lds_slot = nullptr, int lds_n = 0) {
+    const Vec3 inv_d = safe_rcp_dir(ray.d);
+    const Vec3 o_div = ray.o * inv_d;
+    uint64_t stack[BVH4_STACK];
+    int sp = 0;
+    constexpr uint32_t DONE = 0x7fffffffu;
+    uint32_t cur = 0;
+    for (;;) {
+        while (cur < 0x80000000u && cur != DONE) {
+            const BVH4Node nd = nodes[cur];
+            uint32_t next = DONE;
+#pragma unroll
+            for (int c = 0; c < 4; ++c) {
+                float t0x = fmaf(nd.lo_x[c], inv_d.x, -o_div.x);
+                float t1x = fmaf(nd.hi_x[c], inv_d.x, -o_div.x);
+                float t0y = fmaf(nd.lo_y[c], inv_d.y, -o_div.y);
+                float t1y = fmaf(nd.hi_y[c], inv_d.y, -o_div.y);
+                float t0z = fmaf(nd.lo_z[c], inv_d.z, -o_div.z);
+                float t1z = fmaf(nd.hi_z[c], inv_d.z, -o_div.z);
+                float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                    fmaxf(fminf(t0z, t1z), 0.f));
+                float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                    fminf(fmaxf(t0z, t1z), tmax));
+                if (enter > exit_) continue;
+                int ch = nd.child[c];
+                int pc = nd.cnt[c];
+                if (ch < 0 && pc == 0) continue;
+                uint32_t lo = ch < 0
+                    ? (0x80000000u | ((uint32_t)pc << 27) | (uint32_t)(~ch))
+                    : (uint32_t)ch;
+                if (next == DONE) {
+                    next = lo;
+                } else {
+                    uint64_t e = (uint64_t)lo;
+                    if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = e;
+                    else stack[sp - lds_n] = e;
+                    ++sp;
+                }
+            }
+            if (next != DONE) { cur = next; continue; }
+            if (sp == 0) { cur = DONE; break; }
+            --sp;
+            cur = (uint32_t)(sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE]
+                                        : stack[sp - lds_n]);
+        }
+        if (cur == DONE) return false;
+        while (cur >= 0x80000000u) {
+            int base = (int)(cur & 0x07ffffffu);
+            int pc = (int)((cur >> 27) & 0xfu);
+            for (int k = 0; k < pc; ++k) {
+                int pid = base + k;
+                bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+                float u, v;
+                float t = intersect_prim(prims[pid], sph, ray, u, v);
+                if (t > EPSILON && t < tmax) return true;
+            }
+            if (sp == 0) { cur = DONE; break; }
+            --sp;
+            cur = (uint32_t)(sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE]
+                                        : stack[sp - lds_n]);
+        }
+        if (cur == DONE) return false;
+    }
+}
+
 // Node-visit / prim-test counting walk for the BVH-cost visualizer
 // (reference pt_impl/bvh_cost.cu:38-101; counts reflect the traversal that
 // actually runs, i.e. the 4-wide one).

@@ -81,6 +81,7 @@ struct SceneView {
 struct TravCtx {
     uint64_t* lds_slot = nullptr;
     int lds_n = 0;
+    int occ_ww = 0;  // A/B: phase-batched any-hit walk (HIPPT_OCC_WW)
 };
 
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
@@ -97,6 +98,9 @@ HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
 }
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                        TravCtx tc = {}) {
+    if (tc.occ_ww)
+        return occlusion_test_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
+                                      tc.lds_slot, tc.lds_n);
     return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                tc.lds_slot, tc.lds_n);
 }
