@@ -488,6 +488,7 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
         }
         bool occ_a = occlusion_test_bvh(bn, n_nodes, pr, po, r, tmax);
         bool occ_b = occlusion_test_bvh4(n4, pr, po, r, tmax);
+        if (occlusion_test_bvh4_ww(n4, pr, po, r, tmax) != occ_b) { ++bad; continue; }
         // prim index may differ only on exact t ties; compare t and object
         bool hit_match = (a.prim_idx < 0) == (b.prim_idx < 0) &&
                          (a.prim_idx < 0 || fabsf(a.t - b.t) <= 1e-5f * fmaxf(1.f, a.t));

@@ -81,7 +81,6 @@ struct SceneView {
 struct TravCtx {
     uint64_t* lds_slot = nullptr;
     int lds_n = 0;
-    int occ_ww = 0;  // A/B: phase-batched any-hit walk (HIPPT_OCC_WW)
 };
 
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
@@ -96,13 +95,12 @@ HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                  tc.lds_slot, tc.lds_n);
 }
+// Any-hit also runs the phase-batched form (+1-2% measured over the
+// inline-leaf walk; the inline walk stays for host self-tests).
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                        TravCtx tc = {}) {
-    if (tc.occ_ww)
-        return occlusion_test_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                                      tc.lds_slot, tc.lds_n);
-    return occlusion_test_bvh4(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                               tc.lds_slot, tc.lds_n);
+    return occlusion_test_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
+                                  tc.lds_slot, tc.lds_n);
 }
 
 } // namespace hippt

@@ -41,9 +41,9 @@ __attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
 // full 512/MINW VGPRs instead of hoarding occupancy it can't get (measured:
 // plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int occ_ww) {
+              int spp0, int nspp, uint32_t seed, int swiz, int lds_n) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n, occ_ww};
+    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n};
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
     if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
     const int px = (tile % gridDim.x) * 16 + threadIdx.x;
@@ -193,10 +193,6 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         return occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12;
     }();
     const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
-    static int occ_ww_v = [] {
-        const char* e = getenv("HIPPT_OCC_WW");
-        return e ? atoi(e) : 0;
-    }();
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
     dim3 grid((w + 15) / 16, (h + 15) / 16);
@@ -213,15 +209,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     case R_VOLUME_PT:
         if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, occ_ww_v);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, occ_ww_v);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -238,15 +234,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, occ_ww_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
         break;
     }
     }

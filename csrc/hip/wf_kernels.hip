@@ -279,9 +279,9 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
 // then a race-free add into L (exactly one shadow ray per payload per bounce).
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
-void k_wf_shadow(SceneView sv, WfState st, int lds_n, int occ_ww) {
+void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n, occ_ww};
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
     int k = blockIdx.x * blockDim.x + threadIdx.x;
     if (k >= *st.sh_cnt) return;
     float4 od = st.sh_od[k];
@@ -425,10 +425,6 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res;
     }();
     const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
-    static int occ_ww_v = [] {
-        const char* e = getenv("HIPPT_OCC_WW");
-        return e ? atoi(e) : 0;
-    }();
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
         const int nb = st->nb_sort;
@@ -447,7 +443,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
-            hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n, occ_ww_v);
+            hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n);
             hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
