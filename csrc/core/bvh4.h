@@ -157,7 +157,9 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
 HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
                                    const Prim* prims, const uint32_t* prim_obj,
                                    const Ray& ray, float tmax,
-                                   uint64_t* lds_slot = nullptr, int lds_n = 0) {
+                                   uint64_t* lds_slot = nullptr, int lds_n = 0,
+                                   const BVH4Node* top_cache = nullptr,
+                                   int n_cached = 0) {
     HitRecord rec;
     rec.t = tmax;
     const Vec3 inv_d = safe_rcp_dir(ray.d);
@@ -169,7 +171,11 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
     for (;;) {
         // ---- node phase: walk internal nodes until a leaf surfaces
         while (cur < 0x80000000u && cur != DONE) {
-            const BVH4Node nd = nodes[cur];
+            // top-of-tree nodes come from the LDS copy (every walk starts at
+            // the root; 95% L2 hit rate means the bound is hit LATENCY, and
+            // LDS is ~4x closer than L2).  Pointer select, single flat load.
+            const BVH4Node* nsrc = (int)cur < n_cached ? top_cache : nodes;
+            const BVH4Node nd = nsrc[cur];
             uint32_t keys[4];
             int nhit = 0;
 #pragma unroll
@@ -302,7 +308,9 @@ HD bool occlusion_test_bvh4(const BVH4Node* nodes,
 HD bool occlusion_test_bvh4_ww(const BVH4Node* nodes,
                                const Prim* prims, const uint32_t* prim_obj,
                                const Ray& ray, float tmax,
-                               uint64_t* lds_slot = nullptr, int lds_n = 0) {
+                               uint64_t* lds_slot = nullptr, int lds_n = 0,
+                               const BVH4Node* top_cache = nullptr,
+                               int n_cached = 0) {
     const Vec3 inv_d = safe_rcp_dir(ray.d);
     const Vec3 o_div = ray.o * inv_d;
     uint64_t stack[BVH4_STACK];
@@ -311,7 +319,8 @@ HD bool occlusion_test_bvh4_ww(const BVH4Node* nodes,
     uint32_t cur = 0;
     for (;;) {
         while (cur < 0x80000000u && cur != DONE) {
-            const BVH4Node nd = nodes[cur];
+            const BVH4Node* nsrc = (int)cur < n_cached ? top_cache : nodes;
+            const BVH4Node nd = nsrc[cur];
             uint32_t next = DONE;
 #pragma unroll
             for (int c = 0; c < 4; ++c) {

@@ -81,6 +81,8 @@ struct SceneView {
 struct TravCtx {
     uint64_t* lds_slot = nullptr;
     int lds_n = 0;
+    const BVH4Node* top_cache = nullptr;  // LDS copy of nodes4[0..n_cached)
+    int n_cached = 0;
 };
 
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
@@ -93,14 +95,14 @@ struct TravCtx {
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST, TravCtx tc = {}) {
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                                 tc.lds_slot, tc.lds_n);
+                                 tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
 }
 // Any-hit also runs the phase-batched form (+1-2% measured over the
 // inline-leaf walk; the inline walk stays for host self-tests).
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                        TravCtx tc = {}) {
     return occlusion_test_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                                  tc.lds_slot, tc.lds_n);
+                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
 }
 
 } // namespace hippt

@@ -41,9 +41,19 @@ __attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
 // full 512/MINW VGPRs instead of hoarding occupancy it can't get (measured:
 // plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed, int swiz, int lds_n) {
+              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.y * 16 + threadIdx.x], lds_n};
+    // dynamic-LDS layout: [n_cached 128-byte nodes][per-thread stacks].
+    // The top of the tree is copied into LDS once per block: every walk's
+    // first visits are nodes 0..n_cached, and at a 95% L2 hit rate the
+    // bound is hit latency — LDS is ~4x closer.
+    BVH4Node* s_cache = (BVH4Node*)s_stk;
+    uint64_t* s_base = s_stk + (size_t)n_cached * 16;
+    const int tid = threadIdx.y * 16 + threadIdx.x;
+    for (int i = tid; i < n_cached * 16; i += 256)
+        ((uint64_t*)s_cache)[i] = ((const uint64_t*)sv.nodes4)[i];
+    if (n_cached > 0) __syncthreads();
+    TravCtx tc{&s_base[tid], lds_n, s_cache, n_cached};
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
     if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
     const int px = (tile % gridDim.x) * 16 + threadIdx.x;
@@ -187,12 +197,22 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         const char* e = getenv("HIPPT_OCC");
         return e ? atoi(e) : 4;   // ww walk: occ4 measured 140.6 vs 128.9 @6
     }();
-    static int lds_n = [] {
+    // HIPPT_TOPCACHE = nodes of the tree top copied to LDS per block
+    // (default 64 = 8 KB); the stack's LDS share shrinks to keep the same
+    // per-block LDS budget (occupancy unchanged).
+    static int cache_req = [] {
+        const char* e = getenv("HIPPT_TOPCACHE");
+        return e ? atoi(e) : 64;
+    }();
+    const int n_cached = cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4;
+    static int lds_budget = [] {
         const char* e = getenv("HIPPT_STACK");
         if (e && strcmp(e, "scratch") == 0) return 0;   // default: lds
-        return occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12;
+        return (occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12) * 256 * 8;
     }();
-    const uint32_t shmem = (uint32_t)lds_n * 256 * 8;
+    const int lds_n = lds_budget > 0
+        ? (lds_budget - n_cached * (int)sizeof(BVH4Node)) / (256 * 8) : 0;
+    const uint32_t shmem = (uint32_t)(lds_n * 256 * 8 + n_cached * (int)sizeof(BVH4Node));
     const int w = sv.cam.w, h = sv.cam.h;
     dim3 block(16, 16);
     dim3 grid((w + 15) / 16, (h + 15) / 16);
@@ -209,15 +229,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     case R_VOLUME_PT:
         if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         else
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -234,15 +254,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
         break;
     }
     }
