@@ -21,7 +21,8 @@ using namespace hippt;
 
 namespace hippt {
 void render_cpu(const SceneView& sv, float* accum, float* var,
-                int spp0, int nspp, uint32_t seed, int renderer, int n_threads);
+                int spp0, int nspp, uint32_t seed, int renderer, int n_threads,
+                int y0, int y1);
 void render_lt_cpu(const SceneView& sv, float* accum, int spp0, int nspp, uint32_t seed,
                    int spec_constraint, float caustic_scaling, int n_threads);
 }
@@ -347,7 +348,8 @@ struct SceneHolder {
     }
 
     void render_host(farr accum, py::object var, int spp0, int nspp, uint32_t seed,
-                     int renderer, int spec_constraint, float caustic_scaling, int n_threads) {
+                     int renderer, int spec_constraint, float caustic_scaling, int n_threads,
+                     int y0 = 0, int y1 = 0) {
         finalize();
         float* vp = nullptr;
         farr var_arr;
@@ -357,12 +359,13 @@ struct SceneHolder {
             render_lt_cpu(host_sv, accum.mutable_data(), spp0, nspp, seed,
                           spec_constraint, caustic_scaling, n_threads);
         else
-            render_cpu(host_sv, accum.mutable_data(), vp, spp0, nspp, seed, renderer, n_threads);
+            render_cpu(host_sv, accum.mutable_data(), vp, spp0, nspp, seed, renderer, n_threads,
+                       y0, y1);
     }
 
     void render_device(uintptr_t accum_ptr, uintptr_t var_ptr, int spp0, int nspp,
                        uint32_t seed, int renderer, int spec_constraint,
-                       float caustic_scaling, uintptr_t stream) {
+                       float caustic_scaling, uintptr_t stream, int y0 = 0, int y1 = 0) {
         if (!has_dev) throw std::runtime_error("scene not uploaded to device");
         dev_sv.cam = cam;   // camera / depth params may have changed (hot reload)
         dev_sv.md = md;
@@ -378,7 +381,7 @@ struct SceneHolder {
             return;
         }
         HIP_OK(launch_render(dev_sv, (float*)accum_ptr, (float*)var_ptr, spp0, nspp, seed,
-                             renderer, spec_constraint, caustic_scaling, (void*)stream));
+                             renderer, spec_constraint, caustic_scaling, (void*)stream, y0, y1));
     }
 
     py::dict info() {
@@ -522,8 +525,16 @@ PYBIND11_MODULE(_C, m) {
         .def("finalize", &SceneHolder::finalize)
         .def("upload", &SceneHolder::upload)
         .def("release", &SceneHolder::release)
-        .def("render_host", &SceneHolder::render_host)
-        .def("render_device", &SceneHolder::render_device)
+        .def("render_host", &SceneHolder::render_host,
+             py::arg("accum"), py::arg("var"), py::arg("spp0"), py::arg("nspp"),
+             py::arg("seed"), py::arg("renderer"), py::arg("spec_constraint"),
+             py::arg("caustic_scaling"), py::arg("n_threads"),
+             py::arg("y0") = 0, py::arg("y1") = 0)
+        .def("render_device", &SceneHolder::render_device,
+             py::arg("accum_ptr"), py::arg("var_ptr"), py::arg("spp0"), py::arg("nspp"),
+             py::arg("seed"), py::arg("renderer"), py::arg("spec_constraint"),
+             py::arg("caustic_scaling"), py::arg("stream"),
+             py::arg("y0") = 0, py::arg("y1") = 0)
         .def("info", &SceneHolder::info);
 
     m.def("build_bvh", &py_build_bvh, py::arg("prims"), py::arg("prim_obj"),

@@ -32,9 +32,13 @@ static void parallel_rows(int h, int n_threads, const std::function<void(int)>& 
 // and var (h,w,2 float32: lum sum, lum^2 sum).  renderer: 0=PT, 2=VPT,
 // 4=depth, 5=bvh_cost (light tracing has its own entry below).
 void render_cpu(const SceneView& sv, float* accum, float* var,
-                int spp0, int nspp, uint32_t seed, int renderer, int n_threads) {
+                int spp0, int nspp, uint32_t seed, int renderer, int n_threads,
+                int y0, int y1) {
     const int w = sv.cam.w, h = sv.cam.h;
-    parallel_rows(h, n_threads, [&](int y) {
+    if (y1 <= 0 || y1 > h) y1 = h;
+    if (y0 < 0) y0 = 0;
+    parallel_rows(y1 - y0, n_threads, [&](int yr) {
+        int y = yr + y0;
         for (int x = 0; x < w; ++x) {
             size_t pix = size_t(y) * w + x;
             Vec3 Lsum(0.f);

@@ -18,10 +18,12 @@ enum RendererKind : int {
 // Accumulate nspp samples into accum (h*w*4: RGB sum + count) and var
 // (h*w*2: lum sum, lum^2 sum). sv must hold DEVICE pointers.
 // Returns hipError_t as int.
+// y0/y1: optional row band [y0, y1) for tile-split DP (0,0 = full frame);
+// applies to the megakernel renderers (wavefront/lt always render full).
 int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling,
-                  void* stream);
+                  void* stream, int y0 = 0, int y1 = 0);
 
 // Wavefront path tracer (SoA queues + compaction); state owned by WfState.
 struct WfState;

@@ -41,7 +41,7 @@ __attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
 // full 512/MINW VGPRs instead of hoarding occupancy it can't get (measured:
 // plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached) {
+              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached, int y_off, int y_end) {
     extern __shared__ uint64_t s_stk[];
     // dynamic-LDS layout: [n_cached 128-byte nodes][per-thread stacks].
     // The top of the tree is copied into LDS once per block: every walk's
@@ -57,8 +57,8 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
     if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
     const int px = (tile % gridDim.x) * 16 + threadIdx.x;
-    const int py = (tile / gridDim.x) * 16 + threadIdx.y;
-    if (px >= sv.cam.w || py >= sv.cam.h) return;
+    const int py = (tile / gridDim.x) * 16 + threadIdx.y + y_off;
+    if (px >= sv.cam.w || py >= y_end) return;   // y_end = band end (<= h)
     const size_t pix = size_t(py) * sv.cam.w + px;
 
     Vec3 Lsum(0.f);
@@ -176,7 +176,8 @@ __global__ void k_add_count(float* __restrict__ accum, size_t npix, float cnt) {
 // ------------------------------------------------------------------ launches
 int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
-                  int spec_constraint, float caustic_scaling, void* stream) {
+                  int spec_constraint, float caustic_scaling, void* stream,
+                  int y0, int y1) {
     hipStream_t st = (hipStream_t)stream;
     // HIPPT_SWIZZLE=1 enables the XCD band swizzle (measured -14% on the
     // kitchen megakernel — the round-robin XCD dispatch already spreads
@@ -214,8 +215,10 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         ? (lds_budget - n_cached * (int)sizeof(BVH4Node)) / (256 * 8) : 0;
     const uint32_t shmem = (uint32_t)(lds_n * 256 * 8 + n_cached * (int)sizeof(BVH4Node));
     const int w = sv.cam.w, h = sv.cam.h;
+    if (y1 <= 0 || y1 > h) y1 = h;
+    if (y0 < 0) y0 = 0;
     dim3 block(16, 16);
-    dim3 grid((w + 15) / 16, (h + 15) / 16);
+    dim3 grid((w + 15) / 16, (y1 - y0 + 15) / 16);
     switch (renderer) {
     case R_LIGHT_TRACE: {
         long long n_paths = (long long)w * h * nspp;
@@ -229,15 +232,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     case R_VOLUME_PT:
         if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         else
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -254,15 +257,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
         break;
     }
     }

@@ -35,6 +35,12 @@ def build_argparser():
     ap.add_argument("--spp", type=int, default=1024, help="total spp across ranks")
     ap.add_argument("--spp-per-call", type=int, default=4)
     ap.add_argument("--reduce-interval", type=int, default=128)
+    ap.add_argument("--parallelism", type=str, default="sample",
+                    choices=["sample", "tile"],
+                    help="sample = reference-style sample-split (each rank "
+                         "renders the full frame at spp/N); tile = row-band "
+                         "split (each rank renders h/N rows at full spp; "
+                         "megakernel renderers only)")
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--output", type=str, default="ddp_render.png")
     ap.add_argument("--logdir", type=str, default=None, help="TensorBoard logdir")
@@ -119,12 +125,22 @@ def main(argv=None):
         except Exception:
             writer = None
 
-    spp_per_rank = max(1, args.spp // max(world_size, 1))
+    if args.parallelism == "tile":
+        # row-band split on 16-row tile boundaries; SUM all-reduce still
+        # merges exactly (other ranks' rows are zero)
+        h = desc.camera.height
+        tiles = (h + 15) // 16
+        t0_, t1_ = tiles * rank // world_size, tiles * (rank + 1) // world_size
+        band = (t0_ * 16, min(h, t1_ * 16))
+        spp_per_rank = args.spp
+    else:
+        band = (0, 0)
+        spp_per_rank = max(1, args.spp // max(world_size, 1))
     steps = (spp_per_rank + args.spp_per_call - 1) // args.spp_per_call
     t_start = time.perf_counter()
     merged = None
     for k in range(steps):
-        rend.render(args.spp_per_call)
+        rend.render(args.spp_per_call, y0=band[0], y1=band[1])
         if world_size > 1 and ((k + 1) % args.reduce_interval == 0 or k == steps - 1):
             merged, total_spp = reduce_rendered_image(dist, rend, world_size, cpu=args.cpu)
             # frame-time all_gather (ddp_render.py:192-211)

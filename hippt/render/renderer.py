@@ -81,8 +81,10 @@ class Renderer:
     def _seed(self) -> int:
         return (self.seed_offset * 4201) & 0xFFFFFFFF
 
-    def render(self, spp: int = 1):
-        """Accumulate spp more samples (reference render_raw semantics)."""
+    def render(self, spp: int = 1, y0: int = 0, y1: int = 0):
+        """Accumulate spp more samples (reference render_raw semantics).
+        y0/y1 restrict rendering to the row band [y0, y1) — tile-split DP
+        (megakernel renderers only; 0,0 = full frame)."""
         t0 = time.perf_counter()
         if self.device is not None:
             stream = self.torch.cuda.current_stream().cuda_stream
@@ -98,7 +100,8 @@ class Renderer:
             else:
                 self.scene.native.render_device(
                     self.accum.data_ptr(), self.var.data_ptr(), self.accum_cnt, spp,
-                    self._seed(), self.rid, self.spec_constraint, self.caustic_scaling, stream)
+                    self._seed(), self.rid, self.spec_constraint, self.caustic_scaling, stream,
+                    y0, y1)
             self.torch.cuda.synchronize(self.device)
         else:
             var = self.var.reshape(-1)
@@ -114,7 +117,7 @@ class Renderer:
                 self.scene.native.render_host(self.accum.reshape(-1), self.var.reshape(-1),
                                               self.accum_cnt, spp, self._seed(),
                                               self.rid, self.spec_constraint,
-                                              self.caustic_scaling, 0)
+                                              self.caustic_scaling, 0, y0, y1)
         self.accum_cnt += spp
         self.timer.add((time.perf_counter() - t0) * 1000.0)
         return self
