@@ -270,6 +270,7 @@ def sports_car(width=3840, height=2160, spp=64, renderer="pt", seed=7) -> SceneD
     objs.append(ObjectDesc(tris=np.concatenate([
         quad((-8, 0, -8), (8, 0, -8), (8, 0, 8), (-8, 0, 8)),
         quad((-8, 0, 8), (8, 0, 8), (8, 6, 8), (-8, 6, 8)),
+        quad((-8, 0, -8), (-8, 0, 8), (-8, 6, 8), (-8, 6, -8)),
     ]), bsdf=0))
 
     # car body: displaced superellipsoid shell
@@ -283,14 +284,16 @@ def sports_car(width=3840, height=2160, spp=64, renderer="pt", seed=7) -> SceneD
 
     a, b, c = 2.3, 0.62, 1.0   # length, height, width
     e1, e2 = 0.5, 0.85
-    X = a * sgn_pow(np.cos(U), e1)
-    Y = b * sgn_pow(np.sin(np.clip(U * 0.9 + 0.1, -1.4, 1.4)), e1) * \
-        (1.0 + 0.16 * np.cos(V))
-    Z = c * sgn_pow(np.cos(U), e1) * sgn_pow(np.sin(V), e2) * 0.5
-    # cabin bulge
-    bulge = 0.35 * np.exp(-((X + 0.3) ** 2) / 0.8) * np.clip(np.sin(V), 0, 1)
-    Y = np.abs(Y) * 0.5 + bulge + 0.35
-    P = np.stack([X, Y, Z * 2.0], axis=-1).astype(np.float32)
+    # superellipsoid: length along sin(U) (U covers the FULL -a..a span),
+    # cross-section radius along cos(U)
+    X = a * sgn_pow(np.sin(U), e1)
+    cr = sgn_pow(np.cos(U), e1)
+    Yr = b * cr * sgn_pow(np.cos(V), e2)
+    Z = c * cr * sgn_pow(np.sin(V), e2)
+    # cabin bulge on the upper shell, amidships
+    bulge = 0.5 * np.exp(-((X + 0.3) ** 2) / 0.8) * np.clip(np.cos(V), 0, 1) * cr
+    Y = np.abs(Yr) * 0.55 + bulge + 0.30
+    P = np.stack([X, Y, Z], axis=-1).astype(np.float32)
     A = P[:-1, :-1]; B = P[1:, :-1]; Cc = P[1:, 1:]; Dd = P[:-1, 1:]
     body = np.concatenate([
         np.stack([A, B, Cc], axis=2).reshape(-1, 3, 3),
