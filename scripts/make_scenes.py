@@ -149,6 +149,7 @@ def main():
     save_obj(f"{MESH}/floor_big.obj", quad((-6, 0, -6), (-6, 0, 6), (6, 0, 6), (6, 0, -6)))
     save_obj(f"{MESH}/lamp.obj", quad((1.5, 4, -1.5), (1.5, 4, 1.5), (-1.5, 4, 1.5), (-1.5, 4, -1.5)))
     save_obj(f"{MESH}/smoke_bound.obj", box_mesh((-0.6, 0.05, 0.4), (0.6, 1.6, 1.6)))
+    save_obj(f"{MESH}/backdrop.obj", quad((-5, 0, 2.5), (5, 0, 2.5), (5, 5, 2.5), (-5, 5, 2.5)))
     sky_texture()
     with open(os.path.join(ROOT, "scenes", "cornell-box.xml"), "w") as f:
         f.write(CORNELL_XML)
