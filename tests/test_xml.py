@@ -69,7 +69,7 @@ class TestExampleScenes:
 
     def test_tof_cbox_time_gate(self):
         d, img = self._render("tof-cbox.xml", spp=8)
-        assert d.config.use_tof and d.config.min_time == 2.4
+        assert d.config.use_tof and d.config.min_time == 2.8
         assert d.cam_medium == 0
         # gate kills all light outside [2.4, 3.2]: image much darker than
         # the ungated render
