@@ -22,7 +22,7 @@ using namespace hippt;
 namespace hippt {
 void render_cpu(const SceneView& sv, float* accum, float* var,
                 int spp0, int nspp, uint32_t seed, int renderer, int n_threads,
-                int y0, int y1);
+                int y0, int y1, const uint8_t* spp_map);
 void render_lt_cpu(const SceneView& sv, float* accum, int spp0, int nspp, uint32_t seed,
                    int spec_constraint, float caustic_scaling, int n_threads);
 }
@@ -349,23 +349,30 @@ struct SceneHolder {
 
     void render_host(farr accum, py::object var, int spp0, int nspp, uint32_t seed,
                      int renderer, int spec_constraint, float caustic_scaling, int n_threads,
-                     int y0 = 0, int y1 = 0) {
+                     int y0 = 0, int y1 = 0, py::object spp_map = py::none()) {
         finalize();
         float* vp = nullptr;
         farr var_arr;
         if (!var.is_none()) { var_arr = var.cast<farr>(); vp = var_arr.mutable_data(); }
+        const uint8_t* smp = nullptr;
+        py::array_t<uint8_t, py::array::c_style | py::array::forcecast> smp_arr;
+        if (!spp_map.is_none()) {
+            smp_arr = spp_map.cast<py::array_t<uint8_t, py::array::c_style | py::array::forcecast>>();
+            smp = smp_arr.data();
+        }
         py::gil_scoped_release rel;
         if (renderer == R_LIGHT_TRACE)
             render_lt_cpu(host_sv, accum.mutable_data(), spp0, nspp, seed,
                           spec_constraint, caustic_scaling, n_threads);
         else
             render_cpu(host_sv, accum.mutable_data(), vp, spp0, nspp, seed, renderer, n_threads,
-                       y0, y1);
+                       y0, y1, smp);
     }
 
     void render_device(uintptr_t accum_ptr, uintptr_t var_ptr, int spp0, int nspp,
                        uint32_t seed, int renderer, int spec_constraint,
-                       float caustic_scaling, uintptr_t stream, int y0 = 0, int y1 = 0) {
+                       float caustic_scaling, uintptr_t stream, int y0 = 0, int y1 = 0,
+                       uintptr_t spp_map_ptr = 0) {
         if (!has_dev) throw std::runtime_error("scene not uploaded to device");
         dev_sv.cam = cam;   // camera / depth params may have changed (hot reload)
         dev_sv.md = md;
@@ -381,7 +388,8 @@ struct SceneHolder {
             return;
         }
         HIP_OK(launch_render(dev_sv, (float*)accum_ptr, (float*)var_ptr, spp0, nspp, seed,
-                             renderer, spec_constraint, caustic_scaling, (void*)stream, y0, y1));
+                             renderer, spec_constraint, caustic_scaling, (void*)stream, y0, y1,
+                             (const uint8_t*)spp_map_ptr));
     }
 
     py::dict info() {
@@ -529,12 +537,12 @@ PYBIND11_MODULE(_C, m) {
              py::arg("accum"), py::arg("var"), py::arg("spp0"), py::arg("nspp"),
              py::arg("seed"), py::arg("renderer"), py::arg("spec_constraint"),
              py::arg("caustic_scaling"), py::arg("n_threads"),
-             py::arg("y0") = 0, py::arg("y1") = 0)
+             py::arg("y0") = 0, py::arg("y1") = 0, py::arg("spp_map") = py::none())
         .def("render_device", &SceneHolder::render_device,
              py::arg("accum_ptr"), py::arg("var_ptr"), py::arg("spp0"), py::arg("nspp"),
              py::arg("seed"), py::arg("renderer"), py::arg("spec_constraint"),
              py::arg("caustic_scaling"), py::arg("stream"),
-             py::arg("y0") = 0, py::arg("y1") = 0)
+             py::arg("y0") = 0, py::arg("y1") = 0, py::arg("spp_map_ptr") = 0)
         .def("info", &SceneHolder::info);
 
     m.def("build_bvh", &py_build_bvh, py::arg("prims"), py::arg("prim_obj"),

@@ -33,7 +33,7 @@ static void parallel_rows(int h, int n_threads, const std::function<void(int)>& 
 // 4=depth, 5=bvh_cost (light tracing has its own entry below).
 void render_cpu(const SceneView& sv, float* accum, float* var,
                 int spp0, int nspp, uint32_t seed, int renderer, int n_threads,
-                int y0, int y1) {
+                int y0, int y1, const uint8_t* spp_map) {
     const int w = sv.cam.w, h = sv.cam.h;
     if (y1 <= 0 || y1 > h) y1 = h;
     if (y0 < 0) y0 = 0;
@@ -41,9 +41,11 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
         int y = yr + y0;
         for (int x = 0; x < w; ++x) {
             size_t pix = size_t(y) * w + x;
+            const int nspp_px = spp_map ? (int)spp_map[pix] : nspp;
+            if (nspp_px == 0) continue;
             Vec3 Lsum(0.f);
             float lum_s = 0.f, lum_s2 = 0.f;
-            for (int s = 0; s < nspp; ++s) {
+            for (int s = 0; s < nspp_px; ++s) {
                 Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
                 Ray ray = sv.cam.gen_ray(x, y, sp);
                 Vec3 L(0.f);
@@ -58,7 +60,7 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
             accum[pix * 4 + 0] += Lsum.x;
             accum[pix * 4 + 1] += Lsum.y;
             accum[pix * 4 + 2] += Lsum.z;
-            accum[pix * 4 + 3] += (float)nspp;
+            accum[pix * 4 + 3] += (float)nspp_px;
             if (var) { var[pix * 2 + 0] += lum_s; var[pix * 2 + 1] += lum_s2; }
         }
     });

@@ -20,10 +20,13 @@ enum RendererKind : int {
 // Returns hipError_t as int.
 // y0/y1: optional row band [y0, y1) for tile-split DP (0,0 = full frame);
 // applies to the megakernel renderers (wavefront/lt always render full).
+// spp_map: optional per-pixel sample budget for this launch (adaptive
+// sampling; overrides nspp per pixel, 0 = skip pixel).
 int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling,
-                  void* stream, int y0 = 0, int y1 = 0);
+                  void* stream, int y0 = 0, int y1 = 0,
+                  const uint8_t* spp_map = nullptr);
 
 // Wavefront path tracer (SoA queues + compaction); state owned by WfState.
 struct WfState;

@@ -41,7 +41,8 @@ __attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
 // full 512/MINW VGPRs instead of hoarding occupancy it can't get (measured:
 // plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
-              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached, int y_off, int y_end) {
+              int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached, int y_off, int y_end,
+              const uint8_t* __restrict__ spp_map) {
     extern __shared__ uint64_t s_stk[];
     // dynamic-LDS layout: [n_cached 128-byte nodes][per-thread stacks].
     // The top of the tree is copied into LDS once per block: every walk's
@@ -61,6 +62,11 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
     if (px >= sv.cam.w || py >= y_end) return;   // y_end = band end (<= h)
     const size_t pix = size_t(py) * sv.cam.w + px;
 
+    // adaptive sampling: per-pixel sample budget for this launch (variance
+    // routed back into work allocation; accumulator alpha carries per-pixel
+    // counts, so heterogeneous spp still averages exactly)
+    const int nspp_px = spp_map ? (int)spp_map[pix] : nspp;
+    if (nspp_px == 0) return;
     Vec3 Lsum(0.f);
     float lum_s = 0.f, lum_s2 = 0.f;
     if constexpr (RENDERER == R_MEGAKERNEL_PT) {
@@ -79,13 +85,13 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
                 float lum = (L.x + L.y + L.z) * (1.f / 3.f);
                 lum_s += lum;
                 lum_s2 = fmaf(lum, lum, lum_s2);
-                if (++s >= nspp) break;
+                if (++s >= nspp_px) break;
                 sp = Sampler(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
                 ps.reset(sv.cam.gen_ray(px, py, sp));
             }
         }
     } else
-    for (int s = 0; s < nspp; ++s) {
+    for (int s = 0; s < nspp_px; ++s) {
         Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
         Ray ray = sv.cam.gen_ray(px, py, sp);
         Vec3 L(0.f);
@@ -99,7 +105,7 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
         lum_s2 = fmaf(lum, lum, lum_s2);
     }
     float* a = accum + pix * 4;
-    a[0] += Lsum.x; a[1] += Lsum.y; a[2] += Lsum.z; a[3] += (float)nspp;
+    a[0] += Lsum.x; a[1] += Lsum.y; a[2] += Lsum.z; a[3] += (float)nspp_px;
     if (var) { var[pix * 2 + 0] += lum_s; var[pix * 2 + 1] += lum_s2; }
 }
 
@@ -177,7 +183,7 @@ __global__ void k_add_count(float* __restrict__ accum, size_t npix, float cnt) {
 int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling, void* stream,
-                  int y0, int y1) {
+                  int y0, int y1, const uint8_t* spp_map) {
     hipStream_t st = (hipStream_t)stream;
     // HIPPT_SWIZZLE=1 enables the XCD band swizzle (measured -14% on the
     // kitchen megakernel — the round-robin XCD dispatch already spreads
@@ -232,15 +238,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     case R_VOLUME_PT:
         if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         else
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -257,15 +263,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
         break;
     }
     }

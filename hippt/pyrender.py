@@ -41,11 +41,15 @@ class PythonRenderer:
         self._released = False
 
     # -- reference API ------------------------------------------------------
-    def render(self, spp: int = 1):
+    def render(self, spp: int = 1, adaptive: bool = False):
         """Accumulate spp samples; returns deep-copied (H,W,4) torch tensor
-        with mean radiance RGB + accumulated spp in alpha."""
+        with mean radiance RGB + accumulated spp in alpha.  adaptive=True
+        routes the variance buffer into per-pixel sample budgets."""
         self._check()
-        self.renderer.render(spp)
+        if adaptive:
+            self.renderer.render_adaptive(spp)
+        else:
+            self.renderer.render(spp)
         raw = self.renderer.raw()
         if self.renderer.device is not None:
             return raw  # raw() already deep-copies (clone + divide)

@@ -81,7 +81,52 @@ class Renderer:
     def _seed(self) -> int:
         return (self.seed_offset * 4201) & 0xFFFFFFFF
 
-    def render(self, spp: int = 1, y0: int = 0, y1: int = 0):
+    def render_adaptive(self, spp: int, batch: int = 0):
+        """Variance-guided adaptive sampling (extension beyond the
+        reference): the Welford variance buffer is routed back into a
+        per-pixel sample budget, n_i ∝ σ_i (equal-error allocation).  The
+        first batch samples uniformly; later batches concentrate on noisy
+        pixels.  Exact estimator: the accumulator's alpha channel carries
+        per-pixel counts.  Megakernel renderers only."""
+        if self.rid not in (C.R_MEGAKERNEL_PT, C.R_VOLUME_PT):
+            raise ValueError("adaptive sampling needs a megakernel renderer")
+        batch = batch if batch > 0 else max(4, spp // 4)
+        first = min(batch, spp)
+        self.render(first)
+        done = first
+        while done < spp:
+            b = min(batch, spp - done)
+            m = self._spp_budget(b)
+            self.render(b, spp_map=m)
+            done += b
+        return self
+
+    def _spp_budget(self, batch: int):
+        """Per-pixel uint8 budget for `batch` average spp, n_i ∝ σ_i."""
+        if self.device is not None:
+            t = self.torch
+            n = self.accum[:, :, 3].clamp(min=1.0)
+            s, s2 = self.var[:, :, 0], self.var[:, :, 1]
+            var_mean = ((s2 / n - (s / n) ** 2).clamp(min=0.0) / n)
+            w = var_mean.sqrt()
+            # 4x4 box smooth for robustness at low counts
+            w = t.nn.functional.avg_pool2d(w[None, None], 4, 1, 2)[0, 0, :w.shape[0], :w.shape[1]]
+            w = w / w.mean().clamp(min=1e-12)
+            m = (w * batch).round().clamp(0, 255).to(t.uint8)
+            return m.contiguous()
+        import numpy as _np
+        n = _np.maximum(self.accum[:, :, 3], 1.0)
+        s, s2 = self.var[:, :, 0], self.var[:, :, 1]
+        var_mean = _np.maximum(s2 / n - (s / n) ** 2, 0.0) / n
+        w = _np.sqrt(var_mean)
+        k = _np.ones((4, 4), _np.float32) / 16.0
+        from numpy.lib.stride_tricks import sliding_window_view
+        pad = _np.pad(w, 2, mode="edge")
+        w = (sliding_window_view(pad, (4, 4))[:w.shape[0], :w.shape[1]] * k).sum(axis=(2, 3))
+        w = w / max(float(w.mean()), 1e-12)
+        return _np.clip(_np.round(w * batch), 0, 255).astype(_np.uint8)
+
+    def render(self, spp: int = 1, y0: int = 0, y1: int = 0, spp_map=None):
         """Accumulate spp more samples (reference render_raw semantics).
         y0/y1 restrict rendering to the row band [y0, y1) — tile-split DP
         (megakernel renderers only; 0,0 = full frame)."""
@@ -101,7 +146,7 @@ class Renderer:
                 self.scene.native.render_device(
                     self.accum.data_ptr(), self.var.data_ptr(), self.accum_cnt, spp,
                     self._seed(), self.rid, self.spec_constraint, self.caustic_scaling, stream,
-                    y0, y1)
+                    y0, y1, 0 if spp_map is None else spp_map.data_ptr())
             self.torch.cuda.synchronize(self.device)
         else:
             var = self.var.reshape(-1)
@@ -117,8 +162,15 @@ class Renderer:
                 self.scene.native.render_host(self.accum.reshape(-1), self.var.reshape(-1),
                                               self.accum_cnt, spp, self._seed(),
                                               self.rid, self.spec_constraint,
-                                              self.caustic_scaling, 0, y0, y1)
-        self.accum_cnt += spp
+                                              self.caustic_scaling, 0, y0, y1,
+                                              None if spp_map is None else spp_map.reshape(-1))
+        if spp_map is None:
+            self.accum_cnt += spp
+        else:
+            # heterogeneous budgets: advance the sampler-stream base past the
+            # largest per-pixel count so no pixel reuses a sample index
+            m = int(spp_map.max())
+            self.accum_cnt += max(spp, m)
         self.timer.add((time.perf_counter() - t0) * 1000.0)
         return self
 
