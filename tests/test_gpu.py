@@ -188,3 +188,16 @@ class TestExampleScenesGPU:
 
     def test_point_cbox(self):
         self._ab("point-cbox.xml")
+
+
+@pytest.mark.gpu
+def test_adaptive_sampling_gpu():
+    d = cornell_box(width=64, height=64, max_depth=4)
+    r = hippt.PythonRenderer(d, device_id=0)
+    img = r.render(spp=32, adaptive=True).cpu().numpy()
+    assert np.isfinite(img).all()
+    cnt = img[..., 3]
+    assert cnt.std() > 0.5 and cnt.min() >= 8
+    d2 = cornell_box(width=64, height=64, max_depth=4)
+    u = hippt.PythonRenderer(d2, device_id=0).render(spp=32).cpu().numpy()
+    assert abs(img[..., :3].mean() - u[..., :3].mean()) < 0.05 * u[..., :3].mean()
