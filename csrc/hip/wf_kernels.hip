@@ -94,15 +94,23 @@ void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n
 // block-local pixel adjacency): one global 256-bin histogram, a 1-block
 // exclusive scan producing the live count on-device, and an atomic-cursor
 // scatter.  No host round trip; 3 memory-bound passes over 4B/entry.
+// mode 0: key = material byte (grouped shade).  mode 1: key = dead bit only
+// (compaction without ray sorting — reference NO_RAY_SORTING/partition path,
+// defines.cuh:25-37; keeps arrival (pixel) order within live/dead).
+__device__ inline uint32_t sort_key(uint32_t status, int mode) {
+    uint32_t b = status >> 24;
+    return mode ? (b >= DEAD ? (uint32_t)DEAD : 0u) : b;
+}
+
 __global__ __launch_bounds__(SORT_BLOCK)
-void k_sort_hist(const uint32_t* __restrict__ in, int n, uint32_t* __restrict__ hist) {
+void k_sort_hist(const uint32_t* __restrict__ in, int n, uint32_t* __restrict__ hist, int mode) {
     __shared__ uint32_t lh[256];
     for (int t = threadIdx.x; t < 256; t += blockDim.x) lh[t] = 0;
     __syncthreads();
     int base = blockIdx.x * SORT_BLOCK * SORT_ITEMS;
     for (int k = 0; k < SORT_ITEMS; ++k) {
         int i = base + k * SORT_BLOCK + threadIdx.x;
-        if (i < n) atomicAdd(&lh[in[i] >> 24], 1u);
+        if (i < n) atomicAdd(&lh[sort_key(in[i], mode)], 1u);
     }
     __syncthreads();
     for (int t = threadIdx.x; t < 256; t += blockDim.x)
@@ -130,7 +138,7 @@ void k_sort_scan(uint32_t* __restrict__ hist, int* __restrict__ live_out) {
 
 __global__ __launch_bounds__(SORT_BLOCK)
 void k_sort_scatter(const uint32_t* __restrict__ in, int n,
-                    uint32_t* __restrict__ hist, uint32_t* __restrict__ out) {
+                    uint32_t* __restrict__ hist, uint32_t* __restrict__ out, int mode) {
     __shared__ uint32_t lbase[256];
     __shared__ uint32_t lcnt[256];
     for (int t = threadIdx.x; t < 256; t += blockDim.x) lcnt[t] = 0;
@@ -143,7 +151,7 @@ void k_sort_scatter(const uint32_t* __restrict__ in, int n,
         int i = base + k * SORT_BLOCK + threadIdx.x;
         if (i < n) {
             ent[k] = in[i];
-            rank[k] = atomicAdd(&lcnt[ent[k] >> 24], 1u);
+            rank[k] = atomicAdd(&lcnt[sort_key(ent[k], mode)], 1u);
         }
     }
     __syncthreads();
@@ -152,7 +160,7 @@ void k_sort_scatter(const uint32_t* __restrict__ in, int n,
     __syncthreads();
     for (int k = 0; k < SORT_ITEMS; ++k) {
         int i = base + k * SORT_BLOCK + threadIdx.x;
-        if (i < n) out[lbase[ent[k] >> 24] + rank[k]] = ent[k];
+        if (i < n) out[lbase[sort_key(ent[k], mode)] + rank[k]] = ent[k];
     }
 }
 
@@ -398,6 +406,12 @@ void wf_destroy(WfState* s) {
 
 int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, float* var,
                             int spp0, int nspp, uint32_t seed, int sort_mode, void* stream) {
+    if (sort_mode < 0 || sort_mode > 1) sort_mode = 0;
+    static int env_mode = [] {
+        const char* e = getenv("HIPPT_WF_SORT");
+        return (e && strcmp(e, "compact") == 0) ? 1 : -1;
+    }();
+    if (env_mode >= 0) sort_mode = env_mode;
     hipStream_t hs = (hipStream_t)stream;
     const int n = st->n;
     dim3 blk(WF_BLOCK);
@@ -435,11 +449,11 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             // launched full-width and exit by comparing against *live_dev.
             (void)hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
             hipLaunchKernelGGL(k_sort_hist, dim3(nb), dim3(SORT_BLOCK), 0, hs,
-                               st->status, n, st->hist);
+                               st->status, n, st->hist, sort_mode);
             hipLaunchKernelGGL(k_sort_scan, dim3(1), dim3(256), 0, hs,
                                st->hist, st->live_dev);
             hipLaunchKernelGGL(k_sort_scatter, dim3(nb), dim3(SORT_BLOCK), 0, hs,
-                               st->status, n, st->hist, st->order);
+                               st->status, n, st->hist, st->order, sort_mode);
             (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
@@ -448,7 +462,6 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
     }
-    (void)sort_mode;
     return (int)hipGetLastError();
 }
 
