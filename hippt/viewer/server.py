@@ -114,6 +114,7 @@ class ViewerApp:
         self.thread = None
         self.move_speed = 0.15
         self.rot_speed = 0.08
+        self.adaptive = False   # variance-guided per-pixel spp after warmup
 
     # ------------------------------------------------------------ lifecycle
     def start(self):
@@ -129,7 +130,13 @@ class ViewerApp:
     def _loop(self):
         while self.running:
             with self.lock:
-                self.pyr.renderer.render(self.spp_per_frame)
+                r = self.pyr.renderer
+                if (self.adaptive and r.accum_cnt >= 8
+                        and r.rid in (0, 2)):  # megakernel pt/vpt only
+                    m = r._spp_budget(self.spp_per_frame)
+                    r.render(self.spp_per_frame, spp_map=m)
+                else:
+                    r.render(self.spp_per_frame)
             time.sleep(0.0005)
 
     # ------------------------------------------------------------- frames
@@ -327,6 +334,11 @@ def build_app(viewer: ViewerApp):
     def renderer(req: RendererReq):
         viewer.set_renderer(req.kind)
         return {"ok": True}
+
+    @app.post("/api/adaptive")
+    def adaptive(req: AdaptiveReq):
+        viewer.adaptive = bool(req.enabled)
+        return {"ok": True, "adaptive": viewer.adaptive}
 
     @app.post("/api/bsdf")
     def bsdf(req: BsdfReq):

@@ -83,3 +83,15 @@ def test_renderer_switch(client):
     time.sleep(0.4)
     assert v.desc.config.renderer == "depth"
     c.post("/api/renderer", json={"kind": "pt"})
+
+
+def test_adaptive_toggle(client):
+    c, viewer = client
+    r = c.post("/api/adaptive", json={"enabled": True})
+    assert r.status_code == 200 and r.json()["adaptive"] is True
+    import time
+    time.sleep(0.8)   # a few adaptive frames past the warmup threshold
+    stats = c.get("/api/stats").json()
+    assert stats["spp"] > 0
+    r = c.post("/api/adaptive", json={"enabled": False})
+    assert r.json()["adaptive"] is False
