@@ -294,6 +294,9 @@ try:
     class RendererReq(_BaseModel):
         kind: str
 
+    class AdaptiveReq(_BaseModel):
+        enabled: bool = True
+
     class MoveReq(_BaseModel):
         key: str
 except ImportError:  # viewer optional without fastapi/pydantic
