@@ -17,9 +17,11 @@ def main():
     small = "--small" in sys.argv
     dev = -1 if cpu else 0
     os.makedirs(outdir, exist_ok=True)
-    xml_jobs = [("cornell-box.xml", 512), ("balls.xml", 512), ("grid-cbox.xml", 256),
-                ("diamonds.xml", 512), ("env-balls.xml", 256), ("caustics-lt.xml", 512),
-                ("tof-cbox.xml", 384), ("point-cbox.xml", 256)]
+    hq = "--hq" in sys.argv
+    m = 8 if hq else 1
+    xml_jobs = [("cornell-box.xml", 512 * m), ("balls.xml", 512 * m), ("grid-cbox.xml", 256 * m),
+                ("diamonds.xml", 512 * m), ("env-balls.xml", 256 * m), ("caustics-lt.xml", 512 * m),
+                ("tof-cbox.xml", 384 * m), ("point-cbox.xml", 256 * m)]
     for name, spp in xml_jobs:
         d = parse_xml(os.path.join(ROOT, "scenes", name))
         if small:
@@ -32,7 +34,7 @@ def main():
         r.save(out)
         print(f"{name} -> {out} ({r.avg_frame_time():.0f} ms/frame)", flush=True)
         r.release()
-    for gen, nm, spp in [(kitchen, "kitchen", 256), (sports_car, "sports-car", 256)]:
+    for gen, nm, spp in [(kitchen, "kitchen", 256 * m), (sports_car, "sports-car", 256 * m)]:
         w, h = (240, 135) if small else (960, 540)
         d = gen(width=w, height=h)
         r = hippt.PythonRenderer(d, device_id=dev)
