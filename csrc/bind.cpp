@@ -489,6 +489,14 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
         HitRecord w = ray_intersect_bvh4_ww(n4, pr, po, r, tmax);
         if ((w.prim_idx < 0) != (b.prim_idx < 0) ||
             (b.prim_idx >= 0 && fabsf(w.t - b.t) > 1e-5f * fmaxf(1.f, b.t))) { ++bad; continue; }
+        {
+            Bvh4Walk wk;
+            bvh4_walk_init(wk, r, tmax);
+            while (bvh4_walk_step(wk, n4, pr, po, nullptr, 0)) {}
+            if (wk.rec.prim_idx < 0) wk.rec.t = MAX_DIST;
+            if ((wk.rec.prim_idx < 0) != (b.prim_idx < 0) ||
+                (b.prim_idx >= 0 && fabsf(wk.rec.t - b.t) > 1e-5f * fmaxf(1.f, b.t))) { ++bad; continue; }
+        }
         if (n8p) {
             HitRecord w8 = ray_intersect_bvh8_ww(n8p, pr, po, r, tmax);
             bool occ8 = occlusion_test_bvh8(n8p, pr, po, r, tmax);

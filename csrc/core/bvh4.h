@@ -426,4 +426,98 @@ HD Vec2 bvh4_cost(const BVH4Node* nodes, const Prim* prims,
     }
 }
 
+// ----------------------------------------------------------------------
+// Dual-ray while-while walk: one thread advances TWO independent closest-hit
+// walks in lockstep steps, so each lane keeps two node loads in flight (the
+// single-ray walk stalls ~58% on L2-hit latency; a second independent chain
+// per lane doubles memory-level parallelism without needing more waves).
+// Used by the wavefront trace kernel (HIPPT_WF_DUAL).  Each walk gets half
+// of the thread's LDS stack slots.
+struct Bvh4Walk {
+    HitRecord rec;
+    uint32_t cur;       // tagged entry: bit31 leaf, DONE_W = finished
+    int sp;
+    Vec3 inv_d, o_div;
+    Ray ray;
+    uint64_t stack[BVH4_STACK / 2];
+};
+constexpr uint32_t BVH4_DONE_W = 0x7fffffffu;
+
+HD void bvh4_walk_init(Bvh4Walk& w, const Ray& ray, float tmax) {
+    w.rec = HitRecord();
+    w.rec.t = tmax;
+    w.cur = 0;
+    w.sp = 0;
+    w.inv_d = safe_rcp_dir(ray.d);
+    w.o_div = ray.o * w.inv_d;
+    w.ray = ray;
+}
+
+// One step: a node visit (test 4 children, descend/push) OR one leaf batch.
+// Returns true while the walk still has work.
+HD bool bvh4_walk_step(Bvh4Walk& w, const BVH4Node* nodes, const Prim* prims,
+                       const uint32_t* prim_obj, uint64_t* lds_slot, int lds_n) {
+    if (w.cur == BVH4_DONE_W) return false;
+    if (w.cur < 0x80000000u) {
+        const BVH4Node nd = nodes[w.cur];
+        uint32_t keys[4];
+        int nhit = 0;
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+            float t0x = fmaf(nd.lo_x[c], w.inv_d.x, -w.o_div.x);
+            float t1x = fmaf(nd.hi_x[c], w.inv_d.x, -w.o_div.x);
+            float t0y = fmaf(nd.lo_y[c], w.inv_d.y, -w.o_div.y);
+            float t1y = fmaf(nd.hi_y[c], w.inv_d.y, -w.o_div.y);
+            float t0z = fmaf(nd.lo_z[c], w.inv_d.z, -w.o_div.z);
+            float t1z = fmaf(nd.hi_z[c], w.inv_d.z, -w.o_div.z);
+            float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                fmaxf(fminf(t0z, t1z), 0.f));
+            float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                fminf(fmaxf(t0z, t1z), w.rec.t));
+            if (enter <= exit_) keys[nhit++] = (float_as_uint(enter) & ~3u) | (uint32_t)c;
+        }
+        if (nhit > 1) {
+            if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+            if (nhit > 2) {
+                if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                if (nhit > 3) {
+                    if (keys[2] > keys[3]) { uint32_t t = keys[2]; keys[2] = keys[3]; keys[3] = t; }
+                    if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                    if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                }
+            }
+        }
+        uint32_t next = BVH4_DONE_W;
+        for (int k = nhit - 1; k >= 0; --k) {
+            int c = (int)(keys[k] & 3u);
+            int ch = nd.child[c];
+            int pc = nd.cnt[c];
+            if (ch < 0 && pc == 0) continue;
+            uint32_t lo = ch < 0
+                ? (0x80000000u | ((uint32_t)pc << 27) | (uint32_t)(~ch))
+                : (uint32_t)ch;
+            if (k == 0) {
+                next = lo;
+            } else {
+                uint64_t e = ((uint64_t)(keys[k] & ~3u) << 32) | lo;
+                if (w.sp < lds_n) lds_slot[w.sp * BVH4_LDS_STRIDE] = e;
+                else w.stack[w.sp - lds_n] = e;
+                ++w.sp;
+            }
+        }
+        if (next != BVH4_DONE_W) { w.cur = next; return true; }
+    } else {
+        bvh4_leaf_hit(prims, prim_obj, w.ray, (int)(w.cur & 0x07ffffffu),
+                      (int)((w.cur >> 27) & 0xfu), w.rec);
+    }
+    // pop (culling stale subtrees)
+    for (;;) {
+        if (w.sp == 0) { w.cur = BVH4_DONE_W; return false; }
+        --w.sp;
+        uint64_t e = w.sp < lds_n ? lds_slot[w.sp * BVH4_LDS_STRIDE] : w.stack[w.sp - lds_n];
+        if (uint_as_float((uint32_t)(e >> 32)) < w.rec.t) { w.cur = (uint32_t)e; return true; }
+    }
+}
+
 } // namespace hippt

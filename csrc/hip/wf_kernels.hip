@@ -303,21 +303,12 @@ void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
 }
 
 // ------------------------------------------------------- next closest hit
-__global__ __launch_bounds__(256)
-__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
-void k_wf_trace(SceneView sv, WfState st, int lds_n) {
-    extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n};
-    // pixel-order scan: bounce rays of neighboring pixels traverse similar
-    // BVH subtrees, so keeping trace in payload order (NOT material-sorted
-    // order) preserves wave-level spatial coherence — measured 4.5ms -> see
-    // profiles/ for the sorted-order version this replaces.
-    int i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= st.n) return;
-    if (st.status[i] >> 24 >= DEAD) return;  // dead or terminated in shade
-    float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i];
-    Ray ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
-    HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
+// trace epilogue shared by the single- and dual-ray trace kernels: store
+// the hit record, accumulate envmap MIS on miss, refine the status byte.
+__device__ inline void wf_trace_finish(const SceneView& sv, WfState& st, int i,
+                                       const Ray& ray, float prev_pdf,
+                                       HitRecord hit) {
+    if (hit.prim_idx < 0) hit.t = MAX_DIST;
     st.hit[i] = make_float4(hit.t, hit.u, hit.v, int_as_float(hit.prim_idx));
     uint32_t status;
     if (hit.prim_idx < 0) {
@@ -333,7 +324,7 @@ void k_wf_trace(SceneView sv, WfState st, int lds_n) {
                 float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
                 float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d,
                                                   Vec3(pn4.x, pn4.y, pn4.z)) * sel_pdf;
-                w = mis_weight(rd4.w, light_pdf);
+                w = mis_weight(prev_pdf, light_pdf);
             }
             float4 l4 = st.L[i];
             Vec3 L = Vec3(l4.x, l4.y, l4.z) + Vec3(thp4.x, thp4.y, thp4.z) * le * w;
@@ -344,6 +335,65 @@ void k_wf_trace(SceneView sv, WfState st, int lds_n) {
         status = (uint32_t)(sv.objs[oi].bsdf_id & 0x3F);
     }
     st.status[i] = (status << 24) | (uint32_t)i;
+}
+
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
+void k_wf_trace(SceneView sv, WfState st, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
+    // pixel-order scan: bounce rays of neighboring pixels traverse similar
+    // BVH subtrees, so keeping trace in payload order (NOT material-sorted
+    // order) preserves wave-level spatial coherence.
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= st.n) return;
+    if (st.status[i] >> 24 >= DEAD) return;  // dead or terminated in shade
+    float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i];
+    Ray ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
+    HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
+    wf_trace_finish(sv, st, i, ray, rd4.w, hit);
+}
+
+// Dual-ray trace: each lane advances two independent walks in lockstep
+// steps, keeping two node loads in flight per lane (the single walk stalls
+// ~58% of cycles on L2-hit latency).  Rays i and i + n/2 pair up so both
+// halves stay in pixel order.  HIPPT_WF_DUAL selects this kernel.
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(4, 4)))   // 2x walk state needs registers
+void k_wf_trace_dual(SceneView sv, WfState st, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    const int tid = threadIdx.x;
+    const int half = (st.n + 1) >> 1;
+    const int i0 = blockIdx.x * blockDim.x + tid;
+    const int i1 = i0 + half;
+    const int lds_half = lds_n >> 1;
+    uint64_t* slot0 = &s_stk[tid];
+    uint64_t* slot1 = &s_stk[tid + (size_t)lds_half * BVH4_LDS_STRIDE];
+    bool a0 = i0 < half && (st.status[i0] >> 24) < DEAD;
+    bool a1 = i1 < st.n && (st.status[i1] >> 24) < DEAD;
+    if (!a0 && !a1) return;
+    Ray ray0, ray1;
+    float pdf0 = 0.f, pdf1 = 0.f;
+    Bvh4Walk w0, w1;
+    if (a0) {
+        float4 ro = st.ray_o[i0], rd = st.ray_d[i0];
+        ray0 = Ray(Vec3(ro.x, ro.y, ro.z), Vec3(rd.x, rd.y, rd.z));
+        pdf0 = rd.w;
+        bvh4_walk_init(w0, ray0, MAX_DIST);
+    }
+    if (a1) {
+        float4 ro = st.ray_o[i1], rd = st.ray_d[i1];
+        ray1 = Ray(Vec3(ro.x, ro.y, ro.z), Vec3(rd.x, rd.y, rd.z));
+        pdf1 = rd.w;
+        bvh4_walk_init(w1, ray1, MAX_DIST);
+    }
+    bool r0 = a0, r1 = a1;
+    while (r0 | r1) {
+        if (r0) r0 = bvh4_walk_step(w0, sv.nodes4, sv.prims, sv.prim_obj, slot0, lds_half);
+        if (r1) r1 = bvh4_walk_step(w1, sv.nodes4, sv.prims, sv.prim_obj, slot1, lds_half);
+    }
+    if (a0) wf_trace_finish(sv, st, i0, ray0, pdf0, w0.rec);
+    if (a1) wf_trace_finish(sv, st, i1, ray1, pdf1, w1.rec);
 }
 
 // ----------------------------------------------------------------- splat
@@ -439,6 +489,10 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res;
     }();
     const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
+    static int wf_dual = [] {
+        const char* e = getenv("HIPPT_WF_DUAL");
+        return e ? atoi(e) : 0;
+    }();
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
         const int nb = st->nb_sort;
@@ -458,7 +512,12 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
                                bounce);
             hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n);
-            hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
+            if (wf_dual) {
+                dim3 grd_h(((st->n + 1) / 2 + WF_BLOCK - 1) / WF_BLOCK);
+                hipLaunchKernelGGL(k_wf_trace_dual, grd_h, blk, shmem, hs, sv, *st, lds_n);
+            } else {
+                hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
+            }
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
     }
