@@ -516,6 +516,26 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
     return bad;
 }
 
+// Closest-hit query over the BVH4 tree: returns (t, prim) arrays for rays.
+py::tuple py_bvh4_hit(farr prims, uarr prim_obj, farr nodes4,
+                      farr ray_o, farr ray_d) {
+    const BVH4Node* n4 = (const BVH4Node*)nodes4.data();
+    const Prim* pr = (const Prim*)prims.data();
+    const uint32_t* po = prim_obj.data();
+    int nr = (int)ray_o.shape(0);
+    farr t_out(nr);
+    iarr p_out(nr);
+    for (int i = 0; i < nr; ++i) {
+        Ray r;
+        r.o = {ray_o.at(i, 0), ray_o.at(i, 1), ray_o.at(i, 2)};
+        r.d = {ray_d.at(i, 0), ray_d.at(i, 1), ray_d.at(i, 2)};
+        HitRecord h = ray_intersect_bvh4_ww(n4, pr, po, r, MAX_DIST);
+        t_out.mutable_at(i) = h.t;
+        p_out.mutable_at(i) = h.prim_idx;
+    }
+    return py::make_tuple(t_out, p_out);
+}
+
 } // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -558,6 +578,7 @@ PYBIND11_MODULE(_C, m) {
           py::arg("use_sbvh") = false, py::arg("ref_unsplit") = true);
     m.def("collapse_bvh4", &py_collapse_bvh4, py::arg("nodes"));
     m.def("collapse_bvh8", &py_collapse_bvh8, py::arg("nodes"));
+    m.def("bvh4_hit", &py_bvh4_hit);
     m.def("bvh4_selftest", &py_bvh4_selftest,
           py::arg("prims"), py::arg("prim_obj"), py::arg("nodes"), py::arg("nodes4"),
           py::arg("ray_o"), py::arg("ray_d"), py::arg("tmax"),

@@ -203,3 +203,54 @@ class TestBVH4:
         nodes8, depth8 = C.collapse_bvh8(nodes)
         o, d = self._rays(4000, seed=24)
         assert C.bvh4_selftest(prims, pobj, nodes, nodes4, o, d, 1e7, nodes8) == 0
+
+
+class TestTraversalGroundTruth:
+    """BVH4 walk vs O(n) brute force — catches tree-build errors that
+    walk-vs-walk agreement cannot (both walks share the tree)."""
+
+    def test_bvh4_vs_bruteforce(self):
+        rng = np.random.default_rng(31)
+        tris = _random_tris(400, seed=31, spread=1.0)
+        prims = _to_prims(tris)
+        pobj = np.zeros(len(tris), np.uint32)
+        nodes, order, _ = C.build_bvh(prims, pobj, 4, 0.6, False, True)
+        prims_r = np.ascontiguousarray(prims[order])
+        pobj_r = np.ascontiguousarray(pobj[order])
+        nodes4, _ = C.collapse_bvh4(nodes)
+        tris_r = tris[order].astype(np.float64)
+
+        m = 400
+        o = rng.uniform(-3, 3, (m, 3))
+        aim = rng.uniform(-0.8, 0.8, (m, 3))   # aim at the cloud
+        d = aim - o
+        d /= np.linalg.norm(d, axis=1, keepdims=True)
+
+        # vectorized Moller-Trumbore ground truth (float64)
+        v0 = tris_r[:, 0][None]                     # (1,n,3)
+        e1 = (tris_r[:, 1] - tris_r[:, 0])[None]
+        e2 = (tris_r[:, 2] - tris_r[:, 0])[None]
+        D = d[:, None]                              # (m,1,3)
+        O = o[:, None]
+        p = np.cross(D, e2)
+        det = np.einsum("mnk,mnk->mn", np.broadcast_to(e1, p.shape), p)
+        with np.errstate(divide="ignore", invalid="ignore"):
+            inv = np.where(np.abs(det) > 1e-12, 1.0 / det, 0.0)
+            t0 = O - v0
+            u = np.einsum("mnk,mnk->mn", t0, p) * inv
+            q = np.cross(t0, np.broadcast_to(e1, t0.shape))
+            v = np.einsum("mnk,mnk->mn", np.broadcast_to(D, q.shape), q) * inv
+            t = np.einsum("mnk,mnk->mn", np.broadcast_to(e2, q.shape), q) * inv
+        valid = (np.abs(det) > 1e-12) & (u >= 0) & (v >= 0) & (u + v <= 1) & (t > 1e-3)
+        t = np.where(valid, t, np.inf)
+        gt_t = t.min(axis=1)
+
+        bt, bp = C.bvh4_hit(prims_r, pobj_r, nodes4,
+                            o.astype(np.float32), d.astype(np.float32))
+        hits = np.isfinite(gt_t)
+        assert hits.sum() > 100
+        # hit/miss agreement (tolerate fp32-vs-fp64 edge grazing)
+        agree = (bp >= 0) == hits
+        assert agree.mean() > 0.99, f"hit/miss disagreements: {(~agree).sum()}"
+        both = hits & (bp >= 0)
+        np.testing.assert_allclose(bt[both], gt_t[both], rtol=2e-3, atol=1e-3)
