@@ -254,3 +254,48 @@ class TestTraversalGroundTruth:
         assert agree.mean() > 0.99, f"hit/miss disagreements: {(~agree).sum()}"
         both = hits & (bp >= 0)
         np.testing.assert_allclose(bt[both], gt_t[both], rtol=2e-3, atol=1e-3)
+
+
+try:
+    from hypothesis import given, settings, strategies as st
+
+    class TestPropertyBased:
+        """Property-based traversal checks (hypothesis)."""
+
+        @given(seed=st.integers(0, 10_000), n=st.integers(2, 300),
+               use_sbvh=st.booleans())
+        @settings(max_examples=30, deadline=None)
+        def test_any_tree_agrees(self, seed, n, use_sbvh):
+            tris = _random_tris(n, seed=seed, spread=2.0)
+            prims = _to_prims(tris)
+            pobj = np.zeros(n, np.uint32)
+            nodes, order, _ = C.build_bvh(prims, pobj, 4, 0.6, use_sbvh, True)
+            pr = np.ascontiguousarray(prims[order % n])
+            po = np.ascontiguousarray(pobj[order % n])
+            nodes4, depth4 = C.collapse_bvh4(nodes)
+            assert 3 * depth4 <= 64
+            rng = np.random.default_rng(seed + 1)
+            o = rng.uniform(-4, 4, (200, 3)).astype(np.float32)
+            d = rng.normal(size=(200, 3)).astype(np.float32)
+            d /= np.maximum(np.linalg.norm(d, axis=1, keepdims=True), 1e-12)
+            assert C.bvh4_selftest(pr, po, nodes, nodes4, o, d, 1e7) == 0
+
+        @given(lo=st.floats(-100, 100), ext=st.floats(1e-4, 100),
+               seed=st.integers(0, 1000))
+        @settings(max_examples=20, deadline=None)
+        def test_translated_scaled_scene(self, lo, ext, seed):
+            # traversal robust to arbitrary world scales/offsets
+            tris = _random_tris(64, seed=seed, spread=1.0) * ext + lo
+            prims = _to_prims(tris.astype(np.float32))
+            pobj = np.zeros(64, np.uint32)
+            nodes, order, _ = C.build_bvh(prims, pobj, 4, 0.0, False, True)
+            pr = np.ascontiguousarray(prims[order])
+            po = np.ascontiguousarray(pobj[order])
+            nodes4, _ = C.collapse_bvh4(nodes)
+            rng = np.random.default_rng(seed)
+            o = (rng.uniform(-2, 2, (100, 3)) * ext + lo).astype(np.float32)
+            d = rng.normal(size=(100, 3)).astype(np.float32)
+            d /= np.maximum(np.linalg.norm(d, axis=1, keepdims=True), 1e-12)
+            assert C.bvh4_selftest(pr, po, nodes, nodes4, o, d, 1e7) == 0
+except ImportError:  # pragma: no cover
+    pass
