@@ -31,6 +31,8 @@ def main(argv=None):
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--colormap", default="plasma", help="depth/bvh-cost false color")
     ap.add_argument("--variance", default=None, help="also write the variance map here")
+    ap.add_argument("--adaptive", action="store_true",
+                    help="variance-guided per-pixel sample allocation")
     args = ap.parse_args(argv)
 
     import hippt
@@ -69,7 +71,10 @@ def main(argv=None):
     done = 0
     while done < spp:
         step = min(chunk, spp - done)
-        r.renderer.render(step)
+        if args.adaptive and done >= chunk and desc.config.renderer in ("pt", "pt-dyn", "vpt"):
+            r.renderer.render(step, spp_map=r.renderer._spp_budget(step))
+        else:
+            r.renderer.render(step)
         done += step
         el = time.perf_counter() - t0
         print(f"\r[hippt] {done}/{spp} spp, {el:.1f}s, "

@@ -45,6 +45,11 @@ def build_argparser():
     ap.add_argument("--output", type=str, default="ddp_render.png")
     ap.add_argument("--logdir", type=str, default=None, help="TensorBoard logdir")
     ap.add_argument("--cpu", action="store_true", help="gloo/CPU path (tests)")
+    ap.add_argument("--checkpoint", type=str, default=None,
+                    help="periodic per-rank accumulator snapshot path "
+                         "(rank id is appended); resumes from it if present")
+    ap.add_argument("--checkpoint-interval", type=int, default=256,
+                    help="steps between snapshots")
     return ap
 
 
@@ -111,6 +116,12 @@ def main(argv=None):
                              seed_offset=seed_offset)
     rend = r.renderer
 
+    ckpt_path = f"{args.checkpoint}.rank{rank}.npz" if args.checkpoint else None
+    if ckpt_path and os.path.exists(ckpt_path):
+        r.load_state(ckpt_path)
+        if rank == 0:
+            print(f"[ddp] resumed from {ckpt_path} at {r.counter()} spp", flush=True)
+
     stop = {"flag": False}
     if rank == 0:
         def _sigint(sig, frame):
@@ -139,8 +150,11 @@ def main(argv=None):
     steps = (spp_per_rank + args.spp_per_call - 1) // args.spp_per_call
     t_start = time.perf_counter()
     merged = None
-    for k in range(steps):
+    start_step = min(rend.accum_cnt // args.spp_per_call, steps)
+    for k in range(start_step, steps):
         rend.render(args.spp_per_call, y0=band[0], y1=band[1])
+        if ckpt_path and (k + 1) % args.checkpoint_interval == 0:
+            r.save_state(ckpt_path)
         if world_size > 1 and ((k + 1) % args.reduce_interval == 0 or k == steps - 1):
             merged, total_spp = reduce_rendered_image(dist, rend, world_size, cpu=args.cpu)
             # frame-time all_gather (ddp_render.py:192-211)

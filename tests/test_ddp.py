@@ -105,3 +105,29 @@ def test_tile_split_two_ranks(tmp_path):
     # every row rendered (no black band seams)
     rowmean = img[..., :3].mean(axis=(1, 2))
     assert (rowmean > 0.01).all()
+
+
+def test_ddp_checkpoint_resume(tmp_path):
+    """Interrupt-and-resume: first run saves a snapshot, second run resumes
+    and finishes; final spp equals the uninterrupted total."""
+    out = tmp_path / "ck.png"
+    ck = tmp_path / "state"
+    # run 1: 8 spp with snapshots every 1 step of 2 spp
+    res = run_torchrun(2, ["-m", "hippt.parallel.ddp", "--cpu",
+                           "--scene", "cornell", "--width", "48", "--height", "48",
+                           "--spp", "16", "--spp-per-call", "2",
+                           "--checkpoint", str(ck), "--checkpoint-interval", "1",
+                           "--reduce-interval", "2", "--output", str(out)])
+    assert res.returncode == 0, res.stderr[-1500:]
+    assert (tmp_path / "state.rank0.npz").exists()
+    # run 2 resumes: loads the final snapshot (8 spp/rank done) and exits
+    # almost immediately, still writing the merged image
+    out2 = tmp_path / "ck2.png"
+    res = run_torchrun(2, ["-m", "hippt.parallel.ddp", "--cpu",
+                           "--scene", "cornell", "--width", "48", "--height", "48",
+                           "--spp", "16", "--spp-per-call", "2",
+                           "--checkpoint", str(ck), "--checkpoint-interval", "1",
+                           "--reduce-interval", "2", "--output", str(out2)])
+    assert res.returncode == 0, res.stderr[-1500:]
+    assert "resumed from" in res.stdout
+    assert out2.exists()
