@@ -105,3 +105,33 @@ class TestExampleScenes:
         d, img = self._render("point-cbox.xml")
         assert d.emitters[0].type == "point"
         assert img[..., :3].mean() > 0.01
+
+
+class TestParserRobustness:
+    def test_malformed_xml(self, tmp_path):
+        p = tmp_path / "bad.xml"
+        p.write_text("<scene version='1.2'><renderer type='pt'>")
+        import xml.etree.ElementTree as ET
+        import pytest as _pt
+        with _pt.raises(ET.ParseError):
+            parse_xml(str(p))
+
+    def test_missing_mesh(self, tmp_path):
+        p = tmp_path / "m.xml"
+        p.write_text("""<scene version='1.2'>
+          <brdf type='lambertian' id='w'><rgb name='k_d' value='0.5'/></brdf>
+          <shape type='obj'><string name='filename' value='nope.obj'/>
+            <ref type='material' id='w'/></shape></scene>""")
+        import pytest as _pt
+        with _pt.raises(FileNotFoundError):
+            parse_xml(str(p))
+
+    def test_unknown_refs_dont_crash(self, tmp_path):
+        p = tmp_path / "r.xml"
+        p.write_text("""<scene version='1.2'>
+          <brdf type='lambertian' id='w'><rgb name='k_d' value='0.5'/></brdf>
+          <shape type='sphere'><point name='center' value='0,0,0'/>
+            <float name='radius' value='1'/>
+            <ref type='material' id='does-not-exist'/></shape></scene>""")
+        d = parse_xml(str(p))   # falls back to bsdf 0
+        assert len(d.objects) == 1 and d.objects[0].bsdf == 0
