@@ -110,7 +110,11 @@ class Renderer:
             var_mean = ((s2 / n - (s / n) ** 2).clamp(min=0.0) / n)
             w = var_mean.sqrt()
             # 4x4 box smooth for robustness at low counts
-            w = t.nn.functional.avg_pool2d(w[None, None], 4, 1, 2)[0, 0, :w.shape[0], :w.shape[1]]
+            w = t.nn.functional.avg_pool2d(w[None, None], 5, 1, 2)[0, 0]
+            w = w / w.mean().clamp(min=1e-12)
+            # bound the redistribution: heavy-tailed (firefly) pixels would
+            # otherwise capture the whole budget and starve the rest
+            w = w.clamp(0.25, 4.0)
             w = w / w.mean().clamp(min=1e-12)
             m = (w * batch).round().clamp(0, 255).to(t.uint8)
             return m.contiguous()
@@ -119,10 +123,12 @@ class Renderer:
         s, s2 = self.var[:, :, 0], self.var[:, :, 1]
         var_mean = _np.maximum(s2 / n - (s / n) ** 2, 0.0) / n
         w = _np.sqrt(var_mean)
-        k = _np.ones((4, 4), _np.float32) / 16.0
+        k = _np.ones((5, 5), _np.float32) / 25.0
         from numpy.lib.stride_tricks import sliding_window_view
         pad = _np.pad(w, 2, mode="edge")
-        w = (sliding_window_view(pad, (4, 4))[:w.shape[0], :w.shape[1]] * k).sum(axis=(2, 3))
+        w = (sliding_window_view(pad, (5, 5))[:w.shape[0], :w.shape[1]] * k).sum(axis=(2, 3))
+        w = w / max(float(w.mean()), 1e-12)
+        w = _np.clip(w, 0.25, 4.0)          # bounded redistribution (fireflies)
         w = w / max(float(w.mean()), 1e-12)
         return _np.clip(_np.round(w * batch), 0, 255).astype(_np.uint8)
 
@@ -187,14 +193,17 @@ class Renderer:
         return out
 
     def variance(self):
-        """Per-pixel variance of the mean luminance estimate (h,w,1)."""
-        n = max(self.accum_cnt, 1)
+        """Per-pixel variance of the mean luminance estimate (h,w,1).
+        Uses the PER-PIXEL sample count (accumulator alpha) so it stays
+        correct under adaptive sampling and band rendering."""
         if self.device is not None:
+            n = self.accum[:, :, 3].clamp(min=1.0)
             s, s2 = self.var[:, :, 0], self.var[:, :, 1]
-            v = (s2 - s * s / n) / max(n - 1, 1) / n
+            v = (s2 - s * s / n) / (n - 1).clamp(min=1.0) / n
             return v.clamp(min=0).unsqueeze(-1)
+        n = np.maximum(self.accum[:, :, 3], 1.0)
         s, s2 = self.var[:, :, 0], self.var[:, :, 1]
-        v = (s2 - s * s / n) / max(n - 1, 1) / n
+        v = (s2 - s * s / n) / np.maximum(n - 1, 1.0) / n
         return np.clip(v, 0, None)[:, :, None]
 
     # ----------------------------------------------------------- utilities

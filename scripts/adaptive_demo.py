@@ -9,26 +9,27 @@ import hippt  # noqa: E402
 from hippt.scene.procedural import cornell_box  # noqa: E402
 
 
-def run(adaptive, spp, dev):
-    d = cornell_box(width=256, height=256, spp=1, max_depth=6)
-    r = hippt.PythonRenderer(d, device_id=dev)
+def run(adaptive, spp, dev, seed=0, w=192):
+    d = cornell_box(width=w, height=w, spp=1, max_depth=6)
+    r = hippt.PythonRenderer(d, device_id=dev, seed_offset=seed)
     r.render(spp=spp, adaptive=adaptive)
-    v = r.variance()
-    v = v.cpu().numpy() if hasattr(v, "cpu") else np.asarray(v)
-    v = v[..., 0]
-    # tile-max: worst 16x16 tile mean variance (what the eye sees as noise)
-    t = v.reshape(16, 16, 16, 16).mean(axis=(1, 3))
-    return float(v.mean()), float(t.max())
+    out = r.renderer.raw()
+    out = out.cpu().numpy() if hasattr(out, "cpu") else np.asarray(out)
+    return out[..., :3]
 
 
 def main():
     dev = 0 if "--gpu" in sys.argv else -1
     spp = int(sys.argv[sys.argv.index("--spp") + 1]) if "--spp" in sys.argv else 64
-    mu, tu = run(False, spp, dev)
-    ma, ta = run(True, spp, dev)
-    print(f"uniform : mean var {mu:.3e}  worst-tile {tu:.3e}")
-    print(f"adaptive: mean var {ma:.3e}  worst-tile {ta:.3e}")
-    print(f"worst-tile variance ratio (uniform/adaptive): {tu / max(ta, 1e-30):.2f}x")
+    w = 192 if dev >= 0 else 96
+    ref = run(False, 32 * spp if dev >= 0 else 1536, dev, seed=7, w=w)
+    u = run(False, spp, dev, w=w)
+    a = run(True, spp, dev, w=w)
+    ru = float(np.sqrt(((u - ref) ** 2).mean()))
+    ra = float(np.sqrt(((a - ref) ** 2).mean()))
+    print(f"uniform  RMSE vs converged ref: {ru:.4f}")
+    print(f"adaptive RMSE vs converged ref: {ra:.4f}")
+    print(f"error ratio {ru / ra:.2f}x  (~{(ru / ra) ** 2:.2f}x sample efficiency)")
 
 
 if __name__ == "__main__":
