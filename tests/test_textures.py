@@ -33,8 +33,8 @@ def textured_quad_scene(tex_slots, bsdf_kw=None, n=64):
         slots[s] = 1 if s == "normal" and len(d.textures) > 1 else 0
     d.bsdfs = [BsdfDesc(type="lambertian", kd=(1, 1, 1), textures=slots),
                BsdfDesc(type="lambertian", kd=(0.8, 0.8, 0.8))]
-    tris = quad((-1, 0, -1), (1, 0, -1), (1, 0, 1), (-1, 0, 1))
-    uvs = np.array([[(0, 0), (1, 0), (1, 1)], [(0, 0), (1, 1), (0, 1)]], np.float32)
+    tris = quad((-1, 0, -1), (-1, 0, 1), (1, 0, 1), (1, 0, -1))  # +y normal
+    uvs = np.array([[(0, 0), (0, 1), (1, 1)], [(0, 0), (1, 1), (1, 0)]], np.float32)
     d.objects = [ObjectDesc(tris=tris, uvs=uvs, bsdf=0)]
     d.emitters = [EmitterDesc(type="point", pos=(0.0, 3.0, 0.0),
                               emission=(1, 1, 1), scale=20.0)]
