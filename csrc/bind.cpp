@@ -56,6 +56,8 @@ struct SceneHolder {
     std::vector<TexView> tex_host;
     std::vector<int> emitter_prims;
     std::vector<float> emitter_cdf;
+    std::vector<float> env_rows_h, env_cols_h;
+    int env_w = 0, env_h = 0;
     Camera cam{};
     MaxDepthParams md{};
     int env_emitter = -1;
@@ -158,6 +160,15 @@ struct SceneHolder {
     void set_emitter_prims(iarr eprims, farr ecdf) {
         emitter_prims.assign(eprims.data(), eprims.data() + eprims.size());
         emitter_cdf.assign(ecdf.data(), ecdf.data() + ecdf.size());
+    }
+
+    void set_env_cdf(farr rows, farr cols) {
+        if (cols.ndim() != 2) throw std::runtime_error("env cols must be (h,w)");
+        env_h = (int)cols.shape(0);
+        env_w = (int)cols.shape(1);
+        env_rows_h.assign(rows.data(), rows.data() + rows.size());
+        env_cols_h.assign(cols.data(), cols.data() + cols.size());
+        if ((int)env_rows_h.size() != env_h) throw std::runtime_error("env rows/cols mismatch");
     }
 
     int add_phase(int type, float g1, float g2, float wmix) {
@@ -266,6 +277,8 @@ struct SceneHolder {
         sv.n_textures = (int)tex_host.size();
         sv.n_media = (int)media.size();
         sv.env_emitter = env_emitter;
+        sv.env_w = env_w;
+        sv.env_h = env_h;
         sv.cam_medium = cam_medium;
         sv.cam = cam;
         sv.md = md;
@@ -284,6 +297,8 @@ struct SceneHolder {
         host_sv.emitters = emitters.data();
         host_sv.emitter_prims = emitter_prims.data();
         host_sv.emitter_cdf = emitter_cdf.data();
+        host_sv.env_rows = env_rows_h.empty() ? nullptr : env_rows_h.data();
+        host_sv.env_cols = env_cols_h.empty() ? nullptr : env_cols_h.data();
         host_sv.textures = tex_host.data();
         host_sv.media = media.data();
         host_sv.phases = phases.data();
@@ -316,6 +331,8 @@ struct SceneHolder {
         dev_sv.objs = upload_vec((const ObjInfo*)np_objs.data(), np_objs.shape(0));
         dev_sv.emitter_prims = upload_vec(emitter_prims.data(), emitter_prims.size());
         dev_sv.emitter_cdf = upload_vec(emitter_cdf.data(), emitter_cdf.size());
+        dev_sv.env_rows = upload_vec(env_rows_h.data(), env_rows_h.size());
+        dev_sv.env_cols = upload_vec(env_cols_h.data(), env_cols_h.size());
         // textures: RGBA32F rows in device global memory (gfx950 has no
         // device texture units — software bilinear is the CDNA-native path)
         tex_dev.clear();
@@ -550,6 +567,7 @@ PYBIND11_MODULE(_C, m) {
         .def("add_emitter", &SceneHolder::add_emitter)
         .def("update_emitter", &SceneHolder::update_emitter)
         .def("set_emitter_prims", &SceneHolder::set_emitter_prims)
+        .def("set_env_cdf", &SceneHolder::set_env_cdf)
         .def("add_phase", &SceneHolder::add_phase)
         .def("add_medium", &SceneHolder::add_medium)
         .def("update_medium", &SceneHolder::update_medium)

@@ -52,7 +52,24 @@ struct EmitterGeom {
     const int* emitter_prims;     // BVH-reordered primitive ids per emitter
     const float* emitter_cdf;     // per-emitter cumulative areas, normalized [0,1]
     const TexView* textures;
+    // envmap importance sampling (beyond reference: the reference only does
+    // cosine-hemisphere NEE, emitter.cu:25-73): row-marginal and per-row
+    // conditional CDFs over luminance x sin(theta); null = cosine fallback
+    const float* env_rows;        // (h) cumulative
+    const float* env_cols;        // (h*w) cumulative per row
+    int env_w, env_h;
 };
+
+// first index with cdf[i] > u (cdf ascending, cdf[n-1] == 1)
+HD int cdf_find(const float* cdf, int n, float u) {
+    int lo = 0, hi = n - 1;
+    while (lo < hi) {
+        int mid = (lo + hi) >> 1;
+        if (cdf[mid] > u) hi = mid;
+        else lo = mid + 1;
+    }
+    return lo;
+}
 
 HD Vec3 emitter_radiance_tex(const EmitterParams& e, const TexView* textures, Vec2 uv) {
     Vec3 rad = e.emission.xyz() * e.emission.w;
@@ -143,6 +160,35 @@ HD EmitterSampleRec emitter_sample(const EmitterParams& e, const EmitterGeom& g,
         return r;
     }
     case EM_ENVMAP: {
+        if (g.env_rows && e.tex_id >= 0) {
+            // luminance-CDF importance sampling over the lat-long texture
+            const int w = g.env_w, h = g.env_h;
+            int row = cdf_find(g.env_rows, h, sampler.next1f());
+            const float* crow = g.env_cols + (size_t)row * w;
+            int col = cdf_find(crow, w, sampler.next1f());
+            float pr = g.env_rows[row] - (row ? g.env_rows[row - 1] : 0.f);
+            float pc = crow[col] - (col ? crow[col - 1] : 0.f);
+            float uu = (col + sampler.next1f()) / w;
+            float vv = (row + sampler.next1f()) / h;
+            float theta = vv * PI, phi = (uu - 0.5f) * 2.f * PI;
+            float st = sinf(theta);
+            Vec3 dl(st * sinf(phi), cosf(theta), -st * cosf(phi));
+            // invert the envmap rotation applied in envmap_eval
+            Vec3 dir = dl;
+            float az = e.aux.x, ze = e.aux.y;
+            if (az != 0.f || ze != 0.f) {
+                Quat qi = Quat::angle_axis(ze, Vec3(1.f, 0.f, 0.f)) *
+                          Quat::angle_axis(az, Vec3(0.f, 1.f, 0.f));
+                dir = qi.rotate(dl);
+            }
+            r.pos = sp_pos + dir * ENVMAP_DIST;
+            r.normal = -dir;
+            r.radiance = envmap_eval(e, dir, g.textures);
+            r.pdf = pr * pc * (float)w * (float)h /
+                    (2.f * PI * PI * fmaxf(st, 1e-5f));
+            r.delta = false;
+            return r;
+        }
         // cosine-hemisphere NEE around the shading normal (emitter.cu:25-73)
         float pdf;
         Vec3 nn = sp_n;
@@ -163,13 +209,36 @@ HD EmitterSampleRec emitter_sample(const EmitterParams& e, const EmitterGeom& g,
 
 // Solid-angle pdf of hitting this (area/env) emitter with a BSDF ray, for MIS.
 HD float emitter_pdf_hit(const EmitterParams& e, const Vec3& dir, float dist,
-                         const Vec3& light_n, const Vec3& sp_n) {
+                         const Vec3& light_n, const Vec3& sp_n,
+                         const EmitterGeom& g) {
     if (e.type == EM_AREA || e.type == EM_AREA_SPOT) {
         float cos_l = light_n.dot(-dir);
         if (cos_l <= 1e-6f) return 0.f;
         return e.inv_area * dist * dist / cos_l;
     }
     if (e.type == EM_ENVMAP) {
+        if (g.env_rows && e.tex_id >= 0) {
+            // must mirror the tabulated NEE pdf exactly (MIS)
+            Vec3 d = dir;
+            float az = e.aux.x, ze = e.aux.y;
+            if (az != 0.f || ze != 0.f) {
+                Quat q = Quat::angle_axis(-az, Vec3(0.f, 1.f, 0.f)) *
+                         Quat::angle_axis(-ze, Vec3(1.f, 0.f, 0.f));
+                d = q.rotate(d);
+            }
+            Vec2 uv = dir_to_latlong(d);
+            const int w = g.env_w, h = g.env_h;
+            int col = (int)(uv.x * w);
+            int row = (int)(uv.y * h);
+            col = col < 0 ? 0 : (col >= w ? w - 1 : col);
+            row = row < 0 ? 0 : (row >= h ? h - 1 : row);
+            const float* crow = g.env_cols + (size_t)row * w;
+            float pr = g.env_rows[row] - (row ? g.env_rows[row - 1] : 0.f);
+            float pc = crow[col] - (col ? crow[col - 1] : 0.f);
+            float st = sinf(uv.y * PI);
+            return pr * pc * (float)w * (float)h /
+                   (2.f * PI * PI * fmaxf(st, 1e-5f));
+        }
         float c = sp_n.dot(dir);
         return c > 0.f ? c * INV_PI : 0.f;
     }

@@ -80,7 +80,7 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
                 Vec3 le = envmap_eval(env, ray.d, sv.textures);
                 float w = 1.f;
                 if (!prev_delta) {
-                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n) * sel_pdf;
+                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n, sv.emitter_geom()) * sel_pdf;
                     w = mis_weight(prev_pdf, light_pdf);
                 }
                 if (tof_in_range(sv.md, path_time + ENVMAP_DIST)) L += thp * le * w;
@@ -105,7 +105,7 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
             if (!le.is_zero()) {
                 float w = 1.f;
                 if (!prev_delta) {
-                    float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n) * sel_pdf;
+                    float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n, sv.emitter_geom()) * sel_pdf;
                     w = mis_weight(prev_pdf, light_pdf);
                 }
                 if (tof_in_range(sv.md, path_time)) L += thp * le * w;

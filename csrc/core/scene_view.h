@@ -53,6 +53,8 @@ struct SceneView {
     const EmitterParams* emitters; int n_emitters;
     const int* emitter_prims;
     const float* emitter_cdf;
+    // envmap luminance-CDF tables (importance-sampled NEE; null = cosine)
+    const float* env_rows; const float* env_cols; int env_w, env_h;
     const TexView* textures; int n_textures;
     int env_emitter;   // emitter index of the EM_ENVMAP, or -1
     // media
@@ -64,7 +66,8 @@ struct SceneView {
     MaxDepthParams md;
 
     HD EmitterGeom emitter_geom() const {
-        return {prims, attrs, prim_obj, emitter_prims, emitter_cdf, textures};
+        return {prims, attrs, prim_obj, emitter_prims, emitter_cdf, textures,
+                env_rows, env_cols, env_w, env_h};
     }
     HD uint32_t obj_of_prim(int pid) const { return prim_obj[pid] & PRIM_OBJ_MASK; }
     HD bool prim_is_sphere(int pid) const { return (prim_obj[pid] & PRIM_SPHERE_BIT) != 0; }

@@ -204,7 +204,7 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
         if (!le.is_zero()) {
             float w = 1.f;
             if (!prev_delta) {
-                float light_pdf = emitter_pdf_hit(em, ray.d, h4.x, it.shading_n, prev_n) * sel_pdf;
+                float light_pdf = emitter_pdf_hit(em, ray.d, h4.x, it.shading_n, prev_n, sv.emitter_geom()) * sel_pdf;
                 w = mis_weight(prev_pdf, light_pdf);
             }
             L += thp * le * w;
@@ -323,7 +323,8 @@ __device__ inline void wf_trace_finish(const SceneView& sv, WfState& st, int i,
             if (!prev_delta) {
                 float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
                 float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d,
-                                                  Vec3(pn4.x, pn4.y, pn4.z)) * sel_pdf;
+                                                  Vec3(pn4.x, pn4.y, pn4.z),
+                                                  sv.emitter_geom()) * sel_pdf;
                 w = mis_weight(prev_pdf, light_pdf);
             }
             float4 l4 = st.L[i];

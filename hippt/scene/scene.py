@@ -9,6 +9,7 @@ native extension owns device copies and the BVH builder reorders primitives.
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Tuple
 
@@ -369,6 +370,27 @@ class Scene:
                                     e.tex_id, prim_base, prim_cnt, inv_area)
         self.native.set_emitter_prims(np.asarray(eprims, np.int32),
                                       np.asarray(ecdf, np.float32))
+
+        # envmap importance-sampling tables (luminance x sin(theta) CDFs over
+        # the lat-long texture; beyond-reference — cosine NEE is the fallback)
+        for e in d.emitters:
+            if os.environ.get("HIPPT_ENV_IS") == "0":
+                break   # A/B hook: cosine-hemisphere NEE (reference behavior)
+            if e.type == "envmap" and e.tex_id is not None and e.tex_id >= 0 \
+                    and e.tex_id < len(d.textures):
+                img = d.textures[e.tex_id]
+                lum = img[..., :3].mean(axis=2).astype(np.float64)
+                h, w = lum.shape
+                sin_t = np.sin((np.arange(h) + 0.5) * np.pi / h)[:, None]
+                wgt = np.maximum(lum * sin_t, 1e-12)
+                row_sum = wgt.sum(axis=1)
+                rows = np.cumsum(row_sum)
+                rows /= rows[-1]
+                cols = np.cumsum(wgt, axis=1)
+                cols /= cols[:, -1:]
+                self.native.set_env_cdf(rows.astype(np.float32),
+                                        np.ascontiguousarray(cols, np.float32))
+                break
 
         # ---- camera + depth caps
         self._set_camera_native()

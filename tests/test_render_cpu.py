@@ -175,3 +175,55 @@ def test_adaptive_sampling():
     # correlation between allocated spp and (pre-allocation) noise is
     # implicitly tested by the budget math; sanity: no pixel starved
     assert cnt.min() >= 8  # the uniform first batch
+
+
+def test_envmap_importance_sampling():
+    """Textured-envmap NEE uses the luminance-CDF sampler; a constant
+    (all-ones) texture must keep the white furnace white (unbiased, pdf and
+    MIS weights consistent), and a sun texture must cut variance vs the
+    cosine fallback at equal spp."""
+    from hippt.scene.scene import (SceneDesc, ObjectDesc, BsdfDesc, EmitterDesc,
+                                   CameraDesc, RenderConfig)
+    import os
+
+    def sphere_under_env(tex):
+        d = SceneDesc()
+        d.textures = [tex]
+        d.bsdfs = [BsdfDesc(type="lambertian", kd=(0.7, 0.7, 0.7))]
+        d.emitters = [EmitterDesc(type="envmap", emission=(1, 1, 1), scale=1.0,
+                                  tex_id=0)]
+        d.objects = [ObjectDesc(spheres=np.array([[0, 0, 0, 1.0]], np.float32),
+                                bsdf=0)]
+        d.camera = CameraDesc(pos=(0, 0, -4), lookat=(0, 0, 0), fov=35,
+                              width=48, height=48)
+        d.config = RenderConfig(renderer="pt", spp=1, max_depth=8)
+        return d
+
+    ones = np.ones((32, 64, 4), np.float32)
+    img = hippt.PythonRenderer(sphere_under_env(ones), device_id=-1).render(spp=64).numpy()
+    center = img[16:32, 16:32, :3]
+    # white furnace: multi-bounce lambertian ball under unit env -> ~1.0...
+    # with max_depth 8 and albedo 0.7 the Neumann series is close to 1/(1-a)
+    # scaled... just require energy conservation bracket around the analytic
+    # single-sphere value: mean in (0.65, 1.0)
+    assert 0.6 < center.mean() < 1.05, center.mean()
+
+    # sun texture: variance comparison IS vs cosine fallback
+    sun = np.full((32, 64, 4), 0.05, np.float32)
+    sun[6:9, 14:18, :3] = 40.0
+    def var_of(env_is):
+        if not env_is:
+            os.environ["HIPPT_ENV_IS"] = "0"
+        try:
+            r = hippt.PythonRenderer(sphere_under_env(sun), device_id=-1)
+            r.render(spp=24)
+            return float(np.mean(r.renderer.variance())), \
+                float(r.renderer.raw()[..., :3].mean())
+        finally:
+            os.environ.pop("HIPPT_ENV_IS", None)
+    v_is, m_is = var_of(True)
+    v_cos, m_cos = var_of(False)
+    # unbiased: means agree within noise
+    assert abs(m_is - m_cos) < 0.15 * max(m_cos, 1e-9), (m_is, m_cos)
+    # and the CDF sampler is dramatically less noisy
+    assert v_is < 0.5 * v_cos, (v_is, v_cos)
