@@ -58,6 +58,7 @@ struct SceneHolder {
     std::vector<float> emitter_cdf;
     std::vector<float> env_rows_h, env_cols_h;
     int env_w = 0, env_h = 0;
+    std::vector<float> emitter_sel_h;
     Camera cam{};
     MaxDepthParams md{};
     int env_emitter = -1;
@@ -160,6 +161,10 @@ struct SceneHolder {
     void set_emitter_prims(iarr eprims, farr ecdf) {
         emitter_prims.assign(eprims.data(), eprims.data() + eprims.size());
         emitter_cdf.assign(ecdf.data(), ecdf.data() + ecdf.size());
+    }
+
+    void set_emitter_sel(farr cdf) {
+        emitter_sel_h.assign(cdf.data(), cdf.data() + cdf.size());
     }
 
     void set_env_cdf(farr rows, farr cols) {
@@ -299,6 +304,7 @@ struct SceneHolder {
         host_sv.emitter_cdf = emitter_cdf.data();
         host_sv.env_rows = env_rows_h.empty() ? nullptr : env_rows_h.data();
         host_sv.env_cols = env_cols_h.empty() ? nullptr : env_cols_h.data();
+        host_sv.emitter_sel_cdf = emitter_sel_h.empty() ? nullptr : emitter_sel_h.data();
         host_sv.textures = tex_host.data();
         host_sv.media = media.data();
         host_sv.phases = phases.data();
@@ -333,6 +339,7 @@ struct SceneHolder {
         dev_sv.emitter_cdf = upload_vec(emitter_cdf.data(), emitter_cdf.size());
         dev_sv.env_rows = upload_vec(env_rows_h.data(), env_rows_h.size());
         dev_sv.env_cols = upload_vec(env_cols_h.data(), env_cols_h.size());
+        dev_sv.emitter_sel_cdf = upload_vec(emitter_sel_h.data(), emitter_sel_h.size());
         // textures: RGBA32F rows in device global memory (gfx950 has no
         // device texture units — software bilinear is the CDNA-native path)
         tex_dev.clear();
@@ -568,6 +575,7 @@ PYBIND11_MODULE(_C, m) {
         .def("update_emitter", &SceneHolder::update_emitter)
         .def("set_emitter_prims", &SceneHolder::set_emitter_prims)
         .def("set_env_cdf", &SceneHolder::set_env_cdf)
+        .def("set_emitter_sel", &SceneHolder::set_emitter_sel)
         .def("add_phase", &SceneHolder::add_phase)
         .def("add_medium", &SceneHolder::add_medium)
         .def("update_medium", &SceneHolder::update_medium)

@@ -20,15 +20,6 @@ HD bool tof_in_range(const MaxDepthParams& md, float time) {
     return !md.use_tof || (time >= md.min_time && time <= md.max_time);
 }
 
-// Pick an emitter uniformly; returns index and sets pdf.
-HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
-    if (sv.n_emitters <= 0) { pdf = 0.f; return -1; }
-    int i = (int)(sp.next1f() * sv.n_emitters);
-    i = i >= sv.n_emitters ? sv.n_emitters - 1 : i;
-    pdf = 1.f / sv.n_emitters;
-    return i;
-}
-
 // Resumable per-path state: one `path_step` = one bounce of the
 // unidirectional path loop.  Shared by trace_path (per-sample form) and the
 // megakernel's path-regeneration loop (k_render starts the next sample the
@@ -70,7 +61,6 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
     float& path_time = ps.path_time;
     PathStats& st = ps.st;
     int& b = ps.b;
-    const float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
     {
         HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
         if (hit.prim_idx < 0) {
@@ -80,7 +70,8 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
                 Vec3 le = envmap_eval(env, ray.d, sv.textures);
                 float w = 1.f;
                 if (!prev_delta) {
-                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n, sv.emitter_geom()) * sel_pdf;
+                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n, sv.emitter_geom()) *
+                                      emitter_sel_pdf(sv, sv.env_emitter);
                     w = mis_weight(prev_pdf, light_pdf);
                 }
                 if (tof_in_range(sv.md, path_time + ENVMAP_DIST)) L += thp * le * w;
@@ -105,7 +96,8 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
             if (!le.is_zero()) {
                 float w = 1.f;
                 if (!prev_delta) {
-                    float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n, sv.emitter_geom()) * sel_pdf;
+                    float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n, sv.emitter_geom()) *
+                                   emitter_sel_pdf(sv, obj.emitter_id);
                     w = mis_weight(prev_pdf, light_pdf);
                 }
                 if (tof_in_range(sv.md, path_time)) L += thp * le * w;

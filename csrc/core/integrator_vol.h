@@ -76,7 +76,6 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp, TravCtx
     PathStats st;
     VolStack stack;
     if (sv.cam_medium >= 0) stack.push(sv.cam_medium);
-    const float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
 
     int b = 0;
     for (int iter = 0; iter < sv.md.max_depth * 3 + 16 && b < sv.md.max_depth; ++iter) {
@@ -160,7 +159,8 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp, TravCtx
                 if (!le.is_zero()) {
                     float w = 1.f;
                     if (!prev_delta) {
-                        float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n, sv.emitter_geom()) * sel_pdf;
+                        float light_pdf = emitter_pdf_hit(em, ray.d, hit.t, it.shading_n, prev_n, sv.emitter_geom()) *
+                                        emitter_sel_pdf(sv, obj.emitter_id);
                         w = mis_weight(prev_pdf, light_pdf);
                     }
                     if (tof_in_range(sv.md, path_time)) L += thp * le * w;
@@ -212,7 +212,8 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp, TravCtx
                 Vec3 le = envmap_eval(env, ray.d, sv.textures);
                 float w = 1.f;
                 if (!prev_delta) {
-                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n, sv.emitter_geom()) * sel_pdf;
+                    float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d, prev_n, sv.emitter_geom()) *
+                                    emitter_sel_pdf(sv, sv.env_emitter);
                     w = mis_weight(prev_pdf, light_pdf);
                 }
                 if (tof_in_range(sv.md, path_time + ENVMAP_DIST)) L += thp * le * w;

@@ -55,6 +55,8 @@ struct SceneView {
     const float* emitter_cdf;
     // envmap luminance-CDF tables (importance-sampled NEE; null = cosine)
     const float* env_rows; const float* env_cols; int env_w, env_h;
+    // power-proportional light selection CDF (null = uniform, reference)
+    const float* emitter_sel_cdf;
     const TexView* textures; int n_textures;
     int env_emitter;   // emitter index of the EM_ENVMAP, or -1
     // media
@@ -72,6 +74,28 @@ struct SceneView {
     HD uint32_t obj_of_prim(int pid) const { return prim_obj[pid] & PRIM_OBJ_MASK; }
     HD bool prim_is_sphere(int pid) const { return (prim_obj[pid] & PRIM_SPHERE_BIT) != 0; }
 };
+
+// Light selection: power-proportional when the CDF is present (extension;
+// the reference picks uniformly), uniform otherwise.  Any positive weights
+// keep the estimator unbiased because the pick pdf divides the sample.
+HD float emitter_sel_pdf(const SceneView& sv, int i) {
+    if (sv.n_emitters <= 0) return 0.f;
+    if (!sv.emitter_sel_cdf) return 1.f / sv.n_emitters;
+    return sv.emitter_sel_cdf[i] - (i ? sv.emitter_sel_cdf[i - 1] : 0.f);
+}
+
+HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
+    if (sv.n_emitters <= 0) { pdf = 0.f; return -1; }
+    if (sv.emitter_sel_cdf) {
+        int i = cdf_find(sv.emitter_sel_cdf, sv.n_emitters, sp.next1f());
+        pdf = emitter_sel_pdf(sv, i);
+        return i;
+    }
+    int i = (int)(sp.next1f() * sv.n_emitters);
+    i = i >= sv.n_emitters ? sv.n_emitters - 1 : i;
+    pdf = 1.f / sv.n_emitters;
+    return i;
+}
 
 // Traversal entry points used by every integrator/kernel.  The 4-wide
 // ordered walk is the ONLY device path — keeping the binary walk as a

@@ -185,7 +185,6 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
     Vec3 prev_n(pn4.x, pn4.y, pn4.z);
     uint32_t counts = float_as_uint(pn4.w);
     Sampler sp(st.rng[i]);
-    const float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
 
     Vec3 pos = ray.at(h4.x);
     uint32_t po = sv.prim_obj[prim_idx];
@@ -204,7 +203,8 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
         if (!le.is_zero()) {
             float w = 1.f;
             if (!prev_delta) {
-                float light_pdf = emitter_pdf_hit(em, ray.d, h4.x, it.shading_n, prev_n, sv.emitter_geom()) * sel_pdf;
+                float light_pdf = emitter_pdf_hit(em, ray.d, h4.x, it.shading_n, prev_n, sv.emitter_geom()) *
+                                emitter_sel_pdf(sv, obj.emitter_id);
                 w = mis_weight(prev_pdf, light_pdf);
             }
             L += thp * le * w;
@@ -321,10 +321,10 @@ __device__ inline void wf_trace_finish(const SceneView& sv, WfState& st, int i,
             Vec3 le = envmap_eval(env, ray.d, sv.textures);
             float w = 1.f;
             if (!prev_delta) {
-                float sel_pdf = sv.n_emitters > 0 ? 1.f / sv.n_emitters : 0.f;
                 float light_pdf = emitter_pdf_hit(env, ray.d, ENVMAP_DIST, ray.d,
                                                   Vec3(pn4.x, pn4.y, pn4.z),
-                                                  sv.emitter_geom()) * sel_pdf;
+                                                  sv.emitter_geom()) *
+                                  emitter_sel_pdf(sv, sv.env_emitter);
                 w = mis_weight(prev_pdf, light_pdf);
             }
             float4 l4 = st.L[i];

@@ -340,6 +340,7 @@ class Scene:
         # ---- emitters (+ per-emitter area CDF over reordered prims)
         eprims: List[int] = []
         ecdf: List[float] = []
+        emitter_areas = {}
         for ei, e in enumerate(d.emitters):
             etype = {"point": EM_POINT, "area": EM_AREA, "area-spot": EM_AREA_SPOT,
                      "envmap": EM_ENVMAP}[e.type]
@@ -360,6 +361,7 @@ class Scene:
                         old_ids = np.arange(b0, b0 + n)
                         eprims.extend(new_of_old[old_ids].tolist())
                         tot = float(areas.sum())
+                        emitter_areas[ei] = tot
                         inv_area = 1.0 / tot if tot > 0 else 0.0
                         cdf = np.cumsum(areas) / max(tot, 1e-30)
                         ecdf.extend(cdf.tolist())
@@ -370,6 +372,31 @@ class Scene:
                                     e.tex_id, prim_base, prim_cnt, inv_area)
         self.native.set_emitter_prims(np.asarray(eprims, np.int32),
                                       np.asarray(ecdf, np.float32))
+
+        # power-proportional light selection (extension; HIPPT_LIGHT_POWER=0
+        # restores the reference's uniform pick).  Any positive weights are
+        # unbiased; these are proportional to emitted power.
+        if len(d.emitters) > 1 and os.environ.get("HIPPT_LIGHT_POWER") != "0":
+            powers = []
+            for ei, e in enumerate(d.emitters):
+                mean_e = float(np.mean(e.emission)) * float(e.scale)
+                if e.type == "point":
+                    p_ = mean_e * 4.0 * math.pi
+                elif e.type == "envmap":
+                    lum = 1.0
+                    if e.tex_id is not None and 0 <= e.tex_id < len(d.textures):
+                        lum = float(d.textures[e.tex_id][..., :3].mean())
+                    p_ = mean_e * lum * 4.0 * math.pi
+                else:
+                    area = emitter_areas.get(ei, 1.0)
+                    p_ = mean_e * area * math.pi
+                    if e.type == "area-spot":
+                        p_ *= max(1.0 - e.cos_max, 1e-3)
+                powers.append(max(p_, 1e-9))
+            w = np.asarray(powers, np.float64)
+            w = np.maximum(w, 0.05 * w.mean())   # never starve a light
+            cdf = np.cumsum(w) / w.sum()
+            self.native.set_emitter_sel(cdf.astype(np.float32))
 
         # envmap importance-sampling tables (luminance x sin(theta) CDFs over
         # the lat-long texture; beyond-reference — cosine NEE is the fallback)
