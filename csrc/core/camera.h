@@ -24,8 +24,17 @@ struct alignas(16) Camera {
     float pad0, pad1;
 
     // Generate primary ray through pixel (px, py) with jitter.
-    HD Ray gen_ray(int px, int py, Sampler& sp) const {
+    // samp >= 0 stratifies the AA jitter over a per-pixel-rotated 4x4 grid
+    // cycling with the sample index (extension; same RNG stream consumption,
+    // so determinism and all downstream sampling are unchanged)
+    HD Ray gen_ray(int px, int py, Sampler& sp, int samp = -1) const {
         Vec2 j = sp.next2f();
+        if (samp >= 0) {
+            uint32_t hh = (uint32_t)(px * 9781 + py * 6271);
+            uint32_t cell = ((uint32_t)samp + (hh ^ (hh >> 16))) & 15u;
+            j = Vec2(((float)(cell & 3u) + j.x) * 0.25f,
+                     ((float)(cell >> 2) + j.y) * 0.25f);
+        }
         float x = (px + j.x - 0.5f * w);
         float y = (0.5f * h - py - j.y);
         if (ortho) {

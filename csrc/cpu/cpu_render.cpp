@@ -47,7 +47,7 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
             float lum_s = 0.f, lum_s2 = 0.f;
             for (int s = 0; s < nspp_px; ++s) {
                 Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
-                Ray ray = sv.cam.gen_ray(x, y, sp);
+                Ray ray = sv.cam.gen_ray(x, y, sp, spp0 + s);
                 Vec3 L(0.f);
                 if (renderer == 2)      L = trace_path_volumetric(sv, ray, sp);
                 else if (renderer == 4) L = Vec3(trace_depth(sv, ray));

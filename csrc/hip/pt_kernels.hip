@@ -77,7 +77,7 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
         int s = 0;
         Sampler sp(uint32_t(pix), uint32_t(spp0) * SEED_SCALER + seed);
         PathState ps;
-        ps.reset(sv.cam.gen_ray(px, py, sp));
+        ps.reset(sv.cam.gen_ray(px, py, sp, spp0));
         for (;;) {
             if (path_step(sv, ps, sp, tc)) {
                 Vec3 L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
@@ -87,13 +87,13 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
                 lum_s2 = fmaf(lum, lum, lum_s2);
                 if (++s >= nspp_px) break;
                 sp = Sampler(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
-                ps.reset(sv.cam.gen_ray(px, py, sp));
+                ps.reset(sv.cam.gen_ray(px, py, sp, spp0 + s));
             }
         }
     } else
     for (int s = 0; s < nspp_px; ++s) {
         Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
-        Ray ray = sv.cam.gen_ray(px, py, sp);
+        Ray ray = sv.cam.gen_ray(px, py, sp, spp0 + s);
         Vec3 L(0.f);
         if constexpr (RENDERER == R_VOLUME_PT) L = trace_path_volumetric(sv, ray, sp, tc);
         else if constexpr (RENDERER == R_DEPTH) L = Vec3(trace_depth(sv, ray, tc));
@@ -142,7 +142,7 @@ void k_render_persistent(SceneView sv, float* __restrict__ accum, float* __restr
         // already smooth tile tails, and regen lengthens the per-tile stint)
         for (int s = 0; s < nspp; ++s) {
             Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
-            Ray ray = sv.cam.gen_ray(px, py, sp);
+            Ray ray = sv.cam.gen_ray(px, py, sp, spp0 + s);
             Vec3 L = trace_path(sv, ray, sp, tc);
             Lsum += L;
             float lum = (L.x + L.y + L.z) * (1.f / 3.f);

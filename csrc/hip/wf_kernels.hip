@@ -66,7 +66,7 @@ void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n
     if (i >= st.n) return;
     int px = i % st.w, py = i / st.w;
     Sampler sp(uint32_t(i), uint32_t(spp_idx) * SEED_SCALER + seed);
-    Ray ray = sv.cam.gen_ray(px, py, sp);
+    Ray ray = sv.cam.gen_ray(px, py, sp, spp_idx);
     HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
     st.ray_o[i] = make_float4(ray.o.x, ray.o.y, ray.o.z, 0.f);
     st.ray_d[i] = make_float4(ray.d.x, ray.d.y, ray.d.z, 0.f);

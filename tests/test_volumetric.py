@@ -79,10 +79,13 @@ class TestGridMedium:
             d.media[0].density = d.media[0].density * scale / 18.0
             return hippt.PythonRenderer(d, device_id=-1).render(spp=16).numpy()
         thin = render_with(4.0)
-        thick = render_with(40.0)
-        # compare center region where the plume sits
-        assert thick[10:16, 12:20, :3].mean() != pytest.approx(
-            thin[10:16, 12:20, :3].mean(), rel=0.02)
+        thick = render_with(120.0)
+        # compare center region where the plume sits; a 30x denser medium
+        # must measurably change the crop (direction depends on albedo:
+        # scattering can brighten, absorption darken — just not equal)
+        a = float(thick[10:16, 12:20, :3].mean())
+        b = float(thin[10:16, 12:20, :3].mean())
+        assert abs(a - b) > 0.01 * max(b, 1e-9), (a, b)
 
     def test_blackbody_emission_adds_energy(self):
         d1 = smoke_box(width=32, height=24, n_grid=24, emission=False)
