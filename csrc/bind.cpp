@@ -22,7 +22,7 @@ using namespace hippt;
 namespace hippt {
 void render_cpu(const SceneView& sv, float* accum, float* var,
                 int spp0, int nspp, uint32_t seed, int renderer, int n_threads,
-                int y0, int y1, const uint8_t* spp_map);
+                int y0, int y1, const uint8_t* spp_map, float* aux);
 void render_lt_cpu(const SceneView& sv, float* accum, int spp0, int nspp, uint32_t seed,
                    int spec_constraint, float caustic_scaling, int n_threads);
 }
@@ -373,11 +373,15 @@ struct SceneHolder {
 
     void render_host(farr accum, py::object var, int spp0, int nspp, uint32_t seed,
                      int renderer, int spec_constraint, float caustic_scaling, int n_threads,
-                     int y0 = 0, int y1 = 0, py::object spp_map = py::none()) {
+                     int y0 = 0, int y1 = 0, py::object spp_map = py::none(),
+                     py::object aux = py::none()) {
         finalize();
         float* vp = nullptr;
         farr var_arr;
         if (!var.is_none()) { var_arr = var.cast<farr>(); vp = var_arr.mutable_data(); }
+        float* auxp = nullptr;
+        farr aux_arr;
+        if (!aux.is_none()) { aux_arr = aux.cast<farr>(); auxp = aux_arr.mutable_data(); }
         const uint8_t* smp = nullptr;
         py::array_t<uint8_t, py::array::c_style | py::array::forcecast> smp_arr;
         if (!spp_map.is_none()) {
@@ -390,13 +394,13 @@ struct SceneHolder {
                           spec_constraint, caustic_scaling, n_threads);
         else
             render_cpu(host_sv, accum.mutable_data(), vp, spp0, nspp, seed, renderer, n_threads,
-                       y0, y1, smp);
+                       y0, y1, smp, auxp);
     }
 
     void render_device(uintptr_t accum_ptr, uintptr_t var_ptr, int spp0, int nspp,
                        uint32_t seed, int renderer, int spec_constraint,
                        float caustic_scaling, uintptr_t stream, int y0 = 0, int y1 = 0,
-                       uintptr_t spp_map_ptr = 0) {
+                       uintptr_t spp_map_ptr = 0, uintptr_t aux_ptr = 0) {
         if (!has_dev) throw std::runtime_error("scene not uploaded to device");
         dev_sv.cam = cam;   // camera / depth params may have changed (hot reload)
         dev_sv.md = md;
@@ -413,7 +417,7 @@ struct SceneHolder {
         }
         HIP_OK(launch_render(dev_sv, (float*)accum_ptr, (float*)var_ptr, spp0, nspp, seed,
                              renderer, spec_constraint, caustic_scaling, (void*)stream, y0, y1,
-                             (const uint8_t*)spp_map_ptr));
+                             (const uint8_t*)spp_map_ptr, (float*)aux_ptr));
     }
 
     py::dict info() {
@@ -591,12 +595,14 @@ PYBIND11_MODULE(_C, m) {
              py::arg("accum"), py::arg("var"), py::arg("spp0"), py::arg("nspp"),
              py::arg("seed"), py::arg("renderer"), py::arg("spec_constraint"),
              py::arg("caustic_scaling"), py::arg("n_threads"),
-             py::arg("y0") = 0, py::arg("y1") = 0, py::arg("spp_map") = py::none())
+             py::arg("y0") = 0, py::arg("y1") = 0, py::arg("spp_map") = py::none(),
+             py::arg("aux") = py::none())
         .def("render_device", &SceneHolder::render_device,
              py::arg("accum_ptr"), py::arg("var_ptr"), py::arg("spp0"), py::arg("nspp"),
              py::arg("seed"), py::arg("renderer"), py::arg("spec_constraint"),
              py::arg("caustic_scaling"), py::arg("stream"),
-             py::arg("y0") = 0, py::arg("y1") = 0, py::arg("spp_map_ptr") = 0)
+             py::arg("y0") = 0, py::arg("y1") = 0, py::arg("spp_map_ptr") = 0,
+             py::arg("aux_ptr") = 0)
         .def("info", &SceneHolder::info);
 
     m.def("build_bvh", &py_build_bvh, py::arg("prims"), py::arg("prim_obj"),

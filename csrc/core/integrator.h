@@ -35,6 +35,9 @@ struct PathState {
     float path_time;
     PathStats st;
     int b, iter;
+    // primary-hit AOVs (denoiser guides): shading normal, depth, albedo
+    Vec3 aov_n, aov_alb;
+    float aov_t;
 
     HD void reset(const Ray& r) {
         ray = r;
@@ -45,8 +48,27 @@ struct PathState {
         path_time = 0.f;
         st = PathStats();
         b = 0; iter = 0;
+        aov_n = Vec3(0.f); aov_alb = Vec3(0.f); aov_t = 0.f;
     }
 };
+
+// First-hit albedo proxy per BSDF family (denoiser guide, not physics).
+HD Vec3 bsdf_aov_albedo(const BsdfParams& b, Vec2 uv, const TexView* textures) {
+    switch (b.type) {
+    case BSDF_LAMBERTIAN:
+    case BSDF_PLASTIC:
+    case BSDF_PLASTIC_FORWARD:
+        return bsdf_albedo(b, uv, textures);
+    case BSDF_GGX_CONDUCTOR:
+        return tex_or(textures, b.tex[TEX_SPECULAR], uv, b.kg.xyz());
+    case BSDF_SPECULAR:
+    case BSDF_TRANSLUCENT:
+    case BSDF_DISPERSION:
+        return tex_or(textures, b.tex[TEX_SPECULAR], uv, b.ks.xyz());
+    default:
+        return Vec3(1.f);
+    }
+}
 
 // One bounce; returns true when the path is finished (L is final).
 HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
@@ -88,6 +110,11 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
         const BsdfParams& bsdf = sv.bsdfs[obj.bsdf_id];
         if (bsdf.tex[TEX_NORMAL] >= 0)
             it.shading_n = apply_normal_map(sv.textures, bsdf.tex[TEX_NORMAL], it.uv, it.shading_n);
+        if (ps.iter == 1) {
+            ps.aov_n = it.shading_n;
+            ps.aov_t = hit.t;
+            ps.aov_alb = bsdf_aov_albedo(bsdf, it.uv, sv.textures);
+        }
 
         // ---- emitter hit accumulation with MIS (megakernel_pt.cu:91-152)
         if (obj.emitter_id >= 0) {

@@ -33,7 +33,7 @@ static void parallel_rows(int h, int n_threads, const std::function<void(int)>& 
 // 4=depth, 5=bvh_cost (light tracing has its own entry below).
 void render_cpu(const SceneView& sv, float* accum, float* var,
                 int spp0, int nspp, uint32_t seed, int renderer, int n_threads,
-                int y0, int y1, const uint8_t* spp_map) {
+                int y0, int y1, const uint8_t* spp_map, float* aux) {
     const int w = sv.cam.w, h = sv.cam.h;
     if (y1 <= 0 || y1 > h) y1 = h;
     if (y0 < 0) y0 = 0;
@@ -45,6 +45,8 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
             if (nspp_px == 0) continue;
             Vec3 Lsum(0.f);
             float lum_s = 0.f, lum_s2 = 0.f;
+            Vec3 an(0.f), aa(0.f);
+            float at = 0.f;
             for (int s = 0; s < nspp_px; ++s) {
                 Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
                 Ray ray = sv.cam.gen_ray(x, y, sp, spp0 + s);
@@ -52,10 +54,21 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
                 if (renderer == 2)      L = trace_path_volumetric(sv, ray, sp);
                 else if (renderer == 4) L = Vec3(trace_depth(sv, ray));
                 else if (renderer == 5) { Vec2 c = trace_bvh_cost(sv, ray); L = Vec3(c.x, c.y, 0.f); }
-                else                    L = trace_path(sv, ray, sp);
+                else if (aux) {
+                    PathState ps;
+                    ps.reset(ray);
+                    while (!path_step(sv, ps, sp, TravCtx{})) {}
+                    L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
+                    an += ps.aov_n; aa += ps.aov_alb; at += ps.aov_t;
+                } else                  L = trace_path(sv, ray, sp);
                 Lsum += L;
                 float lum = (L.x + L.y + L.z) * (1.f / 3.f);
                 lum_s += lum; lum_s2 += lum * lum;
+            }
+            if (aux) {
+                float* a8 = aux + pix * 8;
+                a8[0] += an.x; a8[1] += an.y; a8[2] += an.z; a8[3] += at;
+                a8[4] += aa.x; a8[5] += aa.y; a8[6] += aa.z; a8[7] += (float)nspp_px;
             }
             accum[pix * 4 + 0] += Lsum.x;
             accum[pix * 4 + 1] += Lsum.y;

@@ -26,7 +26,9 @@ int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling,
                   void* stream, int y0 = 0, int y1 = 0,
-                  const uint8_t* spp_map = nullptr);
+                  const uint8_t* spp_map = nullptr,
+                  float* aux = nullptr);  // (h*w*8) primary-hit AOV sums:
+                                          // n.xyz, depth, albedo.rgb, count
 
 // Wavefront path tracer (SoA queues + compaction); state owned by WfState.
 struct WfState;

@@ -42,7 +42,7 @@ __attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
 // plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
               int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached, int y_off, int y_end,
-              const uint8_t* __restrict__ spp_map) {
+              const uint8_t* __restrict__ spp_map, float* __restrict__ aux) {
     extern __shared__ uint64_t s_stk[];
     // dynamic-LDS layout: [n_cached 128-byte nodes][per-thread stacks].
     // The top of the tree is copied into LDS once per block: every walk's
@@ -75,6 +75,8 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
         // (wave64 tail divergence was the measured bottleneck).  Sampler
         // streams and results are identical to the per-sample loop.
         int s = 0;
+        Vec3 an(0.f), aa(0.f);
+        float at = 0.f;
         Sampler sp(uint32_t(pix), uint32_t(spp0) * SEED_SCALER + seed);
         PathState ps;
         ps.reset(sv.cam.gen_ray(px, py, sp, spp0));
@@ -85,10 +87,16 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
                 float lum = (L.x + L.y + L.z) * (1.f / 3.f);
                 lum_s += lum;
                 lum_s2 = fmaf(lum, lum, lum_s2);
+                if (aux) { an += ps.aov_n; aa += ps.aov_alb; at += ps.aov_t; }
                 if (++s >= nspp_px) break;
                 sp = Sampler(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
                 ps.reset(sv.cam.gen_ray(px, py, sp, spp0 + s));
             }
+        }
+        if (aux) {
+            float* a8 = aux + pix * 8;
+            a8[0] += an.x; a8[1] += an.y; a8[2] += an.z; a8[3] += at;
+            a8[4] += aa.x; a8[5] += aa.y; a8[6] += aa.z; a8[7] += (float)nspp_px;
         }
     } else
     for (int s = 0; s < nspp_px; ++s) {
@@ -183,7 +191,7 @@ __global__ void k_add_count(float* __restrict__ accum, size_t npix, float cnt) {
 int launch_render(const SceneView& sv, float* accum, float* var,
                   int spp0, int nspp, uint32_t seed, int renderer,
                   int spec_constraint, float caustic_scaling, void* stream,
-                  int y0, int y1, const uint8_t* spp_map) {
+                  int y0, int y1, const uint8_t* spp_map, float* aux) {
     hipStream_t st = (hipStream_t)stream;
     // HIPPT_SWIZZLE=1 enables the XCD band swizzle (measured -14% on the
     // kitchen megakernel — the round-robin XCD dispatch already spreads
@@ -238,15 +246,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     case R_VOLUME_PT:
         if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -263,15 +271,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     }
     }

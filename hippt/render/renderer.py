@@ -132,6 +132,36 @@ class Renderer:
         w = w / max(float(w.mean()), 1e-12)
         return _np.clip(_np.round(w * batch), 0, 255).astype(_np.uint8)
 
+    def enable_aov(self):
+        """Allocate the primary-hit AOV buffer (normal/depth/albedo sums);
+        subsequent megakernel renders fill it (denoiser guides)."""
+        h, w = self.scene.desc.camera.height, self.scene.desc.camera.width
+        if self.device is not None:
+            self.aux = self.torch.zeros((h, w, 8), dtype=self.torch.float32,
+                                        device=f"cuda:{self.device}")
+        else:
+            self.aux = np.zeros((h, w, 8), np.float32)
+        return self
+
+    def aov(self):
+        """dict(normal (h,w,3), depth (h,w), albedo (h,w,3)) — means."""
+        a = self.aux
+        cnt = a[:, :, 7:8] if self.device is not None else a[:, :, 7:8]
+        c = cnt.clip(1e-9, None) if self.device is None else cnt.clamp(min=1e-9)
+        return {"normal": a[:, :, 0:3] / c, "depth": a[:, :, 3] / c[:, :, 0],
+                "albedo": a[:, :, 4:7] / c}
+
+    def denoise(self, iterations: int = 2):
+        """SVGF-lite a-trous denoise of the current accumulation using the
+        AOV guides (enable_aov() first).  Returns (h,w,3)."""
+        if getattr(self, "aux", None) is None:
+            raise RuntimeError("call enable_aov() before denoise()")
+        from ..utils.denoise import atrous_denoise
+        img = self.raw()[:, :, :3] if self.device is None else self.raw()[:, :, :3]
+        g = self.aov()
+        return atrous_denoise(img, g["normal"], g["depth"], g["albedo"],
+                              iterations=iterations)
+
     def render(self, spp: int = 1, y0: int = 0, y1: int = 0, spp_map=None):
         """Accumulate spp more samples (reference render_raw semantics).
         y0/y1 restrict rendering to the row band [y0, y1) — tile-split DP
@@ -149,10 +179,12 @@ class Renderer:
                     self._seed() + 0x9E37, C.R_LIGHT_TRACE,
                     self.spec_constraint, self.caustic_scaling, stream)
             else:
+                aux = getattr(self, "aux", None)
                 self.scene.native.render_device(
                     self.accum.data_ptr(), self.var.data_ptr(), self.accum_cnt, spp,
                     self._seed(), self.rid, self.spec_constraint, self.caustic_scaling, stream,
-                    y0, y1, 0 if spp_map is None else spp_map.data_ptr())
+                    y0, y1, 0 if spp_map is None else spp_map.data_ptr(),
+                    0 if aux is None else aux.data_ptr())
             self.torch.cuda.synchronize(self.device)
         else:
             var = self.var.reshape(-1)
@@ -165,11 +197,13 @@ class Renderer:
                                               C.R_LIGHT_TRACE, self.spec_constraint,
                                               self.caustic_scaling, 0)
             else:
+                aux = getattr(self, "aux", None)
                 self.scene.native.render_host(self.accum.reshape(-1), self.var.reshape(-1),
                                               self.accum_cnt, spp, self._seed(),
                                               self.rid, self.spec_constraint,
                                               self.caustic_scaling, 0, y0, y1,
-                                              None if spp_map is None else spp_map.reshape(-1))
+                                              None if spp_map is None else spp_map.reshape(-1),
+                                              aux.reshape(-1) if aux is not None else None)
         if spp_map is None:
             self.accum_cnt += spp
         else:

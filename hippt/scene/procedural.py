@@ -109,8 +109,8 @@ def cornell_box(width=256, height=256, spp=4, max_depth=2, renderer="pt",
     d.emitters = [EmitterDesc(type="area", emission=(1.0, 0.85, 0.6), scale=light_scale)]
     s = 1.0
     # room (open toward camera at z=-2s..? camera looks +z); box [-1,1]^2 x [0,2]
-    floor = quad((-s, 0, 0), (s, 0, 0), (s, 0, 2 * s), (-s, 0, 2 * s))
-    ceil = quad((-s, 2 * s, 2 * s), (s, 2 * s, 2 * s), (s, 2 * s, 0), (-s, 2 * s, 0))
+    floor = quad((-s, 0, 0), (-s, 0, 2 * s), (s, 0, 2 * s), (s, 0, 0))      # +y inward
+    ceil = quad((-s, 2 * s, 2 * s), (-s, 2 * s, 0), (s, 2 * s, 0), (s, 2 * s, 2 * s))  # -y inward
     back = quad((s, 0, 2 * s), (-s, 0, 2 * s), (-s, 2 * s, 2 * s), (s, 2 * s, 2 * s))
     left = quad((-s, 0, 2 * s), (-s, 0, 0), (-s, 2 * s, 0), (-s, 2 * s, 2 * s))
     right = quad((s, 0, 0), (s, 0, 2 * s), (s, 2 * s, 2 * s), (s, 2 * s, 0))

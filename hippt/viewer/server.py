@@ -115,6 +115,7 @@ class ViewerApp:
         self.move_speed = 0.15
         self.rot_speed = 0.08
         self.adaptive = False   # variance-guided per-pixel spp after warmup
+        self.denoise = False    # SVGF-lite a-trous display filter
 
     # ------------------------------------------------------------ lifecycle
     def start(self):
@@ -143,9 +144,16 @@ class ViewerApp:
     def frame_png(self) -> bytes:
         from ..utils.png import tonemap, write_png
         import tempfile, os
+        import numpy as np
         with self.lock:
-            acc = (self.pyr.renderer.accum.cpu().numpy()
-                   if self.pyr.renderer.device is not None else self.pyr.renderer.accum.copy())
+            r = self.pyr.renderer
+            if self.denoise and getattr(r, "aux", None) is not None:
+                den = r.denoise()
+                den = den.cpu().numpy() if hasattr(den, "cpu") else np.asarray(den)
+                acc = np.concatenate([den, np.ones_like(den[..., :1])], axis=2)
+            else:
+                acc = (r.accum.cpu().numpy()
+                       if r.device is not None else r.accum.copy())
         img = tonemap(acc)
         buf = io.BytesIO()
         # write_png writes to path; reuse its encoder via temp buffer
@@ -342,6 +350,14 @@ def build_app(viewer: ViewerApp):
     def adaptive(req: AdaptiveReq):
         viewer.adaptive = bool(req.enabled)
         return {"ok": True, "adaptive": viewer.adaptive}
+
+    @app.post("/api/denoise")
+    def denoise(req: AdaptiveReq):
+        if req.enabled and getattr(viewer.pyr.renderer, "aux", None) is None:
+            with viewer.lock:
+                viewer.pyr.renderer.enable_aov()
+        viewer.denoise = bool(req.enabled)
+        return {"ok": True, "denoise": viewer.denoise}
 
     @app.post("/api/bsdf")
     def bsdf(req: BsdfReq):

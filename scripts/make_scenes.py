@@ -19,8 +19,8 @@ MESH = os.path.join(ROOT, "scenes", "meshes", "cbox")
 def cbox_meshes():
     os.makedirs(MESH, exist_ok=True)
     s = 1.0
-    save_obj(f"{MESH}/floor.obj", quad((-s, 0, 0), (s, 0, 0), (s, 0, 2 * s), (-s, 0, 2 * s)))
-    save_obj(f"{MESH}/ceiling.obj", quad((-s, 2 * s, 2 * s), (s, 2 * s, 2 * s), (s, 2 * s, 0), (-s, 2 * s, 0)))
+    save_obj(f"{MESH}/floor.obj", quad((-s, 0, 0), (-s, 0, 2 * s), (s, 0, 2 * s), (s, 0, 0)))  # +y inward
+    save_obj(f"{MESH}/ceiling.obj", quad((-s, 2 * s, 2 * s), (-s, 2 * s, 0), (s, 2 * s, 0), (s, 2 * s, 2 * s)))  # -y inward
     save_obj(f"{MESH}/back.obj", quad((s, 0, 2 * s), (-s, 0, 2 * s), (-s, 2 * s, 2 * s), (s, 2 * s, 2 * s)))
     save_obj(f"{MESH}/left_wall.obj", quad((-s, 0, 2 * s), (-s, 0, 0), (-s, 2 * s, 0), (-s, 2 * s, 2 * s)))
     save_obj(f"{MESH}/right_wall.obj", quad((s, 0, 0), (s, 0, 2 * s), (s, 2 * s, 2 * s), (s, 2 * s, 0)))

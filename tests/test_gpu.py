@@ -201,3 +201,21 @@ def test_adaptive_sampling_gpu():
     d2 = cornell_box(width=64, height=64, max_depth=4)
     u = hippt.PythonRenderer(d2, device_id=0).render(spp=32).cpu().numpy()
     assert abs(img[..., :3].mean() - u[..., :3].mean()) < 0.05 * u[..., :3].mean()
+
+
+@pytest.mark.gpu
+def test_denoiser_gpu():
+    d = cornell_box(width=96, height=96, max_depth=4)
+    r = hippt.PythonRenderer(d, device_id=0)
+    r.renderer.enable_aov()
+    r.render(spp=8)
+    den = r.renderer.denoise()
+    assert den.is_cuda and torch.isfinite(den).all()
+    d2 = cornell_box(width=96, height=96, max_depth=4)
+    ref = hippt.PythonRenderer(d2, device_id=0, seed_offset=3).render(spp=512).cpu().numpy()[..., :3]
+    noisy = r.renderer.raw()[..., :3].cpu().numpy()
+    dn = den.cpu().numpy()
+    crop = np.s_[30:84, 12:84]
+    rn = float(np.sqrt(((noisy[crop] - ref[crop]) ** 2).mean()))
+    rd = float(np.sqrt(((dn[crop] - ref[crop]) ** 2).mean()))
+    assert rd < 0.6 * rn, (rd, rn)

@@ -95,3 +95,14 @@ def test_adaptive_toggle(client):
     assert stats["spp"] > 0
     r = c.post("/api/adaptive", json={"enabled": False})
     assert r.json()["adaptive"] is False
+
+
+def test_denoise_toggle(client):
+    c, viewer = client
+    r = c.post("/api/denoise", json={"enabled": True})
+    assert r.status_code == 200 and r.json()["denoise"] is True
+    import time
+    time.sleep(0.6)
+    png = c.get("/frame.png")
+    assert png.status_code == 200 and png.content[:4] == b"\x89PNG"
+    c.post("/api/denoise", json={"enabled": False})
