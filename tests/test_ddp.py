@@ -131,3 +131,19 @@ def test_ddp_checkpoint_resume(tmp_path):
     assert res.returncode == 0, res.stderr[-1500:]
     assert "resumed from" in res.stdout
     assert out2.exists()
+
+
+def test_serial_tof_sweep(tmp_path):
+    """Serial batch driver: per-window ToF frames with barriers (reference
+    serial_render.py job_tof_rendering), 2 ranks on gloo."""
+    res = run_torchrun(2, ["-m", "hippt.parallel.serial", "--cpu",
+                           "--job", "tof", "--scene", "cornell",
+                           "--frames", "2", "--spp", "4",
+                           "--width", "32", "--height", "32",
+                           "--time-start", "2.5", "--time-step", "0.6",
+                           "--time-window", "0.6",
+                           "--outdir", str(tmp_path / "tof")])
+    assert res.returncode == 0, res.stderr[-1500:]
+    import os
+    outs = sorted(os.listdir(tmp_path / "tof"))
+    assert len([f for f in outs if f.endswith(".png")]) == 2, outs
