@@ -151,7 +151,7 @@ class Renderer:
         return {"normal": a[:, :, 0:3] / c, "depth": a[:, :, 3] / c[:, :, 0],
                 "albedo": a[:, :, 4:7] / c}
 
-    def denoise(self, iterations: int = 2):
+    def denoise(self, iterations: int = 2, **kw):
         """SVGF-lite a-trous denoise of the current accumulation using the
         AOV guides (enable_aov() first).  Returns (h,w,3)."""
         if getattr(self, "aux", None) is None:
@@ -160,7 +160,7 @@ class Renderer:
         img = self.raw()[:, :, :3] if self.device is None else self.raw()[:, :, :3]
         g = self.aov()
         return atrous_denoise(img, g["normal"], g["depth"], g["albedo"],
-                              iterations=iterations)
+                              iterations=iterations, **kw)
 
     def render(self, spp: int = 1, y0: int = 0, y1: int = 0, spp_map=None):
         """Accumulate spp more samples (reference render_raw semantics).

@@ -6,7 +6,7 @@ from __future__ import annotations
 
 
 def atrous_denoise(color, normal, depth, albedo, iterations: int = 3,
-                   sigma_n: float = 32.0, sigma_z: float = 1.0,
+                   sigma_n: float = 16.0, sigma_z: float = 1.0,
                    sigma_l: float = 4.0):
     """color (h,w,3), normal (h,w,3), depth (h,w), albedo (h,w,3) — torch
     tensors (any device) or numpy arrays.  Returns same-type (h,w,3)."""
