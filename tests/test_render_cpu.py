@@ -244,8 +244,8 @@ def test_power_weighted_light_selection():
         d.emitters = [EmitterDesc(type="area", emission=(1, 1, 1), scale=100.0),
                       EmitterDesc(type="area", emission=(1, 1, 1), scale=0.5)]
         floor = quad((-3, 0, -3), (-3, 0, 3), (3, 0, 3), (3, 0, -3))
-        big = quad((-0.5, 3, -0.5), (-0.5, 3, 0.5), (0.5, 3, 0.5), (0.5, 3, -0.5))[::-1]
-        small = quad((1.5, 3, -0.5), (1.5, 3, 0.5), (2.5, 3, 0.5), (2.5, 3, -0.5))[::-1]
+        big = quad((-0.5, 3, -0.5), (0.5, 3, -0.5), (0.5, 3, 0.5), (-0.5, 3, 0.5))    # -y
+        small = quad((1.5, 3, -0.5), (2.5, 3, -0.5), (2.5, 3, 0.5), (1.5, 3, 0.5))     # -y
         d.objects = [ObjectDesc(tris=floor, bsdf=0),
                      ObjectDesc(tris=big, bsdf=1, emitter=0),
                      ObjectDesc(tris=small, bsdf=1, emitter=1)]
