@@ -199,8 +199,15 @@ HD float bsdf_pdf(const BsdfParams& b, const Vec3& wo, const Vec3& wi, const Int
 }
 
 // ------------------------------------------------------------------- sample
+// lambda_io: per-path wavelength slot for spectral dispersion (0 = not yet
+// sampled).  The wavelength is sampled ONCE per path at the first dispersive
+// transmission and the lambda->RGB basis weight applied exactly once; later
+// dispersive events reuse the same wavelength (consistent IOR) with weight 1.
+// (Per-event resampling squares the non-white basis expectation: measured
+// +36% red in the dispersion furnace.)
 HD BsdfSample bsdf_sample(const BsdfParams& b, const Vec3& wo, const Interaction& it,
-                          Sampler& sp, const TexView* textures) {
+                          Sampler& sp, const TexView* textures,
+                          float* lambda_io = nullptr) {
     BsdfSample s{};
     Vec3 n = it.shading_n;
     float cos_o = n.dot(wo);
@@ -314,8 +321,10 @@ HD BsdfSample bsdf_sample(const BsdfParams& b, const Vec3& wo, const Interaction
         return s;
     }
     case BSDF_DISPERSION: {
-        // spectral glass: one wavelength per path (reference dispersion.cuh)
-        float lambda = LAMBDA_MIN + (LAMBDA_MAX - LAMBDA_MIN) * sp.next1f();
+        // spectral glass: ONE wavelength per path (reference dispersion.cuh)
+        bool have_l = lambda_io && *lambda_io > 0.f;
+        float lambda = have_l ? *lambda_io
+                              : LAMBDA_MIN + (LAMBDA_MAX - LAMBDA_MIN) * sp.next1f();
         float ior = cauchy_ior(b.extra0, b.extra1, lambda);
         bool entering = cos_o > 0.f;
         Vec3 nn = entering ? n : -n;
@@ -324,13 +333,14 @@ HD BsdfSample bsdf_sample(const BsdfParams& b, const Vec3& wo, const Interaction
         Vec3 tint = b.ks.xyz();
         Vec3 wt;
         bool refr_ok = snell_refraction(wo, nn, eta, wt);
-        Vec3 spectral = wavelength_to_rgb(lambda);
         if (!refr_ok || sp.next1f() < F) {
             s.wi = reflect_dir(wo, nn);
             s.pdf = refr_ok ? F : 1.f;
             s.weight = tint;  // reflection is not dispersive
             s.lobe = LOBE_SPECULAR | LOBE_DELTA;
         } else {
+            Vec3 spectral = have_l ? Vec3(1.f) : wavelength_to_rgb(lambda);
+            if (lambda_io) *lambda_io = lambda;
             s.wi = wt.normalized();
             s.pdf = 1.f - F;
             s.weight = tint * spectral * (eta * eta);

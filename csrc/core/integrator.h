@@ -38,6 +38,7 @@ struct PathState {
     // primary-hit AOVs (denoiser guides): shading normal, depth, albedo
     Vec3 aov_n, aov_alb;
     float aov_t;
+    float lambda;   // per-path dispersion wavelength (0 = unsampled)
 
     HD void reset(const Ray& r) {
         ray = r;
@@ -49,6 +50,7 @@ struct PathState {
         st = PathStats();
         b = 0; iter = 0;
         aov_n = Vec3(0.f); aov_alb = Vec3(0.f); aov_t = 0.f;
+        lambda = 0.f;
     }
 };
 
@@ -156,7 +158,7 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
         }
 
         // ---- BSDF sampling
-        BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
+        BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures, &ps.lambda);
         if (bs.pdf <= 0.f || bs.weight.is_zero()) return true;
         if (bs.weight.has_nan() || bs.wi.has_nan()) return true;  // numeric scrub
         thp *= bs.weight;

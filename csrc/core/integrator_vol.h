@@ -68,6 +68,7 @@ HD Vec3 transmittance_estimate(const SceneView& sv, Vec3 from, const Vec3& wi, f
 }
 
 HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
+    float path_lambda = 0.f;
     Vec3 L(0.f), thp(1.f);
     float prev_pdf = 0.f;
     bool prev_delta = true;
@@ -191,7 +192,7 @@ HD Vec3 trace_path_volumetric(const SceneView& sv, Ray ray, Sampler& sp, TravCtx
                 }
             }
 
-            BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
+            BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures, &path_lambda);
             if (bs.pdf <= 0.f || bs.weight.is_zero() || bs.weight.has_nan() || bs.wi.has_nan()) break;
             thp *= bs.weight;
             if ((bs.lobe & LOBE_TRANSMIT) != 0) cross_boundary(stack, obj, bs.wi, geo_n);

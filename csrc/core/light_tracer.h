@@ -29,6 +29,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
     if (!le.valid) return;
 
     const float inv_npix = 1.f / (float(sv.cam.w) * float(sv.cam.h));
+    float path_lambda = 0.f;
     const Vec3 cam_fwd = sv.cam.R * Vec3(0.f, 0.f, 1.f);
 
     // connect a vertex (pos, normal, f_to_cam callback result) to the camera
@@ -86,7 +87,7 @@ HD void trace_light_path_impl(const SceneView& sv, Sampler& sp, SPLAT_FN&& splat
             connect(pos, thp * f, n_spec);
         }
 
-        BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
+        BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures, &path_lambda);
         if (bs.pdf <= 0.f || bs.weight.is_zero() || bs.weight.has_nan() || bs.wi.has_nan()) break;
         thp *= bs.weight;
         if (bs.lobe & (LOBE_SPECULAR | LOBE_TRANSMIT)) ++n_spec;

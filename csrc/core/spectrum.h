@@ -42,9 +42,11 @@ HD Vec3 xyz_to_srgb(const Vec3& c) {
 constexpr float LAMBDA_MIN = 360.f, LAMBDA_MAX = 830.f;
 HD Vec3 wavelength_to_rgb(float lambda_nm) {
     Vec3 xyz{cie_x(lambda_nm), cie_y(lambda_nm), cie_z(lambda_nm)};
-    // normalize so that a flat spectrum integrates to ~white
-    Vec3 rgb = xyz_to_srgb(xyz * ((LAMBDA_MAX - LAMBDA_MIN) / 106.857f));
-    return rgb.maxv(Vec3(0.f));
+    // normalize so a flat spectrum integrates to ~white.  Do NOT clamp the
+    // negative out-of-gamut sRGB lobes: the single-wavelength estimator is
+    // linear and relies on them canceling in expectation — clamping per
+    // sample was measured to ADD ~30% red energy in the dispersion furnace.
+    return xyz_to_srgb(xyz * ((LAMBDA_MAX - LAMBDA_MIN) / 106.857f));
 }
 
 // Planck blackbody radiance -> linear sRGB (normalized to luminance ~1 at the

@@ -239,9 +239,10 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
         }
     }
 
-    // BSDF sample
+    // BSDF sample (per-path dispersion wavelength lives in ray_o.w)
     uint32_t status = DEAD;
-    BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures);
+    float path_lambda = ro4.w;
+    BsdfSample bs = bsdf_sample(bsdf, -ray.d, it, sp, sv.textures, &path_lambda);
     if (bs.pdf > 0.f && !bs.weight.is_zero() && !bs.weight.has_nan() && !bs.wi.has_nan()) {
         thp *= bs.weight;
         int nd = counts & 0xFF, ns = (counts >> 8) & 0xFF, nt = (counts >> 16) & 0xFF,
@@ -269,7 +270,7 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
             prev_pdf = bs.pdf;
             prev_n = it.shading_n;
             Vec3 no = fmadd(bs.wi, EPSILON, pos);
-            st.ray_o[i] = make_float4(no.x, no.y, no.z, 0.f);
+            st.ray_o[i] = make_float4(no.x, no.y, no.z, path_lambda);
             st.ray_d[i] = make_float4(bs.wi.x, bs.wi.y, bs.wi.z, prev_pdf);
         }
     }

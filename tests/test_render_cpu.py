@@ -94,6 +94,37 @@ class TestFurnace:
         # coated diffuse loses a little energy to the interlayer model
         assert 0.8 < m < 1.05, m
 
+    def test_plastic_forward(self):
+        img = render_desc(furnace_scene(
+            BsdfDesc(type="plastic-forward", kd=(1, 1, 1), ior=1.5,
+                     trans_scaler=1.0)), spp=48)
+        m = img[..., :3].mean()
+        assert 0.8 < m < 1.05, m
+
+    def test_forward_null(self):
+        # null boundary passes everything through untouched
+        img = render_desc(furnace_scene(BsdfDesc(type="forward")), spp=24)
+        assert abs(img[..., :3].mean() - 1.0) < 0.02
+
+    def test_dispersion_near_white(self):
+        # spectral glass with unit ks: energy conserved within spectral
+        # noise.  A flat spectrum maps to equal-energy white (1.20, 0.95,
+        # 0.91 in linear sRGB), not D65 — bracket per channel around that.
+        img = render_desc(furnace_scene(
+            BsdfDesc(type="dispersion", ks=(0.99, 0.99, 0.99), preset="bk7")),
+            spp=128)
+        r, g, b = img[..., :3].mean(axis=(0, 1))
+        assert 1.05 < r < 1.35 and 0.85 < g < 1.08 and 0.8 < b < 1.05, (r, g, b)
+        assert 0.9 < (r + g + b) / 3 < 1.12
+
+    def test_translucent_total_internal_reflection(self):
+        # high-IOR glass still conserves energy (TIR paths terminate inside
+        # only via the bounce caps, never create energy)
+        img = render_desc(furnace_scene(
+            BsdfDesc(type="glass", ks=(1, 1, 1), ior=2.4)), spp=64)
+        m = img[..., :3].mean()
+        assert 0.85 < m < 1.05, m
+
 
 class TestMIS:
     def test_nee_vs_bsdf_sampling_agree(self):
