@@ -315,14 +315,10 @@ class Scene:
         # 4-wide collapse of the binary tree: the traversal that actually runs
         # (ordered short-stack walk over 128-byte nodes, csrc/core/bvh4.h)
         nodes4, depth4 = C.collapse_bvh4(nodes)
-        # 8-wide tree A/B (HIPPT_BVH8=1): 256-byte nodes, csrc/core/bvh8.h
-        import os
-        if os.environ.get("HIPPT_BVH8"):
-            nodes8, depth8 = C.collapse_bvh8(nodes)
-        else:
-            nodes8, depth8 = np.zeros((0, 64), np.float32), 0
-        self.bvh_stats = dict(stats, n_nodes4=int(nodes4.shape[0]), depth4=int(depth4),
-                              n_nodes8=int(nodes8.shape[0]), depth8=int(depth8))
+        # (an 8-wide tree was built and measured 2.4x slower on MI355X —
+        # bvh8.h stays host-tested; no device path consumes it)
+        nodes8 = np.zeros((0, 64), np.float32)
+        self.bvh_stats = dict(stats, n_nodes4=int(nodes4.shape[0]), depth4=int(depth4))
         # one zero sentinel row past the tree: traversal speculatively fetches
         # both successor nodes per step (csrc/core/bvh.h)
         nodes = np.concatenate([nodes, np.zeros((1, 8), np.float32)])
