@@ -53,7 +53,13 @@ INDEX_HTML = """<!doctype html>
   <div class="row">max depth: <input id="maxd" size="4" value="8">
     <button onclick="setDepth()">apply</button></div>
   <div class="row"><button onclick="capture()">capture png</button>
-    <button onclick="fetch('/api/reset',{method:'POST'})">reset</button></div>
+    <button onclick="fetch('/api/reset',{method:'POST'})">reset</button>
+    <label><input type="checkbox" onchange="fetch('/api/adaptive',{method:'POST',
+      headers:{'Content-Type':'application/json'},
+      body:JSON.stringify({enabled:this.checked})})"> adaptive spp</label>
+    <label><input type="checkbox" onchange="fetch('/api/denoise',{method:'POST',
+      headers:{'Content-Type':'application/json'},
+      body:JSON.stringify({enabled:this.checked})})"> denoise</label></div>
   <div class="row">WASD move, QE up/down, arrows look</div>
 </div>
 <script>
