@@ -118,7 +118,11 @@ HD MediumSample grid_sample(const MediumParams& m, const Ray& ray, float t_max, 
     float maj = m.majorant;
     if (maj <= 0.f) { r.scattered = false; return r; }
     float inv_maj = 1.f / maj;
-    float albedo = m.sigma_s.x / fmaxf(m.sigma_s.x + m.sigma_a.x, 1e-8f);
+    // colored single-scattering albedo: Russian roulette on the mean, tint
+    // by albedo/mean on scatter (gray media: tint == 1, identical streams)
+    Vec3 alb = m.sigma_s.xyz() /
+               (m.sigma_s.xyz() + m.sigma_a.xyz()).maxv(Vec3(1e-8f));
+    float a_mean = clampv((alb.x + alb.y + alb.z) * (1.f / 3.f), 1e-6f, 1.f);
     float t = 0.f;
     for (int it = 0; it < 4096; ++it) {
         t -= logf(fmaxf(1.f - sp.next1f(), 1e-12f)) * inv_maj;
@@ -127,7 +131,8 @@ HD MediumSample grid_sample(const MediumParams& m, const Ray& ray, float t_max, 
         float sig_t = dens;  // density IS sigma_t (scale folded in)
         if (sp.next1f() < sig_t * inv_maj) {
             // real collision: scatter with albedo, absorb otherwise
-            if (sp.next1f() < albedo) {
+            if (sp.next1f() < a_mean) {
+                r.local_thp = alb * (1.f / a_mean);
                 r.dist = t;
                 r.scattered = true;
                 return r;

@@ -199,3 +199,16 @@ def test_area_spot_cone():
     c = float(lum[38:46, 28:36].mean())   # under the lamp
     edge = float(lum[50:60, 2:10].mean())  # far corner, outside the cone
     assert c > 5 * max(edge, 1e-6), (c, edge)
+
+
+def test_grid_colored_albedo():
+    """Colored grid albedo tints scattered light (the RR-on-mean estimator
+    multiplies albedo/mean per scatter); gray albedo keeps old behavior."""
+    d = smoke_box(width=40, height=30, n_grid=24)
+    d.media[0].sigma_s = (0.9, 0.3, 0.3)   # red-scattering smoke
+    d.media[0].sigma_a = (0.1, 0.7, 0.7)
+    img = hippt.PythonRenderer(d, device_id=-1).render(spp=24).numpy()
+    # plume crop must be red-dominant
+    c = img[8:22, 12:28, :3].mean(axis=(0, 1))
+    assert np.isfinite(img).all()
+    assert c[0] > 1.2 * c[2], c
