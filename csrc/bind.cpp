@@ -564,6 +564,40 @@ py::tuple py_bvh4_hit(farr prims, uarr prim_obj, farr nodes4,
     return py::make_tuple(t_out, p_out);
 }
 
+// BSDF triple-consistency check on bsdf index i of the holder's table:
+//  A = E_sample[weight]                 (hemispherical reflectance via sample)
+//  B = 2pi * E_uniform[eval]            (same integral via eval, uniform dirs)
+//  C = 2pi * E_uniform[pdf]             (pdf normalization, ~1 for non-delta)
+// Mutually consistent sample/eval/pdf must give A ~= B; C ~= (lobe coverage).
+py::tuple py_bsdf_check(SceneHolder& h, int i, float cos_o, uint32_t seed, int n) {
+    if (i < 0 || i >= (int)h.bsdfs.size()) throw std::runtime_error("bad bsdf index");
+    h.finalize();
+    const BsdfParams& b = h.bsdfs[i];
+    Interaction it{};
+    it.shading_n = Vec3(0.f, 0.f, 1.f);
+    float so = sqrtf(fmaxf(0.f, 1.f - cos_o * cos_o));
+    Vec3 wo(so, 0.f, cos_o);
+    Sampler sp(911u, seed);
+    Vec3 A(0.f), B(0.f);
+    double Cp = 0.0;
+    for (int k = 0; k < n; ++k) {
+        BsdfSample s = bsdf_sample(b, wo, it, sp, h.host_sv.textures);
+        if (s.pdf > 0.f) A += s.weight;
+        // uniform hemisphere direction
+        Vec2 u = sp.next2f();
+        float z = u.x, r = sqrtf(fmaxf(0.f, 1.f - z * z)), phi = 2.f * PI * u.y;
+        Vec3 wi(r * cosf(phi), r * sinf(phi), z);
+        B += bsdf_eval(b, wo, wi, it, h.host_sv.textures);
+        Cp += (double)bsdf_pdf(b, wo, wi, it, h.host_sv.textures);
+    }
+    float inv_n = 1.f / (float)n;
+    Vec3 a = A * inv_n;
+    Vec3 bb = B * (2.f * PI * inv_n);
+    double c = Cp * 2.0 * PI * inv_n;
+    return py::make_tuple(py::make_tuple(a.x, a.y, a.z),
+                          py::make_tuple(bb.x, bb.y, bb.z), c);
+}
+
 } // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -611,6 +645,7 @@ PYBIND11_MODULE(_C, m) {
     m.def("collapse_bvh4", &py_collapse_bvh4, py::arg("nodes"));
     m.def("collapse_bvh8", &py_collapse_bvh8, py::arg("nodes"));
     m.def("bvh4_hit", &py_bvh4_hit);
+    m.def("bsdf_check", &py_bsdf_check);
     m.def("bvh4_selftest", &py_bvh4_selftest,
           py::arg("prims"), py::arg("prim_obj"), py::arg("nodes"), py::arg("nodes4"),
           py::arg("ray_o"), py::arg("ray_d"), py::arg("tmax"),

@@ -8,6 +8,7 @@ import numpy as np
 import pytest
 
 import hippt
+import hippt._C as C_mod
 from hippt.scene.scene import (BsdfDesc, CameraDesc, EmitterDesc, ObjectDesc,
                                RenderConfig, SceneDesc)
 from hippt.scene.procedural import cornell_box, uv_sphere_mesh
@@ -332,3 +333,43 @@ def test_denoiser_improves_rmse():
     rn = float(np.sqrt(((noisy[crop] - ref[crop]) ** 2).mean()))
     rd = float(np.sqrt(((den[crop] - ref[crop]) ** 2).mean()))
     assert rd < 0.5 * rn, (rd, rn)
+
+
+class TestBsdfConsistency:
+    """Triple consistency per BSDF: hemispherical reflectance via the sampler
+    (E[f cos/pdf]) must equal the uniform-direction eval integral, and the
+    pdf must integrate to the non-delta lobe probability."""
+
+    def _scene(self, bsdfs):
+        from hippt.scene.scene import Scene, SceneDesc, ObjectDesc, CameraDesc, RenderConfig
+        from hippt.scene.procedural import quad
+        d = SceneDesc()
+        d.bsdfs = bsdfs
+        d.objects = [ObjectDesc(tris=quad((-1, 0, -1), (-1, 0, 1), (1, 0, 1), (1, 0, -1)), bsdf=0)]
+        d.camera = CameraDesc(width=16, height=16)
+        d.config = RenderConfig()
+        return Scene(d)
+
+    def test_lambert_exact(self):
+        sc = self._scene([BsdfDesc(type="lambertian", kd=(0.6, 0.4, 0.2))])
+        A, B, C = C_mod.bsdf_check(sc.native, 0, 0.5, 7, 100000)
+        np.testing.assert_allclose(A, (0.6, 0.4, 0.2), rtol=0.02)
+        np.testing.assert_allclose(B, (0.6, 0.4, 0.2), rtol=0.02)
+        assert abs(C - 1.0) < 0.02
+
+    def test_ggx_iso_and_aniso(self):
+        sc = self._scene([BsdfDesc(type="ggx", metal="Ag", roughness_x=0.3, roughness_y=0.3),
+                          BsdfDesc(type="ggx", metal="Au", roughness_x=0.5, roughness_y=0.1)])
+        for i in range(2):
+            for co in (0.3, 0.8):
+                A, B, C = C_mod.bsdf_check(sc.native, i, co, 11, 200000)
+                np.testing.assert_allclose(A, B, rtol=0.06)
+                assert abs(C - 1.0) < 0.05, C
+
+    def test_plastic_delta_share(self):
+        sc = self._scene([BsdfDesc(type="plastic", kd=(0.7, 0.5, 0.3), ior=1.5)])
+        A, B, C = C_mod.bsdf_check(sc.native, 0, 0.8, 7, 200000)
+        # sample includes the delta coat; eval only the diffuse lobe
+        assert all(a >= b - 0.02 for a, b in zip(A, B))
+        # pdf integrates to the diffuse-lobe probability (1-F) < 1
+        assert 0.8 < C < 1.0, C
