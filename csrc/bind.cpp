@@ -266,9 +266,10 @@ struct SceneHolder {
     }
 
     void set_depths(int max_depth, int max_diffuse, int max_specular, int max_transmit,
-                    int max_volume, float min_time, float max_time, int use_tof) {
+                    int max_volume, float min_time, float max_time, int use_tof,
+                    float radiance_clamp = 0.f) {
         md = {max_depth, max_diffuse, max_specular, max_transmit, max_volume,
-              min_time, max_time, use_tof};
+              min_time, max_time, use_tof, radiance_clamp};
     }
 
     void fill_common(SceneView& sv) {
@@ -619,7 +620,10 @@ PYBIND11_MODULE(_C, m) {
         .def("update_medium", &SceneHolder::update_medium)
         .def("add_texture", &SceneHolder::add_texture)
         .def("set_camera", &SceneHolder::set_camera)
-        .def("set_depths", &SceneHolder::set_depths)
+        .def("set_depths", &SceneHolder::set_depths,
+             py::arg("max_depth"), py::arg("max_diffuse"), py::arg("max_specular"),
+             py::arg("max_transmit"), py::arg("max_volume"), py::arg("min_time"),
+             py::arg("max_time"), py::arg("use_tof"), py::arg("radiance_clamp") = 0.f)
         .def_readwrite("cam_medium", &SceneHolder::cam_medium)
         .def_readwrite("env_emitter", &SceneHolder::env_emitter)
         .def("finalize", &SceneHolder::finalize)

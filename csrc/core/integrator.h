@@ -188,12 +188,20 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
 }
 
 // Full path trace for one camera ray. Returns radiance estimate.
+// Per-sample radiance cap (md.radiance_clamp > 0): the standard biased
+// firefly control — applied at sample commit, never inside the estimator.
+HD Vec3 clamp_radiance(const SceneView& sv, Vec3 L) {
+    float c = sv.md.radiance_clamp;
+    if (c > 0.f) L = L.minv(Vec3(c));
+    return L;
+}
+
 HD Vec3 trace_path(const SceneView& sv, Ray ray, Sampler& sp, TravCtx tc = {}) {
     PathState ps;
     ps.reset(ray);
     while (!path_step(sv, ps, sp, tc)) {}
     if (ps.L.has_nan()) return Vec3(0.f);
-    return ps.L;
+    return clamp_radiance(sv, ps.L);
 }
 
 // Depth renderer: distance of the primary hit (reference pt_impl/depth.cu).

@@ -38,6 +38,7 @@ struct MaxDepthParams {
     float min_time;    // ToF gating window (SUPPORTS_TOF_RENDERING parity)
     float max_time;
     int use_tof;
+    float radiance_clamp;  // per-sample radiance cap, 0 = off (firefly knob)
 };
 
 struct SceneView {

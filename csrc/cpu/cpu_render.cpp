@@ -51,7 +51,7 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
                 Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
                 Ray ray = sv.cam.gen_ray(x, y, sp, spp0 + s);
                 Vec3 L(0.f);
-                if (renderer == 2)      L = trace_path_volumetric(sv, ray, sp);
+                if (renderer == 2)      L = clamp_radiance(sv, trace_path_volumetric(sv, ray, sp));
                 else if (renderer == 4) L = Vec3(trace_depth(sv, ray));
                 else if (renderer == 5) { Vec2 c = trace_bvh_cost(sv, ray); L = Vec3(c.x, c.y, 0.f); }
                 else if (aux) {

@@ -82,7 +82,7 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
         ps.reset(sv.cam.gen_ray(px, py, sp, spp0));
         for (;;) {
             if (path_step(sv, ps, sp, tc)) {
-                Vec3 L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
+                Vec3 L = ps.L.has_nan() ? Vec3(0.f) : clamp_radiance(sv, ps.L);
                 Lsum += L;
                 float lum = (L.x + L.y + L.z) * (1.f / 3.f);
                 lum_s += lum;
@@ -103,7 +103,7 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
         Sampler sp(uint32_t(pix), uint32_t(spp0 + s) * SEED_SCALER + seed);
         Ray ray = sv.cam.gen_ray(px, py, sp, spp0 + s);
         Vec3 L(0.f);
-        if constexpr (RENDERER == R_VOLUME_PT) L = trace_path_volumetric(sv, ray, sp, tc);
+        if constexpr (RENDERER == R_VOLUME_PT) L = clamp_radiance(sv, trace_path_volumetric(sv, ray, sp, tc));
         else if constexpr (RENDERER == R_DEPTH) L = Vec3(trace_depth(sv, ray, tc));
         else if constexpr (RENDERER == R_BVH_COST) { Vec2 c = trace_bvh_cost(sv, ray); L = Vec3(c.x, c.y, 0.f); }
         else L = trace_path(sv, ray, sp, tc);

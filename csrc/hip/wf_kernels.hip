@@ -400,12 +400,13 @@ void k_wf_trace_dual(SceneView sv, WfState st, int lds_n) {
 
 // ----------------------------------------------------------------- splat
 __global__ __launch_bounds__(256)
-void k_wf_splat(WfState st, float* __restrict__ accum, float* __restrict__ var, int nspp_done) {
+void k_wf_splat(SceneView sv, WfState st, float* __restrict__ accum, float* __restrict__ var, int nspp_done) {
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= st.n) return;
     float4 l4 = st.L[i];
     Vec3 L(l4.x, l4.y, l4.z);
     if (L.has_nan()) L = Vec3(0.f);
+    L = clamp_radiance(sv, L);
     float* a = accum + (size_t)i * 4;
     a[0] += L.x; a[1] += L.y; a[2] += L.z; a[3] += 1.f;
     if (var) {
@@ -521,7 +522,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
                 hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
             }
         }
-        hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, *st, accum, var, 1);
+        hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, sv, *st, accum, var, 1);
     }
     return (int)hipGetLastError();
 }

@@ -88,6 +88,20 @@ class PythonRenderer:
     def save(self, path: str, gamma: float = 2.1):
         self.renderer.save(path, gamma)
 
+    def enable_aov(self):
+        """Allocate primary-hit AOV buffers (denoiser guides)."""
+        self.renderer.enable_aov()
+        return self
+
+    def aov(self):
+        """dict(normal, depth, albedo) means over accumulated samples."""
+        return self.renderer.aov()
+
+    def denoise(self, iterations: int = 2, **kw):
+        """SVGF-lite a-trous denoise of the accumulation (enable_aov first);
+        returns an (H, W, 3) tensor/array on the render device."""
+        return self.renderer.denoise(iterations=iterations, **kw)
+
     def save_state(self, path: str):
         """Checkpoint the warm accumulation state (radiance sums, variance
         sums, sample counter) so a long accumulation can resume after a

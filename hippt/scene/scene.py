@@ -157,6 +157,7 @@ class RenderConfig:
     use_sbvh: bool = False
     ref_unsplit: bool = True
     cache_level: int = 6
+    radiance_clamp: float = 0.0        # per-sample firefly cap (0 = off)
     spec_constraint: int = 0
     caustic_scaling: float = 1.0
     bidirectional: bool = False
@@ -419,7 +420,8 @@ class Scene:
         self._set_camera_native()
         self.native.set_depths(cfg.max_depth, cfg.max_diffuse, cfg.max_specular,
                                cfg.max_transmit, cfg.max_volume,
-                               cfg.min_time, cfg.max_time, int(cfg.use_tof))
+                               cfg.min_time, cfg.max_time, int(cfg.use_tof),
+                               cfg.radiance_clamp)
         self.native.cam_medium = d.cam_medium
         self.native.finalize()
 
@@ -521,7 +523,8 @@ class Scene:
             setattr(cfg, k, v)
         self.native.set_depths(cfg.max_depth, cfg.max_diffuse, cfg.max_specular,
                                cfg.max_transmit, cfg.max_volume,
-                               cfg.min_time, cfg.max_time, int(cfg.use_tof))
+                               cfg.min_time, cfg.max_time, int(cfg.use_tof),
+                               cfg.radiance_clamp)
         self.native.finalize()
 
     @property

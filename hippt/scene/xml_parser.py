@@ -163,6 +163,7 @@ def parse_xml(path: str) -> SceneDesc:
         cfg.spec_constraint = max(props.get("specular_constraint", 0), 0)
         cfg.bidirectional = bool(props.get("bidirectional", False))
         cfg.caustic_scaling = props.get("caustic_scaling", 1.0)
+        cfg.radiance_clamp = props.get("radiance_clamp", 0.0)
         if "min_time" in props or "max_time" in props:
             cfg.use_tof = True
             cfg.min_time = props.get("min_time", 0.0)

@@ -299,6 +299,7 @@ try:
         emission_scale: Optional[float] = None
 
     class DepthReq(_BaseModel):
+        radiance_clamp: Optional[float] = None
         max_depth: Optional[int] = None
         max_diffuse: Optional[int] = None
         max_specular: Optional[int] = None

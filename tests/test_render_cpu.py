@@ -373,3 +373,20 @@ class TestBsdfConsistency:
         assert all(a >= b - 0.02 for a, b in zip(A, B))
         # pdf integrates to the diffuse-lobe probability (1-F) < 1
         assert 0.8 < C < 1.0, C
+
+
+def test_radiance_clamp():
+    """radiance_clamp caps per-sample radiance (firefly knob): clamped render
+    mean <= unclamped, and a clamp above the max is a no-op."""
+    from hippt.scene.procedural import cornell_box
+    d = cornell_box(width=32, height=32, spp=1, max_depth=4)
+    base = hippt.PythonRenderer(d, device_id=-1).render(spp=8).numpy()
+    d2 = cornell_box(width=32, height=32, spp=1, max_depth=4)
+    d2.config.radiance_clamp = 2.0
+    lo = hippt.PythonRenderer(d2, device_id=-1).render(spp=8).numpy()
+    d3 = cornell_box(width=32, height=32, spp=1, max_depth=4)
+    d3.config.radiance_clamp = 1e6
+    hi = hippt.PythonRenderer(d3, device_id=-1).render(spp=8).numpy()
+    assert lo[..., :3].max() <= 2.0 + 1e-5
+    assert lo[..., :3].mean() < base[..., :3].mean()      # light pixels capped
+    np.testing.assert_allclose(hi, base, rtol=1e-6)       # no-op at huge clamp
