@@ -599,6 +599,36 @@ py::tuple py_bsdf_check(SceneHolder& h, int i, float cos_o, uint32_t seed, int n
                           py::make_tuple(bb.x, bb.y, bb.z), c);
 }
 
+// Envmap NEE integral check: A = E[radiance/pdf] over emitter_sample draws
+// (hemisphere above n=(0,0,1)) must equal B = the true integral of
+// envmap radiance over that hemisphere (uniform-direction reference).
+// Tight validation of the luminance-CDF sampler's pdf normalization.
+py::tuple py_env_check(SceneHolder& h, uint32_t seed, int n) {
+    h.finalize();
+    if (h.env_emitter < 0) throw std::runtime_error("no envmap emitter");
+    const EmitterParams& e = h.emitters[h.env_emitter];
+    EmitterGeom g = h.host_sv.emitter_geom();
+    Sampler sp(417u, seed);
+    Vec3 A(0.f), B(0.f);
+    Vec3 sp_n(0.f, 0.f, 1.f), sp_pos(0.f);
+    for (int k = 0; k < n; ++k) {
+        EmitterSampleRec r = emitter_sample(e, g, sp_pos, sp_n, sp);
+        if (r.pdf > 0.f) {
+            Vec3 dir = (r.pos - sp_pos).normalized();
+            if (dir.z > 0.f) A += r.radiance / r.pdf;   // hemisphere only
+        }
+        Vec2 u = sp.next2f();
+        float z = u.x, rr = sqrtf(fmaxf(0.f, 1.f - z * z)), phi = 2.f * PI * u.y;
+        Vec3 wi(rr * cosf(phi), rr * sinf(phi), z);
+        B += envmap_eval(e, wi, h.host_sv.textures);
+    }
+    float inv_n = 1.f / (float)n;
+    Vec3 a = A * inv_n;
+    Vec3 b = B * (2.f * PI * inv_n);
+    return py::make_tuple(py::make_tuple(a.x, a.y, a.z),
+                          py::make_tuple(b.x, b.y, b.z));
+}
+
 } // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -650,6 +680,7 @@ PYBIND11_MODULE(_C, m) {
     m.def("collapse_bvh8", &py_collapse_bvh8, py::arg("nodes"));
     m.def("bvh4_hit", &py_bvh4_hit);
     m.def("bsdf_check", &py_bsdf_check);
+    m.def("env_check", &py_env_check);
     m.def("bvh4_selftest", &py_bvh4_selftest,
           py::arg("prims"), py::arg("prim_obj"), py::arg("nodes"), py::arg("nodes4"),
           py::arg("ray_o"), py::arg("ray_d"), py::arg("tmax"),
