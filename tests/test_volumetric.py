@@ -212,3 +212,14 @@ def test_grid_colored_albedo():
     c = img[8:22, 12:28, :3].mean(axis=(0, 1))
     assert np.isfinite(img).all()
     assert c[0] > 1.2 * c[2], c
+
+
+def test_adaptive_vpt():
+    """Adaptive sampling also drives the volumetric megakernel."""
+    d = smoke_box(width=32, height=24, n_grid=16)
+    r = hippt.PythonRenderer(d, device_id=-1)
+    r.render(spp=16, adaptive=True)
+    img = r.renderer.raw()
+    assert np.isfinite(img).all()
+    cnt = img[..., 3]
+    assert cnt.std() > 0.1 and 8 <= cnt.mean() <= 32
