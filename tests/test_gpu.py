@@ -219,3 +219,21 @@ def test_denoiser_gpu():
     rn = float(np.sqrt(((noisy[crop] - ref[crop]) ** 2).mean()))
     rd = float(np.sqrt(((dn[crop] - ref[crop]) ** 2).mean()))
     assert rd < 0.6 * rn, (rd, rn)
+
+
+@pytest.mark.gpu
+def test_band_render_exact():
+    """Tile-band rendering (y0/y1) composes exactly: two half-band renders
+    accumulate to the same buffer as one full render (same per-pixel sampler
+    streams; disjoint rows)."""
+    d = cornell_box(width=64, height=64, max_depth=4)
+    r_full = hippt.PythonRenderer(d, device_id=0)
+    r_full.renderer.render(8)
+    full = r_full.renderer.accum.cpu().numpy()
+    d2 = cornell_box(width=64, height=64, max_depth=4)
+    r_band = hippt.PythonRenderer(d2, device_id=0)
+    r_band.renderer.render(8, y0=0, y1=32)
+    r_band.renderer.accum_cnt -= 8   # same sample indices for the second band
+    r_band.renderer.render(8, y0=32, y1=64)
+    band = r_band.renderer.accum.cpu().numpy()
+    np.testing.assert_array_equal(full, band)
