@@ -50,6 +50,9 @@ def build_argparser():
                          "(rank id is appended); resumes from it if present")
     ap.add_argument("--checkpoint-interval", type=int, default=256,
                     help="steps between snapshots")
+    ap.add_argument("--adaptive", action="store_true",
+                    help="variance-guided per-pixel budgets after the first "
+                         "step (megakernel renderers)")
     return ap
 
 
@@ -152,7 +155,11 @@ def main(argv=None):
     merged = None
     start_step = min(rend.accum_cnt // args.spp_per_call, steps)
     for k in range(start_step, steps):
-        rend.render(args.spp_per_call, y0=band[0], y1=band[1])
+        if args.adaptive and rend.accum_cnt >= args.spp_per_call and rend.rid in (0, 2):
+            m = rend._spp_budget(args.spp_per_call)
+            rend.render(args.spp_per_call, y0=band[0], y1=band[1], spp_map=m)
+        else:
+            rend.render(args.spp_per_call, y0=band[0], y1=band[1])
         if ckpt_path and (k + 1) % args.checkpoint_interval == 0:
             r.save_state(ckpt_path)
         if world_size > 1 and ((k + 1) % args.reduce_interval == 0 or k == steps - 1):

@@ -113,7 +113,7 @@ def test_ddp_checkpoint_resume(tmp_path):
     out = tmp_path / "ck.png"
     ck = tmp_path / "state"
     # run 1: 8 spp with snapshots every 1 step of 2 spp
-    res = run_torchrun(2, ["-m", "hippt.parallel.ddp", "--cpu",
+    res = run_torchrun(2, ["-m", "hippt.parallel.ddp", "--cpu", "--adaptive",
                            "--scene", "cornell", "--width", "48", "--height", "48",
                            "--spp", "16", "--spp-per-call", "2",
                            "--checkpoint", str(ck), "--checkpoint-interval", "1",
