@@ -31,6 +31,15 @@ def main():
         np.savez_compressed(os.path.join(OUT, name.replace(".xml", "") + ".npz"),
                             img=img, spp=spp, w=w)
         print(name, img.shape, float(img[..., :3].mean()))
+    # procedural flagship scenes (tiny CPU renders)
+    import hippt
+    from hippt.scene.procedural import kitchen, sports_car
+    for gen, nm in [(kitchen, "kitchen"), (sports_car, "sports-car")]:
+        d = gen(width=64, height=36)
+        r = hippt.PythonRenderer(d, device_id=-1)
+        img = r.render(spp=8).numpy().astype(np.float32)
+        np.savez_compressed(os.path.join(OUT, nm + ".npz"), img=img, spp=8, w=64)
+        print(nm, img.shape, float(img[..., :3].mean()))
 
 
 if __name__ == "__main__":

@@ -15,6 +15,7 @@ GOLD = os.path.join(ROOT, "tests", "golden")
 
 NAMES = ["cornell-box", "balls", "grid-cbox", "diamonds", "env-balls",
          "point-cbox"]
+PROC = ["kitchen", "sports-car"]
 
 
 @pytest.mark.parametrize("name", NAMES)
@@ -29,3 +30,13 @@ def test_golden(name):
     # same seeds + deterministic CPU path -> near-bit-exact; small rtol for
     # cross-compiler fp differences
     np.testing.assert_allclose(img, ref, rtol=2e-4, atol=2e-4, err_msg=name)
+
+
+@pytest.mark.parametrize("name", PROC)
+def test_golden_procedural(name):
+    from hippt.scene.procedural import kitchen, sports_car
+    z = np.load(os.path.join(GOLD, name + ".npz"))
+    gen = {"kitchen": kitchen, "sports-car": sports_car}[name]
+    d = gen(width=64, height=36)
+    img = hippt.PythonRenderer(d, device_id=-1).render(spp=int(z["spp"])).numpy()
+    np.testing.assert_allclose(img, z["img"], rtol=2e-4, atol=2e-4, err_msg=name)
