@@ -314,6 +314,12 @@ try:
 
     class MoveReq(_BaseModel):
         key: str
+
+    class CamReq(_BaseModel):
+        fov: Optional[float] = None
+        aperture: Optional[float] = None
+        focal_dist: Optional[float] = None
+        ortho: Optional[bool] = None
 except ImportError:  # viewer optional without fastapi/pydantic
     pass
 
@@ -390,6 +396,19 @@ def build_app(viewer: ViewerApp):
     @app.post("/api/depths")
     def depths(req: DepthReq):
         viewer.set_depths(**{k: v for k, v in req.dict().items() if v is not None})
+        return {"ok": True}
+
+    @app.post("/api/camera/params")
+    def camera_params(req: CamReq):
+        with viewer.lock:
+            cam = viewer.pyr.scene.desc.camera
+            for k in ("fov", "aperture", "focal_dist", "ortho"):
+                v = getattr(req, k)
+                if v is not None:
+                    setattr(cam, k, v)
+            viewer.pyr.scene._set_camera_native()
+            viewer.pyr.scene.native.finalize()
+            viewer.pyr.renderer.reset()
         return {"ok": True}
 
     @app.post("/api/camera/move")

@@ -106,3 +106,14 @@ def test_denoise_toggle(client):
     png = c.get("/frame.png")
     assert png.status_code == 200 and png.content[:4] == b"\x89PNG"
     c.post("/api/denoise", json={"enabled": False})
+
+
+def test_camera_params(client):
+    c, viewer = client
+    r = c.post("/api/camera/params", json={"aperture": 0.2, "focal_dist": 3.0})
+    assert r.status_code == 200
+    assert viewer.pyr.scene.desc.camera.aperture == 0.2
+    import time
+    time.sleep(0.3)
+    png = c.get("/frame.png")
+    assert png.status_code == 200
