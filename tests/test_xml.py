@@ -135,3 +135,15 @@ class TestParserRobustness:
             <ref type='material' id='does-not-exist'/></shape></scene>""")
         d = parse_xml(str(p))   # falls back to bsdf 0
         assert len(d.objects) == 1 and d.objects[0].bsdf == 0
+
+
+def test_dof_balls_xml():
+    d, img = TestExampleScenes()._render("dof-balls.xml", spp=8, w=80, h=45)
+    assert d.camera.aperture == 0.12 and d.camera.focal_dist == 4.2
+    assert img[..., :3].mean() > 0.01
+
+
+def test_spot_cbox_xml():
+    d, img = TestExampleScenes()._render("spot-cbox.xml", spp=8)
+    assert d.emitters[0].type == "area-spot"
+    assert img[..., :3].mean() > 0.005
