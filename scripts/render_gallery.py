@@ -21,7 +21,8 @@ def main():
     m = 8 if hq else 1
     xml_jobs = [("cornell-box.xml", 512 * m), ("balls.xml", 512 * m), ("grid-cbox.xml", 256 * m),
                 ("diamonds.xml", 512 * m), ("env-balls.xml", 256 * m), ("caustics-lt.xml", 512 * m),
-                ("tof-cbox.xml", 384 * m), ("point-cbox.xml", 256 * m)]
+                ("tof-cbox.xml", 384 * m), ("point-cbox.xml", 256 * m),
+                ("dof-balls.xml", 256 * m), ("spot-cbox.xml", 256 * m)]
     for name, spp in xml_jobs:
         d = parse_xml(os.path.join(ROOT, "scenes", name))
         if small:
