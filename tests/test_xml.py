@@ -139,7 +139,7 @@ class TestParserRobustness:
 
 def test_dof_balls_xml():
     d, img = TestExampleScenes()._render("dof-balls.xml", spp=8, w=80, h=45)
-    assert d.camera.aperture == 0.12 and d.camera.focal_dist == 4.2
+    assert d.camera.aperture == 0.07 and d.camera.focal_dist == 5.2
     assert img[..., :3].mean() > 0.01
 
 
