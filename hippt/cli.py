@@ -33,6 +33,8 @@ def main(argv=None):
     ap.add_argument("--variance", default=None, help="also write the variance map here")
     ap.add_argument("--adaptive", action="store_true",
                     help="variance-guided per-pixel sample allocation")
+    ap.add_argument("--denoise", action="store_true",
+                    help="SVGF-lite AOV-guided denoise of the final image")
     args = ap.parse_args(argv)
 
     import hippt
@@ -63,6 +65,8 @@ def main(argv=None):
           f"device={'cpu' if device < 0 else device}")
     t0 = time.perf_counter()
     r = hippt.PythonRenderer(desc, device_id=device, seed_offset=args.seed)
+    if args.denoise and desc.config.renderer in ("pt", "pt-dyn", "vpt"):
+        r.renderer.enable_aov()
     info = r.info()
     print(f"[hippt] prims={info['n_prims']} nodes={info['n_nodes']} "
           f"bsdfs={info['n_bsdfs']} emitters={info['n_emitters']} "
@@ -90,6 +94,13 @@ def main(argv=None):
         vals = acc[:, :, 0] / np.maximum(acc[:, :, 3], 1e-9)
         write_png(args.output, false_color(vals, cmap=args.colormap,
                                            log_scale=desc.config.renderer == "bvh-cost"))
+    elif args.denoise and getattr(r.renderer, "aux", None) is not None:
+        import numpy as np
+        from .utils.png import write_png, tonemap
+        den = r.renderer.denoise()
+        den = den.cpu().numpy() if hasattr(den, "cpu") else np.asarray(den)
+        acc = np.concatenate([den, np.ones_like(den[..., :1])], axis=2)
+        write_png(args.output, tonemap(acc, gamma=args.gamma))
     else:
         r.save(args.output, gamma=args.gamma)
     print(f"[hippt] wrote {args.output} ({time.perf_counter() - t0:.1f}s total)")
