@@ -17,8 +17,8 @@
 namespace hippt {
 
 // ------------------------------------------------------------- PT megakernel
-// MINWAVES = __launch_bounds__ waves/SIMD floor (occupancy vs register trade;
-// 0 keeps the compiler's choice).  Runtime-selectable for A/B via HIPPT_WAVES.
+// MINW = exact waves/SIMD residency (amdgpu_waves_per_eu pin; HIPPT_OCC
+// selects the instantiation at launch).
 // XCD-aware tile swizzle: MI355X dispatches consecutive workgroups
 // round-robin over the 8 XCDs (each with its own L2).  Remapping the flat
 // block id so blocks congruent mod 8 cover one contiguous screen band gives
