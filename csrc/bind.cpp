@@ -516,6 +516,9 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
         HitRecord a = ray_intersect_bvh(bn, n_nodes, pr, po, r, tmax);
         HitRecord b = ray_intersect_bvh4(n4, pr, po, r, tmax);
         HitRecord w = ray_intersect_bvh4_ww(n4, pr, po, r, tmax);
+        HitRecord w32 = ray_intersect_bvh4_ww32(n4, pr, po, r, tmax);
+        if ((w32.prim_idx < 0) != (b.prim_idx < 0) ||
+            (b.prim_idx >= 0 && fabsf(w32.t - b.t) > 1e-5f * fmaxf(1.f, b.t))) { ++bad; continue; }
         if ((w.prim_idx < 0) != (b.prim_idx < 0) ||
             (b.prim_idx >= 0 && fabsf(w.t - b.t) > 1e-5f * fmaxf(1.f, b.t))) { ++bad; continue; }
         {

@@ -520,4 +520,87 @@ HD bool bvh4_walk_step(Bvh4Walk& w, const BVH4Node* nodes, const Prim* prims,
     }
 }
 
+// 4-byte-entry variant of the while-while walk: stack entries carry only the
+// tagged node/leaf word (no t_near, so no pop culling).  Halves stack LDS
+// bytes -> twice the LDS entry capacity per thread at the same block budget.
+// A/B hook HIPPT_STACK32; lds_n here counts 4-byte entries.
+HD HitRecord ray_intersect_bvh4_ww32(const BVH4Node* nodes,
+                                     const Prim* prims, const uint32_t* prim_obj,
+                                     const Ray& ray, float tmax,
+                                     uint32_t* lds_slot = nullptr, int lds_n = 0) {
+    HitRecord rec;
+    rec.t = tmax;
+    const Vec3 inv_d = safe_rcp_dir(ray.d);
+    const Vec3 o_div = ray.o * inv_d;
+    uint32_t stack[BVH4_STACK];
+    int sp = 0;
+    constexpr uint32_t DONE = 0x7fffffffu;
+    uint32_t cur = 0;
+    for (;;) {
+        while (cur < 0x80000000u && cur != DONE) {
+            const BVH4Node nd = nodes[cur];
+            uint32_t keys[4];
+            int nhit = 0;
+#pragma unroll
+            for (int c = 0; c < 4; ++c) {
+                float t0x = fmaf(nd.lo_x[c], inv_d.x, -o_div.x);
+                float t1x = fmaf(nd.hi_x[c], inv_d.x, -o_div.x);
+                float t0y = fmaf(nd.lo_y[c], inv_d.y, -o_div.y);
+                float t1y = fmaf(nd.hi_y[c], inv_d.y, -o_div.y);
+                float t0z = fmaf(nd.lo_z[c], inv_d.z, -o_div.z);
+                float t1z = fmaf(nd.hi_z[c], inv_d.z, -o_div.z);
+                float enter = fmaxf(fmaxf(fminf(t0x, t1x), fminf(t0y, t1y)),
+                                    fmaxf(fminf(t0z, t1z), 0.f));
+                float exit_ = fminf(fminf(fmaxf(t0x, t1x), fmaxf(t0y, t1y)),
+                                    fminf(fmaxf(t0z, t1z), rec.t));
+                if (enter <= exit_) keys[nhit++] = (float_as_uint(enter) & ~3u) | (uint32_t)c;
+            }
+            if (nhit > 1) {
+                if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                if (nhit > 2) {
+                    if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                    if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                    if (nhit > 3) {
+                        if (keys[2] > keys[3]) { uint32_t t = keys[2]; keys[2] = keys[3]; keys[3] = t; }
+                        if (keys[1] > keys[2]) { uint32_t t = keys[1]; keys[1] = keys[2]; keys[2] = t; }
+                        if (keys[0] > keys[1]) { uint32_t t = keys[0]; keys[0] = keys[1]; keys[1] = t; }
+                    }
+                }
+            }
+            uint32_t next = DONE;
+            for (int k = nhit - 1; k >= 0; --k) {
+                int c = (int)(keys[k] & 3u);
+                int ch = nd.child[c];
+                int pc = nd.cnt[c];
+                if (ch < 0 && pc == 0) continue;
+                uint32_t lo = ch < 0
+                    ? (0x80000000u | ((uint32_t)pc << 27) | (uint32_t)(~ch))
+                    : (uint32_t)ch;
+                if (k == 0) {
+                    next = lo;
+                } else {
+                    if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = lo;
+                    else stack[sp - lds_n] = lo;
+                    ++sp;
+                }
+            }
+            if (next != DONE) { cur = next; continue; }
+            if (sp == 0) { cur = DONE; break; }
+            --sp;
+            cur = sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE] : stack[sp - lds_n];
+        }
+        if (cur == DONE) break;
+        while (cur >= 0x80000000u) {
+            bvh4_leaf_hit(prims, prim_obj, ray, (int)(cur & 0x07ffffffu),
+                          (int)((cur >> 27) & 0xfu), rec);
+            if (sp == 0) { cur = DONE; break; }
+            --sp;
+            cur = sp < lds_n ? lds_slot[sp * BVH4_LDS_STRIDE] : stack[sp - lds_n];
+        }
+        if (cur == DONE) break;
+    }
+    if (rec.prim_idx < 0) rec.t = MAX_DIST;
+    return rec;
+}
+
 } // namespace hippt

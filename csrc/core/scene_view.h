@@ -111,6 +111,7 @@ struct TravCtx {
     int lds_n = 0;
     const BVH4Node* top_cache = nullptr;  // LDS copy of nodes4[0..n_cached)
     int n_cached = 0;
+    int stack32 = 0;   // A/B: 4-byte stack entries, no pop culling
 };
 
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
@@ -122,6 +123,9 @@ struct TravCtx {
 // cost ~3% of kernel throughput.  bvh8.h stays host-tested for the record.
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST, TravCtx tc = {}) {
+    if (tc.stack32)
+        return ray_intersect_bvh4_ww32(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
+                                       (uint32_t*)tc.lds_slot, tc.lds_n * 2);
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
 }
