@@ -111,7 +111,6 @@ struct TravCtx {
     int lds_n = 0;
     const BVH4Node* top_cache = nullptr;  // LDS copy of nodes4[0..n_cached)
     int n_cached = 0;
-    int stack32 = 0;   // A/B: 4-byte stack entries, no pop culling
 };
 
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
@@ -121,11 +120,11 @@ struct TravCtx {
 // copy = 64 VGPRs of temporaries + SAH dilution at width 8), so the 8-wide
 // walk is NOT wired into the kernels — even a dead `if (sv.nodes8)` branch
 // cost ~3% of kernel throughput.  bvh8.h stays host-tested for the record.
+// (A 4-byte-entry stack variant — double LDS capacity, no pop culling —
+// was measured 3% slower: the t_near culling pays for its 8-byte entries.
+// ray_intersect_bvh4_ww32 stays host-tested for the record.)
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST, TravCtx tc = {}) {
-    if (tc.stack32)
-        return ray_intersect_bvh4_ww32(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                                       (uint32_t*)tc.lds_slot, tc.lds_n * 2);
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
 }

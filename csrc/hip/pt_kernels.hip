@@ -42,7 +42,7 @@ __attribute__((amdgpu_waves_per_eu(MINW, MINW)))   // exact residency: the LDS
 // plain launch_bounds left the kernel at 40 VGPR + ~1 KB/lane of spills)
 void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
               int spp0, int nspp, uint32_t seed, int swiz, int lds_n, int n_cached, int y_off, int y_end,
-              const uint8_t* __restrict__ spp_map, float* __restrict__ aux, int stack32) {
+              const uint8_t* __restrict__ spp_map, float* __restrict__ aux) {
     extern __shared__ uint64_t s_stk[];
     // dynamic-LDS layout: [n_cached 128-byte nodes][per-thread stacks].
     // The top of the tree is copied into LDS once per block: every walk's
@@ -54,7 +54,7 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
     for (int i = tid; i < n_cached * 16; i += 256)
         ((uint64_t*)s_cache)[i] = ((const uint64_t*)sv.nodes4)[i];
     if (n_cached > 0) __syncthreads();
-    TravCtx tc{&s_base[tid], lds_n, s_cache, n_cached, stack32};
+    TravCtx tc{&s_base[tid], lds_n, s_cache, n_cached};
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
     if (swiz) tile = xcd_swizzle(tile, gridDim.x * gridDim.y);
     const int px = (tile % gridDim.x) * 16 + threadIdx.x;
@@ -227,10 +227,6 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }();
     const int lds_n = lds_budget > 0
         ? (lds_budget - n_cached * (int)sizeof(BVH4Node)) / (256 * 8) : 0;
-    static int s32_v = [] {
-        const char* e = getenv("HIPPT_STACK32");
-        return e ? atoi(e) : 0;
-    }();
     const uint32_t shmem = (uint32_t)(lds_n * 256 * 8 + n_cached * (int)sizeof(BVH4Node));
     const int w = sv.cam.w, h = sv.cam.h;
     if (y1 <= 0 || y1 > h) y1 = h;
@@ -250,15 +246,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     case R_VOLUME_PT:
         if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else
-            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_VOLUME_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_DEPTH:
-        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+        hipLaunchKernelGGL((k_render<R_DEPTH, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_BVH_COST:
-        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+        hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_MEGAKERNEL_PT_DYN: {
         static uint32_t* counter = nullptr;
@@ -275,15 +271,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     }
     default: {
         if (occ_v <= 3)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 3>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else if (occ_v == 4)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 4>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else if (occ_v == 5)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 5>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else if (occ_v >= 8)
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 8>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         else
-            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux, s32_v);
+            hipLaunchKernelGGL((k_render<R_MEGAKERNEL_PT, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, swiz(), lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     }
     }
