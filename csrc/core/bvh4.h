@@ -149,7 +149,7 @@ HD HitRecord ray_intersect_bvh4(const BVH4Node* nodes,
 // execute runs of same-kind work instead of interleaving a node step on
 // some lanes with prim tests on others.  Leaf children are POSTPONED onto
 // the stack (tagged entries) instead of intersected inline.  Measured A/B
-// against the inline walk via HIPPT_TRAV=ww (divergence: VALUUtilization
+// against the inline walk (divergence: VALUUtilization
 // 15% on the inline walk).
 //
 // Stack/cur entry low word: bit31 = leaf, [30:27] = prim count (builders
@@ -304,7 +304,7 @@ HD bool occlusion_test_bvh4(const BVH4Node* nodes,
 // closest-hit walk (leaf children postponed onto the stack so a wave runs
 // node steps and prim tests in separate phases), but unordered and with an
 // immediate `true` return on the first blocking hit.  A/B hook
-// HIPPT_OCC_WW selects this against the inline-leaf occlusion walk.
+// Measured +1-2% against the inline-leaf occlusion walk; now the default.
 HD bool occlusion_test_bvh4_ww(const BVH4Node* nodes,
                                const Prim* prims, const uint32_t* prim_obj,
                                const Ray& ray, float tmax,
@@ -523,7 +523,8 @@ HD bool bvh4_walk_step(Bvh4Walk& w, const BVH4Node* nodes, const Prim* prims,
 // 4-byte-entry variant of the while-while walk: stack entries carry only the
 // tagged node/leaf word (no t_near, so no pop culling).  Halves stack LDS
 // bytes -> twice the LDS entry capacity per thread at the same block budget.
-// A/B hook HIPPT_STACK32; lds_n here counts 4-byte entries.
+// Measured 3% SLOWER than the 8-byte culling walk (host-tested record;
+// no device path uses it).  lds_n here counts 4-byte entries.
 HD HitRecord ray_intersect_bvh4_ww32(const BVH4Node* nodes,
                                      const Prim* prims, const uint32_t* prim_obj,
                                      const Ray& ray, float tmax,
