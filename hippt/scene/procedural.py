@@ -161,8 +161,8 @@ def kitchen(width=1920, height=1080, spp=64, renderer="wfpt", seed=5,
         BsdfDesc(type="ggx", metal="Au", roughness_x=0.25, roughness_y=0.08),      # 9 aniso gold
     ]
     d.emitters = [
-        EmitterDesc(type="area", emission=(1.0, 0.92, 0.8), scale=40.0),
-        EmitterDesc(type="area", emission=(0.7, 0.8, 1.0), scale=25.0),
+        EmitterDesc(type="area", emission=(1.0, 0.92, 0.8), scale=22.0),
+        EmitterDesc(type="area", emission=(0.7, 0.8, 1.0), scale=14.0),
     ]
     W, H, D = 6.0, 3.0, 5.0
     room = box_mesh((0, 0, 0), (W, H, D), inward=True)
