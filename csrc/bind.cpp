@@ -42,6 +42,24 @@ using uarr = py::array_t<uint32_t, py::array::c_style | py::array::forcecast>;
 
 Vec3 to_vec3(const std::vector<float>& v) { return {v[0], v[1], v[2]}; }
 
+// Max level count of a 4-wide tree (host walk; used to gate halved-stack
+// traversal kernels like the wavefront dual walk).
+int bvh4_tree_depth(const BVH4Node* nodes, int n) {
+    if (n <= 0) return 0;
+    std::vector<std::pair<int, int>> stk{{0, 1}};
+    int maxd = 1;
+    while (!stk.empty()) {
+        auto [idx, d] = stk.back();
+        stk.pop_back();
+        maxd = std::max(maxd, d);
+        const BVH4Node& nd = nodes[idx];
+        for (int c = 0; c < 4; ++c)
+            if (nd.child[c] >= 0 && !(nd.child[c] == 0 && nd.cnt[c] == 0))
+                stk.push_back({nd.child[c], d + 1});
+    }
+    return maxd;
+}
+
 struct SceneHolder {
     // ----- host data (kept alive / owned)
     farr np_prims, np_attrs, np_nodes, np_nodes4, np_nodes8;
@@ -103,7 +121,12 @@ struct SceneHolder {
         if (nodes8.ndim() == 2 && nodes8.shape(0) > 0 && nodes8.shape(1) != 64)
             throw std::runtime_error("nodes8 must be (m,64)");
         np_nodes8 = std::move(nodes8);
+        bvh4_depth = np_nodes4.ndim() == 2 && np_nodes4.shape(0) > 0
+            ? bvh4_tree_depth((const BVH4Node*)np_nodes4.data(), (int)np_nodes4.shape(0))
+            : 0;
     }
+
+    int bvh4_depth = 0;
 
     void set_objects(iarr objs) {
         if (objs.ndim() != 2 || objs.shape(1) != 8) throw std::runtime_error("objs must be (n,8)");
@@ -288,6 +311,7 @@ struct SceneHolder {
         sv.cam_medium = cam_medium;
         sv.cam = cam;
         sv.md = md;
+        sv.bvh4_depth = bvh4_depth;
     }
 
     void finalize() {

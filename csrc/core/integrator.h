@@ -72,10 +72,13 @@ HD Vec3 bsdf_aov_albedo(const BsdfParams& b, Vec2 uv, const TexView* textures) {
     }
 }
 
-// One bounce; returns true when the path is finished (L is final).
-HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
-    if (ps.iter >= sv.md.max_depth * 2 + 8 || ps.b >= sv.md.max_depth) return true;
-    ++ps.iter;
+// Shade one already-intersected hit (miss handling included): emitter-hit
+// MIS, NEE, BSDF sampling, caps + RR.  Split out of path_step so the
+// wavefront tail kernel (wf_kernels.hip k_wf_tail) can resume a path whose
+// current hit record is already in the payload pool.  Returns true when the
+// path is finished (L is final).
+HD bool path_shade_hit(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc,
+                       const HitRecord& hit) {
     Ray& ray = ps.ray;
     Vec3& L = ps.L;
     Vec3& thp = ps.thp;
@@ -86,7 +89,6 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
     PathStats& st = ps.st;
     int& b = ps.b;
     {
-        HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
         if (hit.prim_idx < 0) {
             // miss -> environment map with MIS against the cosine NEE pdf
             if (sv.env_emitter >= 0) {
@@ -185,6 +187,14 @@ HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
         }
     }
     return false;
+}
+
+// One bounce; returns true when the path is finished (L is final).
+HD bool path_step(const SceneView& sv, PathState& ps, Sampler& sp, TravCtx tc) {
+    if (ps.iter >= sv.md.max_depth * 2 + 8 || ps.b >= sv.md.max_depth) return true;
+    ++ps.iter;
+    HitRecord hit = scene_intersect(sv, ps.ray, MAX_DIST, tc);
+    return path_shade_hit(sv, ps, sp, tc, hit);
 }
 
 // Full path trace for one camera ray. Returns radiance estimate.

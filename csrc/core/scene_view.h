@@ -67,6 +67,9 @@ struct SceneView {
     // camera + depth caps
     Camera cam;
     MaxDepthParams md;
+    // max depth of the 4-wide tree (levels); gates kernels with halved
+    // traversal stacks (wavefront dual-walk trace)
+    int bvh4_depth;
 
     HD EmitterGeom emitter_geom() const {
         return {prims, attrs, prim_obj, emitter_prims, emitter_cdf, textures,

@@ -58,7 +58,7 @@ void render_cpu(const SceneView& sv, float* accum, float* var,
                     PathState ps;
                     ps.reset(ray);
                     while (!path_step(sv, ps, sp, TravCtx{})) {}
-                    L = ps.L.has_nan() ? Vec3(0.f) : ps.L;
+                    L = ps.L.has_nan() ? Vec3(0.f) : clamp_radiance(sv, ps.L);
                     an += ps.aov_n; aa += ps.aov_alb; at += ps.aov_t;
                 } else                  L = trace_path(sv, ray, sp);
                 Lsum += L;

@@ -130,7 +130,22 @@ void build_rec(const Ctx& ctx, std::vector<Refr>&& refs, int depth, BuildOut& ou
         out.leaf_refs.insert(out.leaf_refs.end(), refs.begin(), refs.end());
         ++out.n_leaves;
     };
-    if (n <= 1 || depth > 60) { make_leaf(); return; }
+    // Forced median split for every fallback: the ww traversal packs the
+    // leaf prim count into 4 bits (bvh4.h, cnt<<27), so a leaf may hold at
+    // most 15 refs.  A median split always halves n, so recursion from any
+    // fallback site terminates with all leaves <= 15.
+    auto median_split = [&] {
+        std::vector<Refr> l(refs.begin(), refs.begin() + n / 2);
+        std::vector<Refr> r(refs.begin() + n / 2, refs.end());
+        refs.clear(); refs.shrink_to_fit();
+        build_rec(ctx, std::move(l), depth + 1, out);
+        int li = my + 1;
+        int ri = (int)out.nodes.size();
+        build_rec(ctx, std::move(r), depth + 1, out);
+        out.nodes[my].left = li; out.nodes[my].right = ri;
+    };
+    if (n <= 1 || (depth > 60 && n <= 15)) { make_leaf(); return; }
+    if (depth > 60) { median_split(); return; }
 
     const float leaf_cost = box.area() * n;
 
@@ -217,15 +232,7 @@ void build_rec(const Ctx& ctx, std::vector<Refr>&& refs, int depth, BuildOut& ou
     bool use_spatial = best_sp_axis >= 0 && best_sp_cost < best_obj_cost;
     if (!use_spatial && best_obj_axis < 0) {
         if (n <= 15) { make_leaf(); return; }  // leaf cap: ww walk packs cnt in 4 bits
-        // degenerate: median split
-        std::vector<Refr> l(refs.begin(), refs.begin() + n / 2);
-        std::vector<Refr> r(refs.begin() + n / 2, refs.end());
-        refs.clear(); refs.shrink_to_fit();
-        build_rec(ctx, std::move(l), depth + 1, out);
-        int li = my + 1;
-        int ri = (int)out.nodes.size();
-        build_rec(ctx, std::move(r), depth + 1, out);
-        out.nodes[my].left = li; out.nodes[my].right = ri;
+        median_split();
         return;
     }
     if (n <= ctx.max_leaf &&
@@ -278,7 +285,10 @@ void build_rec(const Ctx& ctx, std::vector<Refr>&& refs, int depth, BuildOut& ou
             (b <= best_obj_bin ? lrefs : rrefs).push_back(r);
         }
     }
-    if (lrefs.empty() || rrefs.empty()) { make_leaf(); return; }
+    if (lrefs.empty() || rrefs.empty()) {
+        if (n <= 15) make_leaf(); else median_split();
+        return;
+    }
     refs.clear(); refs.shrink_to_fit();
 
     int li, ri;

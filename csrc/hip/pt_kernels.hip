@@ -257,7 +257,15 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         hipLaunchKernelGGL((k_render<R_BVH_COST, 6>), grid, block, shmem, st, sv, accum, var, spp0, nspp, seed, 0, lds_n, n_cached, y0, y1, spp_map, aux);
         break;
     case R_MEGAKERNEL_PT_DYN: {
-        static uint32_t* counter = nullptr;
+        // per-device work counters, freed at process exit (one process may
+        // drive several devices: one slot per device id)
+        static struct Counters {
+            uint32_t* p[64] = {};
+            ~Counters() { for (uint32_t* q : p) if (q) (void)hipFree(q); }
+        } counters;
+        int dev = 0;
+        (void)hipGetDevice(&dev);
+        uint32_t*& counter = counters.p[dev & 63];
         if (!counter) {
             if (hipMalloc((void**)&counter, 4) != hipSuccess) return (int)hipGetLastError();
         }

@@ -13,8 +13,19 @@
 //
 // MI355X-native design: the sort is a 3-kernel one-byte counting sort
 // (per-block LDS histograms -> single-block scan -> scatter), entirely
-// memory-bound, with the live count produced as a by-product on-device and
-// read back once per bounce (4 bytes) instead of a device-wide binary search.
+// memory-bound.  The live count is produced on-device by the scan and read
+// back once per bounce (4 bytes, like the reference's post-sort
+// lower_bound readback, wf_path_tracer.cu:199) so that
+//   * the bounce loop BREAKS the moment live == 0,
+//   * shade/shadow/trace launch live-sized grids instead of full-width
+//     ones, and
+//   * from bounce 1 on, the sort passes GATHER through the previous
+//     bounce's compacted order array — their cost scales with the live
+//     count, not with W*H.
+// Below HIPPT_WF_TAIL live rays (default 128K: the point where 256 CUs run
+// out of resident waves anyway), one fused tail kernel finishes every
+// surviving path megakernel-style (integrator.h path_shade_hit/path_step),
+// eliminating all remaining per-bounce sorts and launches.
 #include <hip/hip_runtime.h>
 #include "kernels.h"
 #include <cstdlib>
@@ -40,6 +51,7 @@ struct WfState {
     unsigned long long* rng = nullptr;
     uint32_t* status = nullptr;  // (status<<24)|payload, PIXEL order (trace input)
     uint32_t* order = nullptr;   // compacted material-sorted view (shade input)
+    uint32_t* order2 = nullptr;  // double buffer: bounce b gathers via b-1's view
     uint32_t* hist = nullptr;    // nb_sort * 256
     int* live_dev = nullptr;
     int* live_host = nullptr;    // pinned
@@ -102,15 +114,24 @@ __device__ inline uint32_t sort_key(uint32_t status, int mode) {
     return mode ? (b >= DEAD ? (uint32_t)DEAD : 0u) : b;
 }
 
+// `gather` non-null: the pass runs over the previous bounce's compacted
+// order view (n = prev live count) and re-reads each entry's FRESH status
+// byte from the pixel-order status array — so from bounce 1 on, sort cost
+// scales with the live count, not W*H.
+__device__ __forceinline__ uint32_t sort_load(const uint32_t* in, const uint32_t* gather, int i) {
+    return gather ? in[gather[i] & 0x00FFFFFFu] : in[i];
+}
+
 __global__ __launch_bounds__(SORT_BLOCK)
-void k_sort_hist(const uint32_t* __restrict__ in, int n, uint32_t* __restrict__ hist, int mode) {
+void k_sort_hist(const uint32_t* __restrict__ in, const uint32_t* __restrict__ gather,
+                 int n, uint32_t* __restrict__ hist, int mode) {
     __shared__ uint32_t lh[256];
     for (int t = threadIdx.x; t < 256; t += blockDim.x) lh[t] = 0;
     __syncthreads();
     int base = blockIdx.x * SORT_BLOCK * SORT_ITEMS;
     for (int k = 0; k < SORT_ITEMS; ++k) {
         int i = base + k * SORT_BLOCK + threadIdx.x;
-        if (i < n) atomicAdd(&lh[sort_key(in[i], mode)], 1u);
+        if (i < n) atomicAdd(&lh[sort_key(sort_load(in, gather, i), mode)], 1u);
     }
     __syncthreads();
     for (int t = threadIdx.x; t < 256; t += blockDim.x)
@@ -137,8 +158,8 @@ void k_sort_scan(uint32_t* __restrict__ hist, int* __restrict__ live_out) {
 }
 
 __global__ __launch_bounds__(SORT_BLOCK)
-void k_sort_scatter(const uint32_t* __restrict__ in, int n,
-                    uint32_t* __restrict__ hist, uint32_t* __restrict__ out, int mode) {
+void k_sort_scatter(const uint32_t* __restrict__ in, const uint32_t* __restrict__ gather,
+                    int n, uint32_t* __restrict__ hist, uint32_t* __restrict__ out, int mode) {
     __shared__ uint32_t lbase[256];
     __shared__ uint32_t lcnt[256];
     for (int t = threadIdx.x; t < 256; t += blockDim.x) lcnt[t] = 0;
@@ -150,7 +171,7 @@ void k_sort_scatter(const uint32_t* __restrict__ in, int n,
     for (int k = 0; k < SORT_ITEMS; ++k) {
         int i = base + k * SORT_BLOCK + threadIdx.x;
         if (i < n) {
-            ent[k] = in[i];
+            ent[k] = sort_load(in, gather, i);
             rank[k] = atomicAdd(&lcnt[sort_key(ent[k], mode)], 1u);
         }
     }
@@ -158,9 +179,14 @@ void k_sort_scatter(const uint32_t* __restrict__ in, int n,
     for (int t = threadIdx.x; t < 256; t += blockDim.x)
         lbase[t] = lcnt[t] ? atomicAdd(&hist[t], lcnt[t]) : 0;
     __syncthreads();
+    // dead entries are dropped (nothing downstream reads past the live
+    // prefix), so the compacted view holds live rays only
     for (int k = 0; k < SORT_ITEMS; ++k) {
         int i = base + k * SORT_BLOCK + threadIdx.x;
-        if (i < n) out[lbase[sort_key(ent[k], mode)] + rank[k]] = ent[k];
+        if (i < n) {
+            uint32_t key = sort_key(ent[k], mode);
+            if (key < DEAD) out[lbase[key] + rank[k]] = ent[k];
+        }
     }
 }
 
@@ -168,9 +194,9 @@ void k_sort_scatter(const uint32_t* __restrict__ in, int n,
 // NEE + emitter-hit MIS + BSDF sample for live rays (current hit record).
 __global__ __launch_bounds__(256)
 void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
-                int bounce) {
+                int live, int bounce) {
     int k = blockIdx.x * blockDim.x + threadIdx.x;
-    if (k >= *st.live_dev) return;
+    if (k >= live) return;
     uint32_t entry = order[k];
     int i = (int)(entry & 0x00FFFFFFu);
     float4 h4 = st.hit[i];
@@ -339,17 +365,21 @@ __device__ inline void wf_trace_finish(const SceneView& sv, WfState& st, int i,
     st.status[i] = (status << 24) | (uint32_t)i;
 }
 
+// Live-prefix trace: scans the compacted order view (grid sized from the
+// live count).  Entries within a material bin keep block-local pixel
+// adjacency from the scatter, so wave-level spatial coherence survives the
+// material grouping.  Shade may have terminated a ray after the sort ran,
+// so the fresh status byte is still checked.
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
-void k_wf_trace(SceneView sv, WfState st, int lds_n) {
+void k_wf_trace(SceneView sv, WfState st, const uint32_t* __restrict__ order,
+                int live, int lds_n) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc{&s_stk[threadIdx.x], lds_n};
-    // pixel-order scan: bounce rays of neighboring pixels traverse similar
-    // BVH subtrees, so keeping trace in payload order (NOT material-sorted
-    // order) preserves wave-level spatial coherence.
-    int i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= st.n) return;
-    if (st.status[i] >> 24 >= DEAD) return;  // dead or terminated in shade
+    int k = blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= live) return;
+    int i = (int)(order[k] & 0x00FFFFFFu);
+    if (st.status[i] >> 24 >= DEAD) return;  // terminated in shade
     float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i];
     Ray ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
     HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
@@ -358,21 +388,26 @@ void k_wf_trace(SceneView sv, WfState st, int lds_n) {
 
 // Dual-ray trace: each lane advances two independent walks in lockstep
 // steps, keeping two node loads in flight per lane (the single walk stalls
-// ~58% of cycles on L2-hit latency).  Rays i and i + n/2 pair up so both
-// halves stay in pixel order.  HIPPT_WF_DUAL selects this kernel.
+// ~58% of cycles on L2-hit latency).  Order entries k and k + live/2 pair
+// up so both halves stay in compaction order.  HIPPT_WF_DUAL selects this
+// kernel; the launcher refuses it unless 3*bvh4_depth fits the halved
+// stack (BVH4_STACK/2 private + lds_n/2 LDS entries per walk).
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(4, 4)))   // 2x walk state needs registers
-void k_wf_trace_dual(SceneView sv, WfState st, int lds_n) {
+void k_wf_trace_dual(SceneView sv, WfState st, const uint32_t* __restrict__ order,
+                     int live, int lds_n) {
     extern __shared__ uint64_t s_stk[];
     const int tid = threadIdx.x;
-    const int half = (st.n + 1) >> 1;
-    const int i0 = blockIdx.x * blockDim.x + tid;
-    const int i1 = i0 + half;
+    const int half = (live + 1) >> 1;
+    const int k0 = blockIdx.x * blockDim.x + tid;
+    const int k1 = k0 + half;
     const int lds_half = lds_n >> 1;
     uint64_t* slot0 = &s_stk[tid];
     uint64_t* slot1 = &s_stk[tid + (size_t)lds_half * BVH4_LDS_STRIDE];
-    bool a0 = i0 < half && (st.status[i0] >> 24) < DEAD;
-    bool a1 = i1 < st.n && (st.status[i1] >> 24) < DEAD;
+    const int i0 = k0 < half ? (int)(order[k0] & 0x00FFFFFFu) : 0;
+    const int i1 = k1 < live ? (int)(order[k1] & 0x00FFFFFFu) : 0;
+    bool a0 = k0 < half && (st.status[i0] >> 24) < DEAD;
+    bool a1 = k1 < live && (st.status[i1] >> 24) < DEAD;
     if (!a0 && !a1) return;
     Ray ray0, ray1;
     float pdf0 = 0.f, pdf1 = 0.f;
@@ -396,6 +431,52 @@ void k_wf_trace_dual(SceneView sv, WfState st, int lds_n) {
     }
     if (a0) wf_trace_finish(sv, st, i0, ray0, pdf0, w0.rec);
     if (a1) wf_trace_finish(sv, st, i1, ray1, pdf1, w1.rec);
+}
+
+// ------------------------------------------------------------- tail fuse
+// Below HIPPT_WF_TAIL live rays the chip is under-occupied anyway, so
+// per-bounce sorting + 4 kernel launches cost more than the wave64
+// divergence they avoid.  This kernel finishes every surviving path
+// megakernel-style: shade the stored hit (integrator.h path_shade_hit),
+// then loop full path_step bounces (inline NEE occlusion, no shadow
+// queue) until the path dies.  Mirrors the reference's early loop exit at
+// live == 0 (wf_path_tracer.cu:199-210) taken one step further.
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
+void k_wf_tail(SceneView sv, WfState st, const uint32_t* __restrict__ order,
+               int live, int lds_n) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc{&s_stk[threadIdx.x], lds_n};
+    int k = blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= live) return;
+    int i = (int)(order[k] & 0x00FFFFFFu);
+    float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i], thp4 = st.thp[i], l4 = st.L[i];
+    float4 h4 = st.hit[i], pn4 = st.prevn[i];
+    PathState ps;
+    ps.ray = Ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
+    ps.L = Vec3(l4.x, l4.y, l4.z);
+    ps.thp = Vec3(thp4.x, thp4.y, thp4.z);
+    ps.prev_pdf = rd4.w;
+    ps.prev_delta = (float_as_uint(thp4.w) & 1u) != 0;
+    ps.prev_n = Vec3(pn4.x, pn4.y, pn4.z);
+    ps.path_time = 0.f;
+    uint32_t counts = float_as_uint(pn4.w);
+    ps.st.n_diffuse = (int)(counts & 0xFF);
+    ps.st.n_specular = (int)((counts >> 8) & 0xFF);
+    ps.st.n_transmit = (int)((counts >> 16) & 0xFF);
+    ps.st.n_volume = 0;
+    ps.b = (int)((counts >> 24) & 0xFF);
+    ps.iter = ps.b + 1;  // null bounces before the handoff are not replayed
+    ps.aov_n = Vec3(0.f); ps.aov_alb = Vec3(0.f); ps.aov_t = 0.f;
+    ps.lambda = ro4.w;
+    Sampler sp(st.rng[i]);
+    HitRecord hit;
+    hit.t = h4.x; hit.u = h4.y; hit.v = h4.z;
+    hit.prim_idx = float_as_int(h4.w);
+    bool done = path_shade_hit(sv, ps, sp, tc, hit);
+    while (!done) done = path_step(sv, ps, sp, tc);
+    st.L[i] = make_float4(ps.L.x, ps.L.y, ps.L.z, l4.w);
+    st.status[i] = (DEAD << 24) | (uint32_t)i;
 }
 
 // ----------------------------------------------------------------- splat
@@ -435,6 +516,7 @@ WfState* wf_create(int width, int height) {
     e |= wf_alloc(&s->rng, s->n);
     e |= wf_alloc(&s->status, s->n);
     e |= wf_alloc(&s->order, s->n);
+    e |= wf_alloc(&s->order2, s->n);
     e |= wf_alloc(&s->hist, (size_t)s->nb_sort * 256);
     e |= wf_alloc(&s->live_dev, 1);
     e |= wf_alloc(&s->sh_od, s->n);
@@ -450,7 +532,8 @@ void wf_destroy(WfState* s) {
     if (!s) return;
     (void)hipFree(s->ray_o); (void)hipFree(s->ray_d); (void)hipFree(s->thp); (void)hipFree(s->L);
     (void)hipFree(s->hit); (void)hipFree(s->prevn); (void)hipFree(s->rng);
-    (void)hipFree(s->status); (void)hipFree(s->order); (void)hipFree(s->hist);
+    (void)hipFree(s->status); (void)hipFree(s->order); (void)hipFree(s->order2);
+    (void)hipFree(s->hist);
     (void)hipFree(s->live_dev);
     (void)hipFree(s->sh_od); (void)hipFree(s->sh_dir); (void)hipFree(s->sh_val); (void)hipFree(s->sh_cnt);
     if (s->live_host) (void)hipHostFree(s->live_host);
@@ -484,6 +567,10 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
                 hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
             (void)hipFuncSetAttribute((const void*)&k_wf_shadow,
                 hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
+            (void)hipFuncSetAttribute((const void*)&k_wf_trace_dual,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
+            (void)hipFuncSetAttribute((const void*)&k_wf_tail,
+                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
         }
         return v;
     }();
@@ -496,31 +583,63 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         const char* e = getenv("HIPPT_WF_DUAL");
         return e ? atoi(e) : 0;
     }();
+    // dual-walk stack safety (each walk: BVH4_STACK/2 private + lds_n/2 LDS
+    // entries; a node visit pushes at most 3 entries per level)
+    if (wf_dual && 3 * sv.bvh4_depth > BVH4_STACK / 2 + lds_n / 2) wf_dual = 0;
+    static int tail_thresh = [] {
+        const char* e = getenv("HIPPT_WF_TAIL");
+        return e ? atoi(e) : 128 * 1024;
+    }();
     for (int s = 0; s < nspp; ++s) {
         hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
-        const int nb = st->nb_sort;
+        // Per bounce: build the compacted material-sorted live view, read the
+        // live count back (4 bytes — the price the reference also pays,
+        // wf_path_tracer.cu:199), then launch live-sized grids.  Bounce 0
+        // sorts the full pixel-order status array; later bounces gather
+        // through the previous view, so the sort scales with live too.
+        const uint32_t* gather = nullptr;
+        int prev_live = n;
+        uint32_t* order_cur = st->order;
+        uint32_t* order_prev = st->order2;
         for (int bounce = 0; bounce < sv.md.max_depth + 1; ++bounce) {
-            // build the compacted material-sorted view over the FULL status
-            // array (pixel order preserved there for the trace kernel);
-            // the live count never leaves the device: shade/shadow/trace are
-            // launched full-width and exit by comparing against *live_dev.
+            const int scan_n = prev_live;
+            const int nb = (scan_n + SORT_BLOCK * SORT_ITEMS - 1) / (SORT_BLOCK * SORT_ITEMS);
             (void)hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
             hipLaunchKernelGGL(k_sort_hist, dim3(nb), dim3(SORT_BLOCK), 0, hs,
-                               st->status, n, st->hist, sort_mode);
+                               st->status, gather, scan_n, st->hist, sort_mode);
             hipLaunchKernelGGL(k_sort_scan, dim3(1), dim3(256), 0, hs,
                                st->hist, st->live_dev);
             hipLaunchKernelGGL(k_sort_scatter, dim3(nb), dim3(SORT_BLOCK), 0, hs,
-                               st->status, n, st->hist, st->order, sort_mode);
-            (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
-            hipLaunchKernelGGL(k_wf_shade, grd_n, blk, 0, hs, sv, *st, st->order,
-                               bounce);
-            hipLaunchKernelGGL(k_wf_shadow, grd_n, blk, shmem, hs, sv, *st, lds_n);
-            if (wf_dual) {
-                dim3 grd_h(((st->n + 1) / 2 + WF_BLOCK - 1) / WF_BLOCK);
-                hipLaunchKernelGGL(k_wf_trace_dual, grd_h, blk, shmem, hs, sv, *st, lds_n);
-            } else {
-                hipLaunchKernelGGL(k_wf_trace, grd_n, blk, shmem, hs, sv, *st, lds_n);
+                               st->status, gather, scan_n, st->hist, order_cur, sort_mode);
+            (void)hipMemcpyAsync(st->live_host, st->live_dev, sizeof(int),
+                                 hipMemcpyDeviceToHost, hs);
+            int err = (int)hipStreamSynchronize(hs);
+            if (err) return err;
+            const int live = *st->live_host;
+            if (live == 0) break;                     // reference: break at live==0
+            dim3 grd_live((live + WF_BLOCK - 1) / WF_BLOCK);
+            if (live <= tail_thresh) {
+                hipLaunchKernelGGL(k_wf_tail, grd_live, blk, shmem, hs, sv, *st,
+                                   order_cur, live, lds_n);
+                break;
             }
+            (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
+            hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, order_cur,
+                               live, bounce);
+            hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, shmem, hs, sv, *st, lds_n);
+            if (bounce < sv.md.max_depth) {           // reference skips last-bounce trace
+                if (wf_dual) {
+                    dim3 grd_h(((live + 1) / 2 + WF_BLOCK - 1) / WF_BLOCK);
+                    hipLaunchKernelGGL(k_wf_trace_dual, grd_h, blk, shmem, hs, sv, *st,
+                                       order_cur, live, lds_n);
+                } else {
+                    hipLaunchKernelGGL(k_wf_trace, grd_live, blk, shmem, hs, sv, *st,
+                                       order_cur, live, lds_n);
+                }
+            }
+            gather = order_cur;
+            prev_live = live;
+            uint32_t* t = order_cur; order_cur = order_prev; order_prev = t;
         }
         hipLaunchKernelGGL(k_wf_splat, grd_n, blk, 0, hs, sv, *st, accum, var, 1);
     }

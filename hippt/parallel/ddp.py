@@ -8,10 +8,12 @@ accumulator (ddp_render.py:70-81), frame-time all_gather for logging
 (:192-211), SIGINT graceful shutdown (:51-57).
 
 MI355X-native notes: the all-reduce runs on our own accumulation tensors
-(which are sums, so SUM-reduce is exact, no spp weighting error), on a side
-stream so the next frame's megakernel overlaps the collective; xGMI ring
-all-reduce of a 1080p fp32 accumulator (~33 MB) costs ~1 ms/GPU-pair and is
-negligible at any reasonable reduce cadence.
+(which are sums, so SUM-reduce is exact, no spp weighting error).  The
+collective is issued with async_op=True — RCCL runs it on its internal
+comm stream — and is only waited on AFTER the next step's render kernels
+have been launched on the compute stream, so the xGMI transfer overlaps the
+next frame's megakernel (start_reduce/finish_reduce below; an all-reduce of
+a 1080p fp32 accumulator is ~33 MB and costs ~1 ms/GPU-pair on xGMI).
 """
 from __future__ import annotations
 
@@ -78,18 +80,33 @@ def load_scene(args):
     return desc
 
 
-def reduce_rendered_image(dist, rend, world_size, cpu=False):
-    """spp-weighted all-reduce (ddp_render.py:70-81).  Our accumulators are
-    radiance SUMS with the sample count in alpha, so a plain SUM all-reduce is
-    exactly the spp-weighted average; returns the merged mean image."""
+def start_reduce(dist, rend, cpu=False):
+    """Snapshot the accumulator on the compute stream and launch the SUM
+    all-reduce asynchronously (RCCL comm stream).  The returned work handle
+    is waited on by finish_reduce — after the caller has already queued the
+    next frame's kernels, so the xGMI collective overlaps them."""
     import torch
     t = rend.accum if not cpu else torch.from_numpy(rend.accum.copy())
     merged = t.clone()
-    dist.all_reduce(merged, op=dist.ReduceOp.SUM)
+    work = dist.all_reduce(merged, op=dist.ReduceOp.SUM, async_op=True)
+    return work, merged
+
+
+def finish_reduce(work, merged):
+    work.wait()
     cnt = merged[:, :, 3:4].clamp(min=1e-9)
     out = merged.clone()
     out[:, :, :3] /= cnt
     return out, float(merged[0, 0, 3])
+
+
+def reduce_rendered_image(dist, rend, world_size, cpu=False):
+    """Synchronous spp-weighted all-reduce (ddp_render.py:70-81).  Our
+    accumulators are radiance SUMS with the sample count in alpha, so a plain
+    SUM all-reduce is exactly the spp-weighted average; returns the merged
+    mean image.  The main loop uses the split start_reduce/finish_reduce pair
+    for comm/compute overlap; this wrapper serves tests and one-shot merges."""
+    return finish_reduce(*start_reduce(dist, rend, cpu=cpu))
 
 
 def main(argv=None):
@@ -120,8 +137,14 @@ def main(argv=None):
     rend = r.renderer
 
     ckpt_path = f"{args.checkpoint}.rank{rank}.npz" if args.checkpoint else None
+    ckpt_step = None
     if ckpt_path and os.path.exists(ckpt_path):
-        r.load_state(ckpt_path)
+        extra = r.load_state(ckpt_path)
+        # the step index is persisted explicitly (deriving it from accum_cnt
+        # over-counts under --adaptive, whose per-call spp can exceed
+        # spp_per_call)
+        if extra and "step" in extra:
+            ckpt_step = int(extra["step"])
         if rank == 0:
             print(f"[ddp] resumed from {ckpt_path} at {r.counter()} spp", flush=True)
 
@@ -153,31 +176,50 @@ def main(argv=None):
     steps = (spp_per_rank + args.spp_per_call - 1) // args.spp_per_call
     t_start = time.perf_counter()
     merged = None
-    start_step = min(rend.accum_cnt // args.spp_per_call, steps)
+    pending = None  # (work, merged_tensor, step_idx) of an in-flight reduce
+    if ckpt_step is not None:
+        start_step = min(ckpt_step, steps)
+    else:
+        start_step = min(rend.accum_cnt // args.spp_per_call, steps)
+
+    def _finish_pending(pending):
+        """Complete an async reduce (after the NEXT step's kernels were
+        launched — this is where the xGMI collective overlaps compute) and
+        do the frame-time all_gather + logging (ddp_render.py:192-211)."""
+        work, m, kidx = pending
+        merged, total_spp = finish_reduce(work, m)
+        ft = torch.tensor([rend.avg_frame_time()])
+        fts = [torch.zeros_like(ft) for _ in range(world_size)]
+        dist.all_gather(fts, ft)
+        if rank == 0:
+            times = [float(t.item()) for t in fts]
+            print(f"[ddp] step {kidx+1}/{steps} total_spp={total_spp:.0f} "
+                  f"frame_ms={times} avg={np.mean(times):.1f}", flush=True)
+            if writer is not None:
+                img = (merged[:, :, :3].clamp(min=0) ** (1 / 2.1)).clamp(max=1)
+                writer.add_image("render", img.permute(2, 0, 1).cpu(), kidx)
+                writer.add_scalar("frame_ms/avg", float(np.mean(times)), kidx)
+        return merged
+
     for k in range(start_step, steps):
         if args.adaptive and rend.accum_cnt >= args.spp_per_call and rend.rid in (0, 2):
             m = rend._spp_budget(args.spp_per_call)
             rend.render(args.spp_per_call, y0=band[0], y1=band[1], spp_map=m)
         else:
             rend.render(args.spp_per_call, y0=band[0], y1=band[1])
+        if pending is not None:
+            # this step's kernels are queued; now drain last step's collective
+            merged = _finish_pending(pending)
+            pending = None
         if ckpt_path and (k + 1) % args.checkpoint_interval == 0:
-            r.save_state(ckpt_path)
+            r.save_state(ckpt_path, step=np.int64(k + 1))
         if world_size > 1 and ((k + 1) % args.reduce_interval == 0 or k == steps - 1):
-            merged, total_spp = reduce_rendered_image(dist, rend, world_size, cpu=args.cpu)
-            # frame-time all_gather (ddp_render.py:192-211)
-            ft = torch.tensor([rend.avg_frame_time()])
-            fts = [torch.zeros_like(ft) for _ in range(world_size)]
-            dist.all_gather(fts, ft)
-            if rank == 0:
-                times = [float(t.item()) for t in fts]
-                print(f"[ddp] step {k+1}/{steps} total_spp={total_spp:.0f} "
-                      f"frame_ms={times} avg={np.mean(times):.1f}", flush=True)
-                if writer is not None:
-                    img = (merged[:, :, :3].clamp(min=0) ** (1 / 2.1)).clamp(max=1)
-                    writer.add_image("render", img.permute(2, 0, 1).cpu(), k)
-                    writer.add_scalar("frame_ms/avg", float(np.mean(times)), k)
+            pending = (*start_reduce(dist, rend, cpu=args.cpu), k)
         if stop["flag"]:
             break
+    if pending is not None:
+        merged = _finish_pending(pending)
+        pending = None
 
     if world_size > 1:
         dist.barrier()

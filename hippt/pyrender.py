@@ -102,22 +102,24 @@ class PythonRenderer:
         returns an (H, W, 3) tensor/array on the render device."""
         return self.renderer.denoise(iterations=iterations, **kw)
 
-    def save_state(self, path: str):
+    def save_state(self, path: str, **extra):
         """Checkpoint the warm accumulation state (radiance sums, variance
         sums, sample counter) so a long accumulation can resume after a
         restart.  The reference keeps this state implicit and unpersisted
         (tracer_base.cuh:135-158 accum buffer + accum_cnt); here it is a
-        first-class .npz snapshot."""
+        first-class .npz snapshot.  Extra scalar/array keys (e.g. the driver
+        loop's step index) ride along and come back from load_state()."""
         self._check()
         import numpy as np
         r = self.renderer
         accum = r.accum.cpu().numpy() if r.device is not None else r.accum
         var = r.var.cpu().numpy() if r.device is not None else r.var
         np.savez_compressed(path, accum=accum, var=var,
-                            accum_cnt=np.int64(r.accum_cnt))
+                            accum_cnt=np.int64(r.accum_cnt), **extra)
 
     def load_state(self, path: str):
-        """Resume from a save_state() snapshot (shape-checked)."""
+        """Resume from a save_state() snapshot (shape-checked).  Returns the
+        dict of extra keys stored by save_state (empty if none)."""
         self._check()
         import numpy as np
         z = np.load(path)
@@ -133,6 +135,7 @@ class PythonRenderer:
             r.accum[:] = z["accum"]
             r.var[:] = z["var"]
         r.accum_cnt = int(z["accum_cnt"])
+        return {k: z[k] for k in z.files if k not in ("accum", "var", "accum_cnt")}
 
     def _check(self):
         if self._released:
