@@ -124,9 +124,30 @@ struct SceneHolder {
         bvh4_depth = np_nodes4.ndim() == 2 && np_nodes4.shape(0) > 0
             ? bvh4_tree_depth((const BVH4Node*)np_nodes4.data(), (int)np_nodes4.shape(0))
             : 0;
+        // scene bounding sphere (envmap sample_le disk origin)
+        const Prim* pr = (const Prim*)np_prims.data();
+        const uint32_t* po = np_prim_obj.data();
+        int n = (int)np_prims.shape(0);
+        AABB sb;
+        for (int i = 0; i < n; ++i) {
+            if (po[i] & PRIM_SPHERE_BIT) {
+                Vec3 c = pr[i].v0.xyz();
+                float rad = pr[i].v0.w;
+                sb.grow(c - Vec3(rad)); sb.grow(c + Vec3(rad));
+            } else {
+                Vec3 v0 = pr[i].v0.xyz();
+                sb.grow(v0);
+                sb.grow(v0 + pr[i].e1.xyz());
+                sb.grow(v0 + pr[i].e2.xyz());
+            }
+        }
+        Vec3 c = (sb.lo + sb.hi) * 0.5f;
+        float rad = n > 0 ? (sb.hi - sb.lo).length() * 0.5f : 1.f;
+        scene_bound = Vec4(c, rad);
     }
 
     int bvh4_depth = 0;
+    Vec4 scene_bound{0.f, 0.f, 0.f, 1.f};
 
     void set_objects(iarr objs) {
         if (objs.ndim() != 2 || objs.shape(1) != 8) throw std::runtime_error("objs must be (n,8)");
@@ -312,6 +333,7 @@ struct SceneHolder {
         sv.cam = cam;
         sv.md = md;
         sv.bvh4_depth = bvh4_depth;
+        sv.scene_bound = scene_bound;
     }
 
     void finalize() {

@@ -58,6 +58,9 @@ struct EmitterGeom {
     const float* env_rows;        // (h) cumulative
     const float* env_cols;        // (h*w) cumulative per row
     int env_w, env_h;
+    // scene bounding sphere (envmap light-path emission: disk sampling)
+    Vec3 scene_center;
+    float scene_radius;
 };
 
 // first index with cdf[i] > u (cdf ascending, cdf[n-1] == 1)
@@ -101,6 +104,34 @@ HD Vec3 envmap_eval(const EmitterParams& e, const Vec3& dir, const TexView* text
     Vec3 rad = e.emission.xyz() * e.emission.w;
     if (e.tex_id >= 0) rad *= textures[e.tex_id].sample(dir_to_latlong(d)).xyz();
     return rad;
+}
+
+// Importance-sample a world direction TOWARD the environment from its
+// luminance CDF; solid-angle pdf out.  Shared by NEE (emitter_sample) and
+// light-path emission (emitter_sample_le).  Requires g.env_rows + texture.
+HD Vec3 envmap_sample_dir(const EmitterParams& e, const EmitterGeom& g,
+                          Sampler& sampler, float& pdf) {
+    const int w = g.env_w, h = g.env_h;
+    int row = cdf_find(g.env_rows, h, sampler.next1f());
+    const float* crow = g.env_cols + (size_t)row * w;
+    int col = cdf_find(crow, w, sampler.next1f());
+    float pr = g.env_rows[row] - (row ? g.env_rows[row - 1] : 0.f);
+    float pc = crow[col] - (col ? crow[col - 1] : 0.f);
+    float uu = (col + sampler.next1f()) / w;
+    float vv = (row + sampler.next1f()) / h;
+    float theta = vv * PI, phi = (uu - 0.5f) * 2.f * PI;
+    float st = sinf(theta);
+    Vec3 dl(st * sinf(phi), cosf(theta), -st * cosf(phi));
+    // invert the envmap rotation applied in envmap_eval
+    Vec3 dir = dl;
+    float az = e.aux.x, ze = e.aux.y;
+    if (az != 0.f || ze != 0.f) {
+        Quat qi = Quat::angle_axis(ze, Vec3(1.f, 0.f, 0.f)) *
+                  Quat::angle_axis(az, Vec3(0.f, 1.f, 0.f));
+        dir = qi.rotate(dl);
+    }
+    pdf = pr * pc * (float)w * (float)h / (2.f * PI * PI * fmaxf(st, 1e-5f));
+    return dir;
 }
 
 // NEE sample toward emitter e from shading point `sp_pos` with normal `sp_n`.
@@ -162,30 +193,12 @@ HD EmitterSampleRec emitter_sample(const EmitterParams& e, const EmitterGeom& g,
     case EM_ENVMAP: {
         if (g.env_rows && e.tex_id >= 0) {
             // luminance-CDF importance sampling over the lat-long texture
-            const int w = g.env_w, h = g.env_h;
-            int row = cdf_find(g.env_rows, h, sampler.next1f());
-            const float* crow = g.env_cols + (size_t)row * w;
-            int col = cdf_find(crow, w, sampler.next1f());
-            float pr = g.env_rows[row] - (row ? g.env_rows[row - 1] : 0.f);
-            float pc = crow[col] - (col ? crow[col - 1] : 0.f);
-            float uu = (col + sampler.next1f()) / w;
-            float vv = (row + sampler.next1f()) / h;
-            float theta = vv * PI, phi = (uu - 0.5f) * 2.f * PI;
-            float st = sinf(theta);
-            Vec3 dl(st * sinf(phi), cosf(theta), -st * cosf(phi));
-            // invert the envmap rotation applied in envmap_eval
-            Vec3 dir = dl;
-            float az = e.aux.x, ze = e.aux.y;
-            if (az != 0.f || ze != 0.f) {
-                Quat qi = Quat::angle_axis(ze, Vec3(1.f, 0.f, 0.f)) *
-                          Quat::angle_axis(az, Vec3(0.f, 1.f, 0.f));
-                dir = qi.rotate(dl);
-            }
+            float pdf;
+            Vec3 dir = envmap_sample_dir(e, g, sampler, pdf);
             r.pos = sp_pos + dir * ENVMAP_DIST;
             r.normal = -dir;
             r.radiance = envmap_eval(e, dir, g.textures);
-            r.pdf = pr * pc * (float)w * (float)h /
-                    (2.f * PI * PI * fmaxf(st, 1e-5f));
+            r.pdf = pdf;
             r.delta = false;
             return r;
         }
@@ -305,6 +318,35 @@ HD EmitterLeRec emitter_sample_le(const EmitterParams& e, const EmitterGeom& g, 
         Vec3 rad = emitter_radiance_tex(e, g.textures, uv);
         // pdf_pos = inv_area; throughput = L * cos / (pdf_pos * pdf_dir)
         r.throughput = rad * (cos_l / fmaxf(e.inv_area * pdf_dir, 1e-12f));
+        r.valid = true;
+        return r;
+    }
+    case EM_ENVMAP: {
+        // Light-path emission from the environment (reference
+        // EnvMapEmitter::sample_le, emitter.cuh:338): pick a direction toward
+        // the env (importance-sampled from the luminance CDF when present,
+        // else uniform sphere), then a point on the disk of the scene
+        // bounding sphere perpendicular to it, and shoot the ray INWARD.
+        // pdf_pos = 1/(pi R^2) over the disk; the disk is perpendicular to
+        // the ray so the cosine is 1:
+        //   throughput = Le(dir) * pi R^2 / pdf_dir.
+        float pdf_dir;
+        Vec3 dir;  // from scene toward env
+        if (g.env_rows && e.tex_id >= 0) {
+            dir = envmap_sample_dir(e, g, sampler, pdf_dir);
+        } else {
+            dir = sample_uniform_sphere(sampler.next2f(), pdf_dir);
+        }
+        if (pdf_dir <= 0.f) return r;
+        float R = fmaxf(g.scene_radius, 1e-4f);
+        Vec2 u = sampler.next2f();
+        float rr = sqrtf(u.x), phi = 2.f * PI * u.y;
+        Frame f = Frame::from_n(dir);
+        Vec3 offset = f.to_world(Vec3(rr * cosf(phi), rr * sinf(phi), 0.f)) * R;
+        Vec3 origin = g.scene_center + dir * (2.f * R) + offset;
+        r.ray = Ray(origin, -dir);
+        r.normal = -dir;
+        r.throughput = envmap_eval(e, dir, g.textures) * (PI * R * R / pdf_dir);
         r.valid = true;
         return r;
     }

@@ -70,10 +70,13 @@ struct SceneView {
     // max depth of the 4-wide tree (levels); gates kernels with halved
     // traversal stacks (wavefront dual-walk trace)
     int bvh4_depth;
+    // scene bounding sphere (xyz = center, w = radius); envmap sample_le
+    Vec4 scene_bound;
 
     HD EmitterGeom emitter_geom() const {
         return {prims, attrs, prim_obj, emitter_prims, emitter_cdf, textures,
-                env_rows, env_cols, env_w, env_h};
+                env_rows, env_cols, env_w, env_h,
+                scene_bound.xyz(), scene_bound.w};
     }
     HD uint32_t obj_of_prim(int pid) const { return prim_obj[pid] & PRIM_OBJ_MASK; }
     HD bool prim_is_sphere(int pid) const { return (prim_obj[pid] & PRIM_SPHERE_BIT) != 0; }
