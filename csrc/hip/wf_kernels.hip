@@ -68,12 +68,29 @@ __device__ __forceinline__ uint32_t pack_counts(int d, int s, int t, int b) {
     return (uint32_t)(d | (s << 8) | (t << 16) | (b << 24));
 }
 
+// Dynamic-LDS layout for all traversal kernels here, same as the megakernel
+// (pt_kernels.hip k_render): [n_cached 128-byte top-tree nodes][stacks].
+// Every walk's first visits are nodes 0..n_cached and the walk is bound by
+// hit LATENCY (95% L2 hit rate) — LDS is ~4x closer than L2.  Round-1 WFPT
+// omitted this cache; it was the bulk of the megakernel-vs-wavefront gap
+// (trace alone cost as much as the whole megakernel, profiles/README.md r02).
+__device__ __forceinline__ TravCtx wf_lds_ctx(const SceneView& sv, uint64_t* s_stk,
+                                              int lds_n, int n_cached) {
+    BVH4Node* s_cache = (BVH4Node*)s_stk;
+    uint64_t* s_base = s_stk + (size_t)n_cached * 16;
+    for (int i = threadIdx.x; i < n_cached * 16; i += 256)
+        ((uint64_t*)s_cache)[i] = ((const uint64_t*)sv.nodes4)[i];
+    if (n_cached > 0) __syncthreads();
+    return TravCtx{&s_base[threadIdx.x], lds_n, s_cache, n_cached};
+}
+
 // ----------------------------------------------------------------- raygen
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
-void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n) {
+void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n,
+                 int n_cached) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n};
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
     int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= st.n) return;
     int px = i % st.w, py = i / st.w;
@@ -314,9 +331,9 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
 // then a race-free add into L (exactly one shadow ray per payload per bounce).
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
-void k_wf_shadow(SceneView sv, WfState st, int lds_n) {
+void k_wf_shadow(SceneView sv, WfState st, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n};
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
     int k = blockIdx.x * blockDim.x + threadIdx.x;
     if (k >= *st.sh_cnt) return;
     float4 od = st.sh_od[k];
@@ -373,9 +390,9 @@ __device__ inline void wf_trace_finish(const SceneView& sv, WfState& st, int i,
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
 void k_wf_trace(SceneView sv, WfState st, const uint32_t* __restrict__ order,
-                int live, int lds_n) {
+                int live, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n};
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
     int k = blockIdx.x * blockDim.x + threadIdx.x;
     if (k >= live) return;
     int i = (int)(order[k] & 0x00FFFFFFu);
@@ -444,9 +461,9 @@ void k_wf_trace_dual(SceneView sv, WfState st, const uint32_t* __restrict__ orde
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
 void k_wf_tail(SceneView sv, WfState st, const uint32_t* __restrict__ order,
-               int live, int lds_n) {
+               int live, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
-    TravCtx tc{&s_stk[threadIdx.x], lds_n};
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
     int k = blockIdx.x * blockDim.x + threadIdx.x;
     if (k >= live) return;
     int i = (int)(order[k] & 0x00FFFFFFu);
@@ -574,11 +591,21 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         }
         return v;
     }();
-    static int lds_n = [] {
+    static int lds_budget = [] {
         const char* e = getenv("HIPPT_WF_STACK");
-        return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res;
+        return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res * WF_BLOCK * 8;
     }();
-    const uint32_t shmem = (uint32_t)lds_n * WF_BLOCK * 8;
+    // LDS top-tree cache shares the block budget with the stacks (same
+    // occupancy, fewer LDS stack entries; overflow spills to scratch)
+    static int cache_req = [] {
+        const char* e = getenv("HIPPT_TOPCACHE");
+        return e ? atoi(e) : 64;
+    }();
+    const int n_cached = lds_budget > 0
+        ? (cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4) : 0;
+    const int lds_n = lds_budget > 0
+        ? (lds_budget - n_cached * (int)sizeof(BVH4Node)) / (WF_BLOCK * 8) : 0;
+    const uint32_t shmem = (uint32_t)(lds_n * WF_BLOCK * 8 + n_cached * (int)sizeof(BVH4Node));
     static int wf_dual = [] {
         const char* e = getenv("HIPPT_WF_DUAL");
         return e ? atoi(e) : 0;
@@ -590,8 +617,13 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         const char* e = getenv("HIPPT_WF_TAIL");
         return e ? atoi(e) : 128 * 1024;
     }();
+    static int wf_log = [] {
+        const char* e = getenv("HIPPT_WF_LOG");
+        return e ? atoi(e) : 0;
+    }();
     for (int s = 0; s < nspp; ++s) {
-        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed, lds_n);
+        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed,
+                           lds_n, n_cached);
         // Per bounce: build the compacted material-sorted live view, read the
         // live count back (4 bytes — the price the reference also pays,
         // wf_path_tracer.cu:199), then launch live-sized grids.  Bounce 0
@@ -616,25 +648,30 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             int err = (int)hipStreamSynchronize(hs);
             if (err) return err;
             const int live = *st->live_host;
+            if (wf_log && s == 0)
+                printf("[wf] spp %d bounce %d live %d (%.1f%%)\n", spp0 + s, bounce,
+                       live, 100.0 * live / n);
             if (live == 0) break;                     // reference: break at live==0
             dim3 grd_live((live + WF_BLOCK - 1) / WF_BLOCK);
             if (live <= tail_thresh) {
                 hipLaunchKernelGGL(k_wf_tail, grd_live, blk, shmem, hs, sv, *st,
-                                   order_cur, live, lds_n);
+                                   order_cur, live, lds_n, n_cached);
                 break;
             }
             (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, order_cur,
                                live, bounce);
-            hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, shmem, hs, sv, *st, lds_n);
+            hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, shmem, hs, sv, *st,
+                               lds_n, n_cached);
             if (bounce < sv.md.max_depth) {           // reference skips last-bounce trace
                 if (wf_dual) {
                     dim3 grd_h(((live + 1) / 2 + WF_BLOCK - 1) / WF_BLOCK);
-                    hipLaunchKernelGGL(k_wf_trace_dual, grd_h, blk, shmem, hs, sv, *st,
+                    hipLaunchKernelGGL(k_wf_trace_dual, grd_h, blk,
+                                       (uint32_t)(lds_n * WF_BLOCK * 8), hs, sv, *st,
                                        order_cur, live, lds_n);
                 } else {
                     hipLaunchKernelGGL(k_wf_trace, grd_live, blk, shmem, hs, sv, *st,
-                                       order_cur, live, lds_n);
+                                       order_cur, live, lds_n, n_cached);
                 }
             }
             gather = order_cur;
