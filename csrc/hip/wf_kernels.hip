@@ -85,8 +85,11 @@ __device__ __forceinline__ TravCtx wf_lds_ctx(const SceneView& sv, uint64_t* s_s
 }
 
 // ----------------------------------------------------------------- raygen
+template <int MINW = 4>
 __global__ __launch_bounds__(256)
-__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))  // exact residency = what the
+// LDS stack budget allows; frees 512/MINW VGPRs so the walk state does not
+// spill (same fix as the megakernel, measured +9% there)
 void k_wf_raygen(SceneView sv, WfState st, int spp_idx, uint32_t seed, int lds_n,
                  int n_cached) {
     extern __shared__ uint64_t s_stk[];
@@ -329,8 +332,11 @@ void k_wf_shade(SceneView sv, WfState st, const uint32_t* __restrict__ order,
 // ---------------------------------------------------- shadow-ray resolve
 // Lean traversal-only kernel (58 VGPR class -> 8 waves/SIMD): any-hit test,
 // then a race-free add into L (exactly one shadow ray per payload per bounce).
+template <int MINW = 4>
 __global__ __launch_bounds__(256)
-__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))  // exact residency = what the
+// LDS stack budget allows; frees 512/MINW VGPRs so the walk state does not
+// spill (same fix as the megakernel, measured +9% there)
 void k_wf_shadow(SceneView sv, WfState st, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
@@ -387,8 +393,11 @@ __device__ inline void wf_trace_finish(const SceneView& sv, WfState& st, int i,
 // adjacency from the scatter, so wave-level spatial coherence survives the
 // material grouping.  Shade may have terminated a ray after the sort ran,
 // so the fresh status byte is still checked.
+template <int MINW = 4>
 __global__ __launch_bounds__(256)
-__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))  // exact residency = what the
+// LDS stack budget allows; frees 512/MINW VGPRs so the walk state does not
+// spill (same fix as the megakernel, measured +9% there)
 void k_wf_trace(SceneView sv, WfState st, const uint32_t* __restrict__ order,
                 int live, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
@@ -450,26 +459,13 @@ void k_wf_trace_dual(SceneView sv, WfState st, const uint32_t* __restrict__ orde
     if (a1) wf_trace_finish(sv, st, i1, ray1, pdf1, w1.rec);
 }
 
-// ------------------------------------------------------------- tail fuse
-// Below HIPPT_WF_TAIL live rays the chip is under-occupied anyway, so
-// per-bounce sorting + 4 kernel launches cost more than the wave64
-// divergence they avoid.  This kernel finishes every surviving path
-// megakernel-style: shade the stored hit (integrator.h path_shade_hit),
-// then loop full path_step bounces (inline NEE occlusion, no shadow
-// queue) until the path dies.  Mirrors the reference's early loop exit at
-// live == 0 (wf_path_tracer.cu:199-210) taken one step further.
-__global__ __launch_bounds__(256)
-__attribute__((amdgpu_waves_per_eu(6, 6)))   // LDS stack caps at 6 waves/SIMD
-void k_wf_tail(SceneView sv, WfState st, const uint32_t* __restrict__ order,
-               int live, int lds_n, int n_cached) {
-    extern __shared__ uint64_t s_stk[];
-    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
-    int k = blockIdx.x * blockDim.x + threadIdx.x;
-    if (k >= live) return;
-    int i = (int)(order[k] & 0x00FFFFFFu);
+// --------------------------------------------- payload <-> PathState glue
+// The wavefront payload pool resumes paths through the SAME single-source
+// integrator as the megakernel (path_shade_hit / path_step).
+__device__ __forceinline__ void wf_load_ps(const WfState& st, int i, PathState& ps,
+                                           Sampler& sp, HitRecord& hit) {
     float4 ro4 = st.ray_o[i], rd4 = st.ray_d[i], thp4 = st.thp[i], l4 = st.L[i];
     float4 h4 = st.hit[i], pn4 = st.prevn[i];
-    PathState ps;
     ps.ray = Ray(Vec3(ro4.x, ro4.y, ro4.z), Vec3(rd4.x, rd4.y, rd4.z));
     ps.L = Vec3(l4.x, l4.y, l4.z);
     ps.thp = Vec3(thp4.x, thp4.y, thp4.z);
@@ -486,13 +482,94 @@ void k_wf_tail(SceneView sv, WfState st, const uint32_t* __restrict__ order,
     ps.iter = ps.b + 1;  // null bounces before the handoff are not replayed
     ps.aov_n = Vec3(0.f); ps.aov_alb = Vec3(0.f); ps.aov_t = 0.f;
     ps.lambda = ro4.w;
-    Sampler sp(st.rng[i]);
-    HitRecord hit;
+    sp = Sampler(st.rng[i]);
     hit.t = h4.x; hit.u = h4.y; hit.v = h4.z;
     hit.prim_idx = float_as_int(h4.w);
+}
+
+__device__ __forceinline__ void wf_store_ps(WfState& st, int i, const PathState& ps,
+                                            const Sampler& sp) {
+    st.ray_o[i] = make_float4(ps.ray.o.x, ps.ray.o.y, ps.ray.o.z, ps.lambda);
+    st.ray_d[i] = make_float4(ps.ray.d.x, ps.ray.d.y, ps.ray.d.z, ps.prev_pdf);
+    st.thp[i] = make_float4(ps.thp.x, ps.thp.y, ps.thp.z,
+                            uint_as_float(ps.prev_delta ? 1u : 0u));
+    float lw = st.L[i].w;
+    st.L[i] = make_float4(ps.L.x, ps.L.y, ps.L.z, lw);
+    st.prevn[i] = make_float4(ps.prev_n.x, ps.prev_n.y, ps.prev_n.z,
+                              uint_as_float(pack_counts(ps.st.n_diffuse, ps.st.n_specular,
+                                                        ps.st.n_transmit, ps.b)));
+    st.rng[i] = sp.state;
+}
+
+// ------------------------------------------------------- fused bounce step
+// One kernel per bounce: shade the stored hit (NEE occlusion inline — the
+// megakernel-measured winner at occ 4) and immediately trace the sampled
+// ray while it is still in registers.  Versus the split
+// shade/shadow-queue/trace pipeline this removes two kernel boundaries
+// worth of payload HBM round trips per bounce and gives the latency-bound
+// BVH walk shade VALU work to overlap.  The sort/compaction stays — the
+// wavefront's actual value on this hardware.  HIPPT_WF_FUSE=0 restores the
+// split pipeline for A/B.
+template <int MINW = 4>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))
+void k_wf_step(SceneView sv, WfState st, const uint32_t* __restrict__ order,
+               int live, int do_trace, int lds_n, int n_cached) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
+    int k = blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= live) return;
+    int i = (int)(order[k] & 0x00FFFFFFu);
+    PathState ps;
+    Sampler sp(0, 0);
+    HitRecord hit;
+    wf_load_ps(st, i, ps, sp, hit);
+    bool done = path_shade_hit(sv, ps, sp, tc, hit);
+    if (!done && ps.b >= sv.md.max_depth) done = true;  // path_step entry cap
+    uint32_t status = DEAD;
+    if (!done && do_trace) {
+        HitRecord nh = scene_intersect(sv, ps.ray, MAX_DIST, tc);
+        if (nh.prim_idx < 0) {
+            nh.t = MAX_DIST;
+            path_shade_hit(sv, ps, sp, tc, nh);  // miss branch: envmap + MIS
+        } else {
+            uint32_t oi = sv.prim_obj[nh.prim_idx] & PRIM_OBJ_MASK;
+            status = (uint32_t)(sv.objs[oi].bsdf_id & 0x3F);
+        }
+        st.hit[i] = make_float4(nh.t, nh.u, nh.v, int_as_float(nh.prim_idx));
+    }
+    wf_store_ps(st, i, ps, sp);
+    st.status[i] = (status << 24) | (uint32_t)i;
+}
+
+// ------------------------------------------------------------- tail fuse
+// Below HIPPT_WF_TAIL live rays the chip is under-occupied anyway, so
+// per-bounce sorting + kernel launches cost more than the wave64
+// divergence they avoid.  This kernel finishes every surviving path
+// megakernel-style: shade the stored hit (integrator.h path_shade_hit),
+// then loop full path_step bounces (inline NEE occlusion, no shadow
+// queue) until the path dies.  Mirrors the reference's early loop exit at
+// live == 0 (wf_path_tracer.cu:199-210) taken one step further.
+template <int MINW = 4>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))  // exact residency = what the
+// LDS stack budget allows; frees 512/MINW VGPRs so the walk state does not
+// spill (same fix as the megakernel, measured +9% there)
+void k_wf_tail(SceneView sv, WfState st, const uint32_t* __restrict__ order,
+               int live, int lds_n, int n_cached) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
+    int k = blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= live) return;
+    int i = (int)(order[k] & 0x00FFFFFFu);
+    PathState ps;
+    Sampler sp(0, 0);
+    HitRecord hit;
+    wf_load_ps(st, i, ps, sp, hit);
     bool done = path_shade_hit(sv, ps, sp, tc, hit);
     while (!done) done = path_step(sv, ps, sp, tc);
-    st.L[i] = make_float4(ps.L.x, ps.L.y, ps.L.z, l4.w);
+    float lw = st.L[i].w;
+    st.L[i] = make_float4(ps.L.x, ps.L.y, ps.L.z, lw);
     st.status[i] = (DEAD << 24) | (uint32_t)i;
 }
 
@@ -569,31 +646,38 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     const int n = st->n;
     dim3 blk(WF_BLOCK);
     dim3 grd_n((n + WF_BLOCK - 1) / WF_BLOCK);
-    // HIPPT_WF_OCC = waves/SIMD cap (reserved LDS) and HIPPT_WF_STACK =
-    // lds | scratch for the traversal kernels (defaults: occ 6, lds —
-    // measured best for the lean trace/shadow kernels, profiles/README.md).
-    static int occ_res = [] {
+    // HIPPT_WF_OCC = exact waves/SIMD for the traversal kernels (both the
+    // amdgpu_waves_per_eu instantiation — frees 512/occ VGPRs so the walk
+    // state does not spill — AND the matching LDS block budget).  Default 4,
+    // the megakernel's measured best.  HIPPT_WF_STACK=scratch disables the
+    // LDS stack + cache entirely (A/B hook).
+    static int occ_v = [] {
         const char* e = getenv("HIPPT_WF_OCC");
-        int occ = e ? atoi(e) : 5;   // ww walk: occ4/5 ~100 vs occ6 98.5 Msps
-        int m[7] = {20, 20, 40, 26, 20, 16, 12};
-        int v = m[occ < 0 ? 0 : (occ > 6 ? 6 : occ)];
-        if (v * WF_BLOCK * 8 > 65536) {
-            (void)hipFuncSetAttribute((const void*)&k_wf_raygen,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
-            (void)hipFuncSetAttribute((const void*)&k_wf_trace,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
-            (void)hipFuncSetAttribute((const void*)&k_wf_shadow,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
-            (void)hipFuncSetAttribute((const void*)&k_wf_trace_dual,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
-            (void)hipFuncSetAttribute((const void*)&k_wf_tail,
-                hipFuncAttributeMaxDynamicSharedMemorySize, v * WF_BLOCK * 8);
-        }
-        return v;
+        int occ = e ? atoi(e) : 4;
+        return occ < 3 ? 3 : (occ > 6 ? 6 : occ);
     }();
+    using RaygenFn = void (*)(SceneView, WfState, int, uint32_t, int, int);
+    using TraceFn = void (*)(SceneView, WfState, const uint32_t*, int, int, int);
+    using ShadowFn = void (*)(SceneView, WfState, int, int);
+    static RaygenFn f_raygen = occ_v == 3 ? k_wf_raygen<3> : occ_v == 4 ? k_wf_raygen<4>
+                             : occ_v == 5 ? k_wf_raygen<5> : k_wf_raygen<6>;
+    static TraceFn f_trace = occ_v == 3 ? k_wf_trace<3> : occ_v == 4 ? k_wf_trace<4>
+                           : occ_v == 5 ? k_wf_trace<5> : k_wf_trace<6>;
+    static TraceFn f_tail = occ_v == 3 ? k_wf_tail<3> : occ_v == 4 ? k_wf_tail<4>
+                          : occ_v == 5 ? k_wf_tail<5> : k_wf_tail<6>;
+    using StepFn = void (*)(SceneView, WfState, const uint32_t*, int, int, int, int);
+    static StepFn f_step = occ_v == 3 ? k_wf_step<3> : occ_v == 4 ? k_wf_step<4>
+                         : occ_v == 5 ? k_wf_step<5> : k_wf_step<6>;
+    static int wf_fuse = [] {
+        const char* e = getenv("HIPPT_WF_FUSE");
+        return e ? atoi(e) : 1;
+    }();
+    static ShadowFn f_shadow = occ_v == 3 ? k_wf_shadow<3> : occ_v == 4 ? k_wf_shadow<4>
+                             : occ_v == 5 ? k_wf_shadow<5> : k_wf_shadow<6>;
     static int lds_budget = [] {
         const char* e = getenv("HIPPT_WF_STACK");
-        return (e && strcmp(e, "scratch") == 0) ? 0 : occ_res * WF_BLOCK * 8;
+        if (e && strcmp(e, "scratch") == 0) return 0;
+        return (occ_v == 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12) * WF_BLOCK * 8;
     }();
     // LDS top-tree cache shares the block budget with the stacks (same
     // occupancy, fewer LDS stack entries; overflow spills to scratch)
@@ -622,7 +706,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return e ? atoi(e) : 0;
     }();
     for (int s = 0; s < nspp; ++s) {
-        hipLaunchKernelGGL(k_wf_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed,
+        hipLaunchKernelGGL(f_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed,
                            lds_n, n_cached);
         // Per bounce: build the compacted material-sorted live view, read the
         // live count back (4 bytes — the price the reference also pays,
@@ -654,14 +738,18 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
             if (live == 0) break;                     // reference: break at live==0
             dim3 grd_live((live + WF_BLOCK - 1) / WF_BLOCK);
             if (live <= tail_thresh) {
-                hipLaunchKernelGGL(k_wf_tail, grd_live, blk, shmem, hs, sv, *st,
+                hipLaunchKernelGGL(f_tail, grd_live, blk, shmem, hs, sv, *st,
                                    order_cur, live, lds_n, n_cached);
                 break;
             }
+            if (wf_fuse) {
+                hipLaunchKernelGGL(f_step, grd_live, blk, shmem, hs, sv, *st, order_cur,
+                                   live, bounce < sv.md.max_depth ? 1 : 0, lds_n, n_cached);
+            } else {
             (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, order_cur,
                                live, bounce);
-            hipLaunchKernelGGL(k_wf_shadow, grd_live, blk, shmem, hs, sv, *st,
+            hipLaunchKernelGGL(f_shadow, grd_live, blk, shmem, hs, sv, *st,
                                lds_n, n_cached);
             if (bounce < sv.md.max_depth) {           // reference skips last-bounce trace
                 if (wf_dual) {
@@ -670,9 +758,10 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
                                        (uint32_t)(lds_n * WF_BLOCK * 8), hs, sv, *st,
                                        order_cur, live, lds_n);
                 } else {
-                    hipLaunchKernelGGL(k_wf_trace, grd_live, blk, shmem, hs, sv, *st,
+                    hipLaunchKernelGGL(f_trace, grd_live, blk, shmem, hs, sv, *st,
                                        order_cur, live, lds_n, n_cached);
                 }
+            }
             }
             gather = order_cur;
             prev_live = live;
