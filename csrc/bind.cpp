@@ -482,13 +482,14 @@ struct SceneHolder {
 
 // ------------------------------------------------------------- BVH builder
 py::tuple py_build_bvh(farr prims, uarr prim_obj, int max_leaf, float overlap_w,
-                       bool use_sbvh, bool ref_unsplit) {
+                       bool use_sbvh, bool ref_unsplit, float trav_cost) {
     int n = (int)prims.shape(0);
     BVHBuildConfig cfg;
     cfg.max_leaf_prims = max_leaf;
     cfg.overlap_w = overlap_w;
     cfg.use_sbvh = use_sbvh;
     cfg.ref_unsplit = ref_unsplit;
+    cfg.trav_cost = trav_cost;
     BVHBuildResult res;
     {
         py::gil_scoped_release rel;
@@ -724,7 +725,8 @@ PYBIND11_MODULE(_C, m) {
 
     m.def("build_bvh", &py_build_bvh, py::arg("prims"), py::arg("prim_obj"),
           py::arg("max_leaf") = 4, py::arg("overlap_w") = 0.f,
-          py::arg("use_sbvh") = false, py::arg("ref_unsplit") = true);
+          py::arg("use_sbvh") = false, py::arg("ref_unsplit") = true,
+          py::arg("trav_cost") = 0.f);
     m.def("collapse_bvh4", &py_collapse_bvh4, py::arg("nodes"));
     m.def("collapse_bvh8", &py_collapse_bvh8, py::arg("nodes"));
     m.def("bvh4_hit", &py_bvh4_hit);

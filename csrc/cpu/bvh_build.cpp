@@ -37,6 +37,7 @@ struct Builder {
     std::mutex pool_mu;
     int max_leaf;
     float overlap_w;
+    float trav_cost;
 
     int alloc_node() {
         std::lock_guard<std::mutex> g(pool_mu);
@@ -84,7 +85,8 @@ struct Builder {
             if (cost < best) { best = cost; best_b = b; rbox_best = rboxes[b + 1]; }
         }
         float leaf_cost = box.area() * n;
-        if (best_b < 0 || (n <= max_leaf && best >= leaf_cost)) return -1;
+        if (best_b < 0 || (n <= max_leaf && best + trav_cost * box.area() >= leaf_cost))
+            return -1;
         // partition
         auto mid = std::partition(bp.begin() + lo, bp.begin() + hi, [&](const BuildPrim& p) {
             int b = std::min(N_BINS - 1, (int)((p.centroid[axis] - cmin) * inv));
@@ -194,6 +196,7 @@ BVHBuildResult build_bvh(const Prim* prims, const uint32_t* prim_obj, int n,
     Builder b;
     b.max_leaf = std::max(1, std::min(cfg.max_leaf_prims, 15));
     b.overlap_w = cfg.overlap_w;
+    b.trav_cost = cfg.trav_cost;
     b.bp.resize(n);
     for (int i = 0; i < n; ++i) {
         AABB box;

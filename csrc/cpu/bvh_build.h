@@ -24,6 +24,11 @@ struct BVHBuildResult {
 struct BVHBuildConfig {
     int max_leaf_prims = 4;
     float overlap_w = 0.f;       // SAH overlap penalty weight (reference bvh_overlap_w)
+    // SAH traversal-cost constant (units of one prim test): split only when
+    // Ct*A_parent + sum A_i*N_i beats A_parent*N.  0 = reference behavior
+    // (always split while area*count drops), which over-fragments leaves for
+    // a 4-wide walk whose node step costs ~a dependent 128-B load.
+    float trav_cost = 0.f;
     bool use_sbvh = false;       // spatial splits (SBVH)
     bool ref_unsplit = true;     // SBVH reference unsplitting
     int n_threads = 8;

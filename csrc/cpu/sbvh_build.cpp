@@ -42,6 +42,7 @@ struct Ctx {
     int max_leaf;
     bool ref_unsplit;
     float inv_root_area;
+    float trav_cost = 0.f;
     // final leaves' reference order (append-only under mutex-free ownership:
     // built single-threaded per subtree, merged bottom-up)
 };
@@ -236,7 +237,10 @@ void build_rec(const Ctx& ctx, std::vector<Refr>&& refs, int depth, BuildOut& ou
         return;
     }
     if (n <= ctx.max_leaf &&
-        std::min(best_obj_cost, best_sp_cost) >= leaf_cost) { make_leaf(); return; }
+        std::min(best_obj_cost, best_sp_cost) + ctx.trav_cost * box.area() >= leaf_cost) {
+        make_leaf();
+        return;
+    }
 
     std::vector<Refr> lrefs, rrefs;
     if (use_spatial) {
@@ -354,7 +358,7 @@ BVHBuildResult build_sbvh(const Prim* prims, const uint32_t* prim_obj, int n,
         root_box.grow(box);
     }
     Ctx ctx{prims, prim_obj, std::max(1, std::min(cfg.max_leaf_prims, 15)), cfg.ref_unsplit,
-            root_box.area() > 0.f ? 1.f / root_box.area() : 0.f};
+            root_box.area() > 0.f ? 1.f / root_box.area() : 0.f, cfg.trav_cost};
     BuildOut out;
     out.nodes.reserve(2 * n);
     out.leaf_refs.reserve((size_t)(n * 1.3));

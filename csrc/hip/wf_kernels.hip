@@ -22,10 +22,11 @@
 //   * from bounce 1 on, the sort passes GATHER through the previous
 //     bounce's compacted order array — their cost scales with the live
 //     count, not with W*H.
-// Below HIPPT_WF_TAIL live rays (default 128K: the point where 256 CUs run
-// out of resident waves anyway), one fused tail kernel finishes every
-// surviving path megakernel-style (integrator.h path_shade_hit/path_step),
-// eliminating all remaining per-bounce sorts and launches.
+// Below HIPPT_WF_TAIL live rays (default 1.5M, measured: per-bounce pass
+// overhead outweighs the divergence it avoids well before the chip runs
+// dry), one fused tail kernel finishes every surviving path
+// megakernel-style (integrator.h path_shade_hit/path_step), eliminating
+// all remaining per-bounce sorts and launches.
 #include <hip/hip_runtime.h>
 #include "kernels.h"
 #include <cstdlib>
@@ -653,7 +654,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     // LDS stack + cache entirely (A/B hook).
     static int occ_v = [] {
         const char* e = getenv("HIPPT_WF_OCC");
-        int occ = e ? atoi(e) : 4;
+        int occ = e ? atoi(e) : 3;   // measured best for the fused step (r02)
         return occ < 3 ? 3 : (occ > 6 ? 6 : occ);
     }();
     using RaygenFn = void (*)(SceneView, WfState, int, uint32_t, int, int);
@@ -698,8 +699,11 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     // entries; a node visit pushes at most 3 entries per level)
     if (wf_dual && 3 * sv.bvh4_depth > BVH4_STACK / 2 + lds_n / 2) wf_dual = 0;
     static int tail_thresh = [] {
+        // Hand off to the fused tail once <=1.5M rays are live (measured
+        // best on kitchen 1080p: per-bounce pipeline overhead outweighs the
+        // wave64 divergence it avoids well before the chip runs dry).
         const char* e = getenv("HIPPT_WF_TAIL");
-        return e ? atoi(e) : 128 * 1024;
+        return e ? atoi(e) : 1500 * 1024;
     }();
     static int wf_log = [] {
         const char* e = getenv("HIPPT_WF_LOG");

@@ -154,6 +154,9 @@ class RenderConfig:
     renderer: str = "pt"               # pt | wfpt | vpt | lt | depth | bvh-cost | pt-dyn
     max_leaf: int = 4
     overlap_w: float = 0.6   # SAH overlap penalty (reference accelerator default)
+    # SAH traversal-cost constant for the 4-wide tree (units of one prim
+    # test); >0 coarsens leaves so the latency-bound walk visits fewer nodes
+    bvh_trav_cost: float = 0.0
     use_sbvh: bool = False
     ref_unsplit: bool = True
     cache_level: int = 6
@@ -310,8 +313,9 @@ class Scene:
         import os as _os
         max_leaf = int(_os.environ.get("HIPPT_MAX_LEAF", cfg.max_leaf))
         overlap_w = float(_os.environ.get("HIPPT_OVERLAP_W", cfg.overlap_w))
+        trav_cost = float(_os.environ.get("HIPPT_BVH_CT", cfg.bvh_trav_cost))
         nodes, order, stats = C.build_bvh(prims, prim_obj, max_leaf, overlap_w,
-                                          cfg.use_sbvh, cfg.ref_unsplit)
+                                          cfg.use_sbvh, cfg.ref_unsplit, trav_cost)
         self.bvh_stats = stats
         # 4-wide collapse of the binary tree: the traversal that actually runs
         # (ordered short-stack walk over 128-byte nodes, csrc/core/bvh4.h)
