@@ -40,9 +40,10 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world_size > 1
 
+    backend = os.environ.get("HIPPT_DDP_BACKEND", "gloo" if args.cpu else "nccl")
     if distributed:
         import torch.distributed as dist
-        dist.init_process_group(backend="gloo" if args.cpu else "nccl")
+        dist.init_process_group(backend=backend)
 
     import hippt
     from hippt.scene import procedural
@@ -65,10 +66,11 @@ def main():
     if args.sbvh or os.environ.get("HIPPT_SBVH"):
         desc.config.use_sbvh = True
 
-    device = None if args.cpu else local_rank
+    # more ranks than GPUs (single-GPU RCCL/gloo smoke): share devices
+    dev_id = local_rank if args.cpu else local_rank % max(1, torch.cuda.device_count())
     if not args.cpu:
-        torch.cuda.set_device(local_rank)
-    r = hippt.PythonRenderer(desc, device_id=-1 if args.cpu else local_rank,
+        torch.cuda.set_device(dev_id)
+    r = hippt.PythonRenderer(desc, device_id=-1 if args.cpu else dev_id,
                              seed_offset=rank)
     rend = r.renderer
 
@@ -89,7 +91,13 @@ def main():
             return
         import torch.distributed as dist
         t = rend.accum if not args.cpu else torch.from_numpy(rend.accum)
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        if backend != "nccl" and hasattr(t, "is_cuda") and t.is_cuda:
+            # gloo smoke (more ranks than GPUs): collective over a host copy
+            h = t.cpu()
+            dist.all_reduce(h, op=dist.ReduceOp.SUM)
+            t.copy_(h.to(t.device))
+        else:
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
         # keep per-rank accumulators independent afterwards for weak scaling:
         # divide back so ranks continue from their own estimate
         t /= world_size

@@ -84,10 +84,17 @@ def start_reduce(dist, rend, cpu=False):
     """Snapshot the accumulator on the compute stream and launch the SUM
     all-reduce asynchronously (RCCL comm stream).  The returned work handle
     is waited on by finish_reduce — after the caller has already queued the
-    next frame's kernels, so the xGMI collective overlaps them."""
+    next frame's kernels, so the xGMI collective overlaps them.  When the
+    backend cannot take CUDA tensors (gloo smoke with more ranks than
+    GPUs), the collective runs over a host copy."""
     import torch
-    t = rend.accum if not cpu else torch.from_numpy(rend.accum.copy())
-    merged = t.clone()
+    import torch.distributed as tdist
+    a = rend.accum
+    if isinstance(a, torch.Tensor):
+        gloo = tdist.get_backend() == "gloo" if tdist.is_initialized() else cpu
+        merged = a.cpu().clone() if (gloo and a.is_cuda) else a.clone()
+    else:
+        merged = torch.from_numpy(a.copy())
     work = dist.all_reduce(merged, op=dist.ReduceOp.SUM, async_op=True)
     return work, merged
 
@@ -124,7 +131,7 @@ def main(argv=None):
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    backend = "gloo" if args.cpu else "nccl"
+    backend = os.environ.get("HIPPT_DDP_BACKEND", "gloo" if args.cpu else "nccl")
     if world_size > 1:
         dist.init_process_group(backend=backend)
 
@@ -132,8 +139,8 @@ def main(argv=None):
     desc = load_scene(args)
     # per-rank decorrelated seeds (ddp_render.py:141-143)
     seed_offset = local_rank + args.seed * world_size
-    r = hippt.PythonRenderer(desc, device_id=-1 if args.cpu else local_rank,
-                             seed_offset=seed_offset)
+    dev_id = -1 if args.cpu else local_rank % max(1, torch.cuda.device_count())
+    r = hippt.PythonRenderer(desc, device_id=dev_id, seed_offset=seed_offset)
     rend = r.renderer
 
     ckpt_path = f"{args.checkpoint}.rank{rank}.npz" if args.checkpoint else None
