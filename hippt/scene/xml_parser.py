@@ -225,13 +225,21 @@ def parse_xml(path: str) -> SceneDesc:
             m.sigma_a = props.get("sigma_a", (0.1,) * 3)
             m.sigma_s = props.get("sigma_s", (1.0,) * 3)
         else:
-            # grid medium: .nvdb is not loadable offline (NanoVDB assets do not
-            # ship); native .npy/.vgrid grids are; missing file -> procedural
+            # grid medium: .nvdb (NanoVDB, reference vol_grid.cu:216-342 —
+            # converted to dense on the host, hippt/scene/nvdb.py) and native
+            # .npy grids load from disk; missing file -> procedural fallback
             dens_path = props.get("density", "")
             full = os.path.normpath(os.path.join(base, dens_path)) if dens_path else ""
             dens = None
-            if full and os.path.exists(full) and full.endswith(".npy"):
-                dens = np.load(full).astype(np.float32)
+            nvdb_bounds = None
+            if full and os.path.exists(full):
+                if full.endswith(".npy"):
+                    dens = np.load(full).astype(np.float32)
+                elif full.endswith(".nvdb"):
+                    from .nvdb import read_nvdb
+                    g = read_nvdb(full)[0]
+                    dens = g["dense"]
+                    nvdb_bounds = (g["world_min"], g["world_max"])
             if dens is None:
                 from .procedural import smoke_density
                 dens = smoke_density(n=96) * 12.0
@@ -241,10 +249,18 @@ def parse_xml(path: str) -> SceneDesc:
             m.sigma_a = tuple(1.0 - a for a in albedo)
             temp_path = props.get("emission", "")
             fullt = os.path.normpath(os.path.join(base, temp_path)) if temp_path else ""
-            if fullt and os.path.exists(fullt) and fullt.endswith(".npy"):
-                m.temperature = np.load(fullt).astype(np.float32)
-            m.grid_lo = props.get("grid_lo", (0.0, 0.0, 0.0))
-            m.grid_hi = props.get("grid_hi", (1.0, 1.0, 1.0))
+            if fullt and os.path.exists(fullt):
+                if fullt.endswith(".npy"):
+                    m.temperature = np.load(fullt).astype(np.float32)
+                elif fullt.endswith(".nvdb"):
+                    from .nvdb import read_nvdb
+                    m.temperature = read_nvdb(fullt)[0]["dense"]
+            # explicit grid_lo/hi win; else .nvdb world bounds; else unit cube
+            if "grid_lo" in props or nvdb_bounds is None:
+                m.grid_lo = props.get("grid_lo", (0.0, 0.0, 0.0))
+                m.grid_hi = props.get("grid_hi", (1.0, 1.0, 1.0))
+            else:
+                m.grid_lo, m.grid_hi = nvdb_bounds
         d.media.append(m)
         medium_ids[mid] = len(d.media) - 1
 

@@ -144,8 +144,20 @@ def sky_texture():
     write_png(os.path.join(tdir, "sky.png"), img)
 
 
+def smoke_nvdb():
+    """NanoVDB smoke asset for grid-cbox-nvdb.xml (hippt/scene/nvdb.py)."""
+    from hippt.scene.procedural import smoke_density
+    from hippt.scene.nvdb import write_nvdb
+    adir = os.path.join(ROOT, "scenes", "assets")
+    os.makedirs(adir, exist_ok=True)
+    d = (smoke_density(n=64) * 12.0).astype(np.float32)
+    write_nvdb(os.path.join(adir, "smoke.nvdb"), d, voxel_size=1.2 / 64,
+               world_origin=(-0.6, 0.05, 0.4), grid_name="density")
+
+
 def main():
     cbox_meshes()
+    smoke_nvdb()
     save_obj(f"{MESH}/floor_big.obj", quad((-6, 0, -6), (-6, 0, 6), (6, 0, 6), (6, 0, -6)))
     save_obj(f"{MESH}/lamp.obj", quad((1.5, 4, -1.5), (1.5, 4, 1.5), (-1.5, 4, 1.5), (-1.5, 4, -1.5)))
     save_obj(f"{MESH}/smoke_bound.obj", box_mesh((-0.6, 0.05, 0.4), (0.6, 1.6, 1.6)))
