@@ -70,6 +70,7 @@ struct SceneHolder {
     std::vector<PhaseParams> phases;
     std::vector<MediumParams> media;          // host pointers inside
     std::vector<farr> media_density, media_temp;
+    std::vector<std::vector<float>> media_super;   // majorant supergrids (raw max)
     std::vector<farr> tex_data;
     std::vector<TexView> tex_host;
     std::vector<int> emitter_prims;
@@ -235,7 +236,9 @@ struct SceneHolder {
         m.type = type; m.phase_id = phase_id;
         m.scale = scale; m.emission_scale = emission_scale; m.temp_scale = temp_scale;
         m.density = nullptr; m.temperature = nullptr;
+        m.super = nullptr;
         m.nx = m.ny = m.nz = 0;
+        m.sx = m.sy = m.sz = 0;
         if (type == MED_GRID) {
             farr d = density.cast<farr>();
             if (d.ndim() != 3) throw std::runtime_error("density must be (nz,ny,nx)");
@@ -248,6 +251,41 @@ struct SceneHolder {
             for (size_t i = 0; i < n; ++i) { mx = fmaxf(mx, dd[i]); sum += dd[i]; }
             m.majorant = mx * scale;
             m.avg_density = (float)(sum / std::max<size_t>(n, 1)) * scale;
+            // majorant supergrid: per-8^3-supercell RAW density max with a
+            // 1-voxel dilation (the stochastic-offset lookup samples up to
+            // +-0.5 voxel outside the cell); scale is applied in the walk so
+            // update_medium's scale changes need no rebuild
+            constexpr int SUP = 8;
+            if (getenv("HIPPT_NO_SUPER")) {
+                // A/B hook: single global majorant (round-1 behavior)
+                m.sx = m.sy = m.sz = 0;
+                m.super = nullptr;
+                goto super_done;
+            }
+            m.sx = (m.nx + SUP - 1) / SUP;
+            m.sy = (m.ny + SUP - 1) / SUP;
+            m.sz = (m.nz + SUP - 1) / SUP;
+            {
+            std::vector<float> sup((size_t)m.sx * m.sy * m.sz, 0.f);
+            for (int iz = 0; iz < m.nz; ++iz)
+                for (int iy = 0; iy < m.ny; ++iy)
+                    for (int ix = 0; ix < m.nx; ++ix) {
+                        float v = dd[(size_t(iz) * m.ny + iy) * m.nx + ix];
+                        if (v <= 0.f) continue;
+                        int cx0 = std::max(0, (ix - 1) / SUP), cx1 = std::min(m.sx - 1, (ix + 1) / SUP);
+                        int cy0 = std::max(0, (iy - 1) / SUP), cy1 = std::min(m.sy - 1, (iy + 1) / SUP);
+                        int cz0 = std::max(0, (iz - 1) / SUP), cz1 = std::min(m.sz - 1, (iz + 1) / SUP);
+                        for (int cz = cz0; cz <= cz1; ++cz)
+                            for (int cy = cy0; cy <= cy1; ++cy)
+                                for (int cx = cx0; cx <= cx1; ++cx) {
+                                    float& s = sup[(size_t(cz) * m.sy + cy) * m.sx + cx];
+                                    s = fmaxf(s, v);
+                                }
+                    }
+            media_super.push_back(std::move(sup));
+            m.super = media_super.back().data();
+            }
+            super_done:;
             Vec3 lo = to_vec3(grid_lo), hi = to_vec3(grid_hi);
             m.grid_lo = Vec4(lo, 0.f);
             Vec3 ext = hi - lo;
@@ -406,6 +444,10 @@ struct SceneHolder {
             if (media[i].temperature) {
                 size_t n = (size_t)media[i].nx * media[i].ny * media[i].nz;
                 media_dev[i].temperature = upload_vec(media[i].temperature, n);
+            }
+            if (media[i].super) {
+                size_t n = (size_t)media[i].sx * media[i].sy * media[i].sz;
+                media_dev[i].super = upload_vec(media[i].super, n);
             }
         }
         dev_sv.media = upload_vec(media_dev.data(), media_dev.size());

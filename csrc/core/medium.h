@@ -32,6 +32,12 @@ struct alignas(16) MediumParams {
     Vec4 grid_inv_extent;  // 1 / (hi - lo)
     const float* density;      // nx*ny*nz density voxels (nullptr for homogeneous)
     const float* temperature;  // optional emission temperature grid
+    // majorant supergrid: per-supercell RAW density max (8^3 voxels/cell,
+    // dilated by 1 voxel for the stochastic-offset lookup); the tracking
+    // walks DDA over it so null collisions never happen in empty space
+    // (reference uses one global majorant from tree extrema, vol_grid.cu:131)
+    const float* super;
+    int32_t sx, sy, sz;
     int32_t nx, ny, nz;
     int32_t type;
     int32_t phase_id;
@@ -40,7 +46,7 @@ struct alignas(16) MediumParams {
     float scale;           // density multiplier
     float emission_scale;  // blackbody emission brightness
     float temp_scale;      // temperature units -> Kelvin
-    float pad0, pad1;
+    float pad0;
 };
 
 struct MediumSample {
@@ -110,8 +116,89 @@ HD Vec3 homogeneous_transmittance(const MediumParams& m, float dist) {
     return (sig_t * -dist).expv();
 }
 
+// ------------------------------------------------------- majorant supergrid
+// Amanatides-Woo DDA over the supercell grid: next() yields the ray
+// segments [s0, s1) with their LOCAL majorants.  Delta/ratio tracking
+// restart their exponential stepping at each boundary (memoryless), so
+// per-segment majorants are unbiased and empty cells cost one DDA step.
+struct SuperDDA {
+    int cx, cy, cz, sx, sy, sz;
+    int stx, sty, stz;
+    float tmx, tmy, tmz;       // t of next boundary crossing per axis
+    float tdx, tdy, tdz;       // t per cell step per axis
+    float t_cur, t_end;
+    const float* super;
+    float scale;
+
+    HD bool init(const MediumParams& m, const Ray& ray, float t_lim) {
+        if (!m.super) return false;
+        super = m.super; scale = m.scale;
+        sx = m.sx; sy = m.sy; sz = m.sz;
+        Vec3 lo = m.grid_lo.xyz();
+        Vec3 inv_ext = m.grid_inv_extent.xyz();
+        Vec3 inv_d = safe_rcp_dir(ray.d);
+        float t0 = 0.f, t1 = t_lim;
+        Vec3 cw;  // world size of one supercell per axis
+        for (int a = 0; a < 3; ++a) {
+            float ext = 1.f / inv_ext[a];
+            float ta = (lo[a] - ray.o[a]) * inv_d[a];
+            float tb = (lo[a] + ext - ray.o[a]) * inv_d[a];
+            if (ta > tb) { float tt = ta; ta = tb; tb = tt; }
+            t0 = fmaxf(t0, ta);
+            t1 = fminf(t1, tb);
+            cw.set(a, ext / (a == 0 ? (float)sx : a == 1 ? (float)sy : (float)sz));
+        }
+        t_cur = t0; t_end = t1;
+        if (t0 >= t1) { t_end = t_cur; return true; }  // no overlap: empty march
+        Vec3 p0 = ray.at(t0 + fminf(1e-5f, 0.5f * (t1 - t0)));
+        int c[3];
+        float tm[3], td[3];
+        int st[3];
+        for (int a = 0; a < 3; ++a) {
+            int n = a == 0 ? sx : a == 1 ? sy : sz;
+            float g = (p0[a] - lo[a]) / cw[a];
+            c[a] = clampv((int)g, 0, n - 1);
+            if (ray.d[a] > 0.f) {
+                st[a] = 1;
+                tm[a] = (lo[a] + (c[a] + 1) * cw[a] - ray.o[a]) * inv_d[a];
+                td[a] = cw[a] * inv_d[a];
+            } else if (ray.d[a] < 0.f) {
+                st[a] = -1;
+                tm[a] = (lo[a] + c[a] * cw[a] - ray.o[a]) * inv_d[a];
+                td[a] = -cw[a] * inv_d[a];
+            } else {
+                st[a] = 0;
+                tm[a] = MAX_DIST;
+                td[a] = MAX_DIST;
+            }
+        }
+        cx = c[0]; cy = c[1]; cz = c[2];
+        stx = st[0]; sty = st[1]; stz = st[2];
+        tmx = tm[0]; tmy = tm[1]; tmz = tm[2];
+        tdx = td[0]; tdy = td[1]; tdz = td[2];
+        return true;
+    }
+
+    HD bool next(float& s0, float& s1, float& maj) {
+        if (t_cur >= t_end) return false;
+        if (cx < 0 || cx >= sx || cy < 0 || cy >= sy || cz < 0 || cz >= sz) return false;
+        maj = super[(size_t(cz) * sy + cy) * sx + cx] * scale;
+        s0 = t_cur;
+        float tn = fminf(fminf(tmx, tmy), tmz);
+        s1 = fminf(tn, t_end);
+        t_cur = s1;
+        if (tn <= t_end) {
+            if (tmx <= tmy && tmx <= tmz) { cx += stx; tmx += tdx; }
+            else if (tmy <= tmz)          { cy += sty; tmy += tdy; }
+            else                          { cz += stz; tmz += tdz; }
+        }
+        return true;
+    }
+};
+
 // ------------------------------------------------------------- grid medium
-// Delta-tracking distance sampling (reference vol_grid.cu:128-150).
+// Delta-tracking distance sampling (reference vol_grid.cu:128-150), DDA'd
+// over the majorant supergrid when present.
 HD MediumSample grid_sample(const MediumParams& m, const Ray& ray, float t_max, Sampler& sp) {
     MediumSample r{};
     r.local_thp = Vec3(1.f);
@@ -123,6 +210,33 @@ HD MediumSample grid_sample(const MediumParams& m, const Ray& ray, float t_max, 
     Vec3 alb = m.sigma_s.xyz() /
                (m.sigma_s.xyz() + m.sigma_a.xyz()).maxv(Vec3(1e-8f));
     float a_mean = clampv((alb.x + alb.y + alb.z) * (1.f / 3.f), 1e-6f, 1.f);
+    SuperDDA dda;
+    if (dda.init(m, ray, t_max)) {
+        float s0, s1, maj_c;
+        int it = 0;
+        while (dda.next(s0, s1, maj_c)) {
+            if (maj_c <= 0.f) continue;               // empty supercell: free skip
+            float inv_c = 1.f / maj_c;
+            float t = s0;
+            for (; it < 16384; ++it) {
+                t -= logf(fmaxf(1.f - sp.next1f(), 1e-12f)) * inv_c;
+                if (t >= s1) break;
+                float dens = grid_density_at(m, ray.at(t), &sp);
+                if (sp.next1f() < dens * inv_c) {
+                    if (sp.next1f() < a_mean) {
+                        r.local_thp = alb * (1.f / a_mean);
+                        r.dist = t;
+                        r.scattered = true;
+                        return r;
+                    }
+                    r.local_thp = Vec3(0.f);          // absorbed
+                    return r;
+                }
+            }
+        }
+        r.scattered = false;
+        return r;
+    }
     float t = 0.f;
     for (int it = 0; it < 4096; ++it) {
         t -= logf(fmaxf(1.f - sp.next1f(), 1e-12f)) * inv_maj;
@@ -152,6 +266,29 @@ HD MediumSample grid_sample(const MediumParams& m, const Ray& ray, float t_max, 
 HD Vec3 grid_transmittance(const MediumParams& m, const Ray& ray, float dist, Sampler& sp) {
     float maj = m.majorant;
     if (maj <= 0.f) return Vec3(1.f);
+    SuperDDA dda;
+    if (dda.init(m, ray, dist)) {
+        float tr = 1.f;
+        float s0, s1, maj_c;
+        int it = 0;
+        while (dda.next(s0, s1, maj_c)) {
+            if (maj_c <= 0.f) continue;
+            float inv_c = 1.f / maj_c;
+            float t = s0;
+            for (; it < 16384; ++it) {
+                t -= logf(fmaxf(1.f - sp.next1f(), 1e-12f)) * inv_c;
+                if (t >= s1) break;
+                float dens = grid_density_at(m, ray.at(t), &sp);
+                tr *= fmaxf(0.f, 1.f - dens * inv_c);
+                if (tr < 0.1f) {  // Russian roulette termination
+                    if (sp.next1f() >= tr * 10.f) return Vec3(0.f);
+                    tr = 0.1f;
+                }
+                if (tr <= 0.f) return Vec3(0.f);
+            }
+        }
+        return Vec3(tr);
+    }
     float inv_maj = 1.f / maj;
     float tr = 1.f;
     float t = 0.f;
