@@ -19,7 +19,7 @@ import numpy as np
 
 from .scene import (BsdfDesc, CameraDesc, EmitterDesc, MediumDesc, ObjectDesc,
                     RenderConfig, SceneDesc)
-from .obj_loader import load_obj
+from .obj_loader import load_obj, load_obj_multi
 
 RENDERER_MAP = {
     "pt": "pt", "megakernel": "pt", "pt-dynamic": "pt-dyn", "dynamic": "pt-dyn",
@@ -331,9 +331,57 @@ def parse_xml(path: str) -> SceneDesc:
         emitter_ids[eid] = len(d.emitters) - 1
 
     # ---- shapes
+    def bsdf_from_mtl(name: str, mat: dict) -> int:
+        """Map a Wavefront MTL material onto the BSDF matrix (multi-material
+        OBJ hero assets; reference loads materials via tinyobjloader,
+        scene.cu:548-660).  Glass: d<1 with ior; metal-ish: illum>=3 with a
+        strong Ks; else (textured) lambertian."""
+        b = BsdfDesc()
+        if mat.get("d", 1.0) < 0.99:
+            b.type = "translucent"
+            b.ks = mat.get("ks", (1.0, 1.0, 1.0))
+            b.ior = mat.get("ni", 1.5)
+        elif mat.get("illum", 2) >= 3 and max(mat.get("ks", (0,) * 3)) > 0.25:
+            b.type = "ggx"
+            b.kg = mat.get("ks", (0.9, 0.9, 0.9))
+            ns = max(mat.get("ns", 10.0), 1.0)
+            r = float(np.sqrt(2.0 / (ns + 2.0)))  # Blinn exponent -> roughness
+            b.roughness_x = b.roughness_y = r
+        else:
+            b.type = "lambertian"
+            b.kd = mat.get("kd", (0.8, 0.8, 0.8))
+            if mat.get("map_kd"):
+                t = load_tex(os.path.relpath(mat["map_kd"], base))
+                if t >= 0:
+                    b.textures["diffuse"] = t
+        d.bsdfs.append(b)
+        bsdf_ids[f"__mtl__{name}"] = len(d.bsdfs) - 1
+        return len(d.bsdfs) - 1
+
     for el in root.findall("shape"):
         stype = el.get("type", "obj")
         props, refs = _props(el)
+        if stype != "sphere" and props.get("use_mtl", False):
+            # multi-material OBJ: one object per usemtl group, BSDFs from the
+            # .mtl library; explicit <ref type="material"> is the fallback
+            # for faces without a material
+            fn = os.path.normpath(os.path.join(base, props.get("filename", "")))
+            if not os.path.exists(fn):
+                raise FileNotFoundError(f"mesh not found: {fn}")
+            groups, mats = load_obj_multi(fn)
+            for mname, tris, normals, uvs in groups:
+                o = ObjectDesc()
+                o.tris, o.normals, o.uvs = tris, normals, uvs
+                key = f"__mtl__{mname}"
+                if key in bsdf_ids:
+                    o.bsdf = bsdf_ids[key]
+                elif mname in mats:
+                    o.bsdf = bsdf_from_mtl(mname, mats[mname])
+                else:
+                    o.bsdf = bsdf_ids.get(refs.get("material", ""), 0)
+                o.cullable = bool(props.get("cullable", False))
+                d.objects.append(o)
+            continue
         o = ObjectDesc()
         if stype == "sphere":
             c = props.get("center", (0.0, 0.0, 0.0))

@@ -14,7 +14,7 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 GOLD = os.path.join(ROOT, "tests", "golden")
 
 NAMES = ["cornell-box", "balls", "grid-cbox", "diamonds", "env-balls",
-         "point-cbox"]
+         "point-cbox", "hero", "grid-cbox-nvdb"]
 PROC = ["kitchen", "sports-car"]
 
 

@@ -8,6 +8,7 @@ import hippt
 from hippt.scene.xml_parser import parse_xml, parse_rgb
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+SCENES = os.path.join(ROOT, "scenes")
 
 
 def test_parse_rgb():
@@ -147,3 +148,23 @@ def test_spot_cbox_xml():
     d, img = TestExampleScenes()._render("spot-cbox.xml", spp=8)
     assert d.emitters[0].type == "area-spot"
     assert img[..., :3].mean() > 0.005
+
+
+def test_multi_material_obj_hero():
+    """scenes/hero.xml: one OBJ, four usemtl groups; BSDFs derive from the
+    .mtl library (glass via d/Ni, metal via illum+Ks+Ns, textured
+    lambertians via map_Kd) — reference tinyobjloader path scene.cu:548-660."""
+    d = parse_xml(os.path.join(SCENES, "hero.xml"))
+    types = [b.type for b in d.bsdfs]
+    assert "ggx" in types and "translucent" in types
+    # two map_Kd textures loaded and bound to diffuse slots
+    mtl_bsdfs = [b for b in d.bsdfs if b.textures.get("diffuse") is not None]
+    assert len([b for b in mtl_bsdfs if "diffuse" in b.textures]) >= 2
+    assert len(d.textures) >= 2
+    # hero groups became separate objects with normals + uvs
+    hero_objs = [o for o in d.objects if o.tris is not None and o.uvs is not None]
+    assert len(hero_objs) >= 4
+    from hippt.scene.obj_loader import load_obj_multi
+    groups, mats = load_obj_multi(os.path.join(SCENES, "meshes", "hero", "hero.obj"))
+    assert [g[0] for g in groups] == ["ceramic", "metal", "glass", "wood"]
+    assert set(mats) == {"ceramic", "metal", "glass", "wood"}
