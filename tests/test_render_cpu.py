@@ -456,3 +456,100 @@ def test_envmap_light_tracing_agrees_with_pt():
     pt_c = pt_img[16:32, 16:32, :3].mean()
     assert lt_c > 1e-3, "LT from envmap produced a black image"
     assert abs(lt_c - pt_c) < 0.25 * pt_c, (lt_c, pt_c)
+
+
+def test_analytic_rectangle_light_oracle():
+    """Independent correctness oracle: direct lighting
+    from a rectangular uniform luminaire over a lambertian floor has the
+    closed-form Lambert contour integral
+        E = L/2 * sum_i acos(v_i . v_j) * normalize(v_i x v_j) . n,
+    computed HERE in numpy — not derived from any csrc/ math.  The renderer
+    (NEE + emitter-hit MIS + BSDF sampling + area-CDF emitter sampling) must
+    reproduce rho/pi * E.  Black-albedo light + single flat floor means the
+    path integral is exactly the direct term."""
+    from hippt.scene.scene import (SceneDesc, ObjectDesc, BsdfDesc, EmitterDesc,
+                                   CameraDesc, RenderConfig)
+
+    L = 5.0          # emitted radiance
+    rho = 0.6        # floor albedo
+    h = 1.25         # light height
+    half = 0.45      # light half-extent
+
+    # floor at y=0 (two triangles, big), light quad at y=h facing down
+    def quad(p0, p1, p2, p3):
+        return np.array([[p0, p1, p2], [p0, p2, p3]], np.float32)
+
+    d = SceneDesc()
+    d.bsdfs = [BsdfDesc(type="lambertian", kd=(rho,) * 3),
+               BsdfDesc(type="lambertian", kd=(0.0, 0.0, 0.0))]
+    d.emitters = [EmitterDesc(type="area", emission=(L, L, L), scale=1.0)]
+    d.objects = [
+        ObjectDesc(tris=quad((-20, 0, -20), (-20, 0, 20), (20, 0, 20), (20, 0, -20)),
+                   bsdf=0),
+        # wind the light so its shading normal points DOWN (toward the floor)
+        ObjectDesc(tris=quad((-half, h, -half), (-half, h, half),
+                             (half, h, half), (half, h, -half))[:, ::-1].copy(),
+                   bsdf=1, emitter=0),
+    ]
+    d.camera = CameraDesc(pos=(0, 0.8, -2.2), lookat=(0, 0, 0), fov=30,
+                          width=64, height=64)
+    d.config = RenderConfig(renderer="pt", max_depth=3)
+    r = hippt.PythonRenderer(d, device_id=-1)
+    img = r.render(spp=512).numpy()
+
+    def analytic_E(p):
+        verts = np.array([(-half, h, -half), (-half, h, half),
+                          (half, h, half), (half, h, -half)], np.float64)
+        v = verts - np.asarray(p, np.float64)
+        v /= np.linalg.norm(v, axis=1, keepdims=True)
+        n = np.array([0.0, 1.0, 0.0])
+        E = 0.0
+        for i in range(4):
+            a, b = v[i], v[(i + 1) % 4]
+            cr = np.cross(a, b)
+            s = np.linalg.norm(cr)
+            if s < 1e-12:
+                continue
+            # edge winding chosen so E > 0 for a light above the floor
+            E += np.arccos(np.clip(np.dot(a, b), -1, 1)) * np.dot(cr / s, n)
+        return abs(E) * L / 2.0
+
+    # compare a few pixels: project pixel centers onto the floor via the
+    # camera (same pinhole model parameters, re-derived here)
+    cam = d.camera
+    # forward/right/up basis from pos/lookat (independent reimplementation)
+    fwd = np.array(cam.lookat, np.float64) - np.array(cam.pos, np.float64)
+    fwd /= np.linalg.norm(fwd)
+    right = np.cross(fwd, (0, 1, 0)); right /= np.linalg.norm(right)
+    up = np.cross(right, fwd)
+    focal = 0.5 * cam.width / np.tan(np.radians(cam.fov) / 2)
+    for px, py in [(32, 32), (20, 40), (44, 26)]:
+        dir_cam = (px + 0.5 - cam.width / 2) * right + \
+                  (cam.height / 2 - (py + 0.5)) * up + focal * fwd
+        dir_cam /= np.linalg.norm(dir_cam)
+        t = -cam.pos[1] / dir_cam[1]
+        p = np.array(cam.pos) + t * dir_cam
+        expect = rho / np.pi * analytic_E(p)
+        got = img[py, px, :3].mean()
+        assert abs(got - expect) < 0.05 * expect + 0.01, \
+            (px, py, got, expect)
+
+
+def test_convex_sphere_albedo_oracle():
+    """Second independent oracle: a single CONVEX lambertian sphere under a
+    unit-radiance constant environment reflects exactly rho toward the
+    camera (one surface hit, reflected rays cannot re-hit a convex body):
+    L = rho * 1.0 analytically, no renderer math reused."""
+    from hippt.scene.scene import (SceneDesc, ObjectDesc, BsdfDesc, EmitterDesc,
+                                   CameraDesc, RenderConfig)
+    rho = 0.37
+    d = SceneDesc()
+    d.bsdfs = [BsdfDesc(type="lambertian", kd=(rho,) * 3)]
+    d.emitters = [EmitterDesc(type="envmap", emission=(1, 1, 1), scale=1.0)]
+    d.objects = [ObjectDesc(spheres=np.array([[0, 0, 0, 1.0]], np.float32), bsdf=0)]
+    d.camera = CameraDesc(pos=(0, 0, -4), lookat=(0, 0, 0), fov=25,
+                          width=48, height=48, )
+    d.config = RenderConfig(renderer="pt", max_depth=4)
+    img = hippt.PythonRenderer(d, device_id=-1).render(spp=256).numpy()
+    center = img[18:30, 18:30, :3].mean()
+    assert abs(center - rho) < 0.015, (center, rho)
