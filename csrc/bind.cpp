@@ -149,6 +149,14 @@ struct SceneHolder {
 
     int bvh4_depth = 0;
     Vec4 scene_bound{0.f, 0.f, 0.f, 1.f};
+    int cache_nodes = 0;
+
+    void set_cache_level(int level) {
+        // accelerator XML cache_level -> top-tree nodes in LDS (2^level,
+        // the reference's binary top-level count; clamped in the launcher
+        // to the LDS budget)
+        cache_nodes = level > 0 ? (1 << (level > 12 ? 12 : level)) : 0;
+    }
 
     void set_objects(iarr objs) {
         if (objs.ndim() != 2 || objs.shape(1) != 8) throw std::runtime_error("objs must be (n,8)");
@@ -372,6 +380,7 @@ struct SceneHolder {
         sv.md = md;
         sv.bvh4_depth = bvh4_depth;
         sv.scene_bound = scene_bound;
+        sv.cache_nodes = cache_nodes;
     }
 
     void finalize() {
@@ -746,6 +755,7 @@ PYBIND11_MODULE(_C, m) {
              py::arg("max_depth"), py::arg("max_diffuse"), py::arg("max_specular"),
              py::arg("max_transmit"), py::arg("max_volume"), py::arg("min_time"),
              py::arg("max_time"), py::arg("use_tof"), py::arg("radiance_clamp") = 0.f)
+        .def("set_cache_level", &SceneHolder::set_cache_level)
         .def_readwrite("cam_medium", &SceneHolder::cam_medium)
         .def_readwrite("env_emitter", &SceneHolder::env_emitter)
         .def("finalize", &SceneHolder::finalize)

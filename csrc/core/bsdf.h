@@ -154,6 +154,9 @@ HD Vec3 bsdf_eval(const BsdfParams& b, const Vec3& wo, const Vec3& wi,
         Vec3 wo_l = fr.to_local(wo), wi_l = fr.to_local(wi);
         Vec3 wh_l = (wo_l + wi_l).normalized();
         float rx = b.extra0, ry = b.extra1;
+        // NOTE: a roughness texture forces an ISOTROPIC lobe (one channel
+        // drives both axes); anisotropy is constant-parameter only.  Same
+        // limitation as the reference's single roughness slot.
         if (b.tex[TEX_ROUGHNESS] >= 0) { rx = ry = textures[b.tex[TEX_ROUGHNESS]].sample(it.uv).x; }
         float ax = roughness_to_alpha(rx), ay = roughness_to_alpha(ry);
         float D = ggx_d(wh_l, ax, ay);
@@ -189,6 +192,9 @@ HD float bsdf_pdf(const BsdfParams& b, const Vec3& wo, const Vec3& wi, const Int
         Vec3 wo_l = fr.to_local(wo), wi_l = fr.to_local(wi);
         Vec3 wh_l = (wo_l + wi_l).normalized();
         float rx = b.extra0, ry = b.extra1;
+        // NOTE: a roughness texture forces an ISOTROPIC lobe (one channel
+        // drives both axes); anisotropy is constant-parameter only.  Same
+        // limitation as the reference's single roughness slot.
         if (b.tex[TEX_ROUGHNESS] >= 0) { rx = ry = textures[b.tex[TEX_ROUGHNESS]].sample(it.uv).x; }
         float ax = roughness_to_alpha(rx), ay = roughness_to_alpha(ry);
         return ggx_pdf_vndf(wo_l, wh_l, ax, ay) / (4.f * fmaxf(fabsf(wo_l.dot(wh_l)), 1e-7f));
@@ -303,6 +309,9 @@ HD BsdfSample bsdf_sample(const BsdfParams& b, const Vec3& wo, const Interaction
         Frame fr = Frame::from_n(nn);
         Vec3 wo_l = fr.to_local(wo);
         float rx = b.extra0, ry = b.extra1;
+        // NOTE: a roughness texture forces an ISOTROPIC lobe (one channel
+        // drives both axes); anisotropy is constant-parameter only.  Same
+        // limitation as the reference's single roughness slot.
         if (b.tex[TEX_ROUGHNESS] >= 0) { rx = ry = textures[b.tex[TEX_ROUGHNESS]].sample(it.uv).x; }
         float ax = roughness_to_alpha(rx), ay = roughness_to_alpha(ry);
         Vec3 wh_l = ggx_sample_wh(wo_l, ax, ay, sp.next2f());

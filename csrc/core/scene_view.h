@@ -72,6 +72,10 @@ struct SceneView {
     int bvh4_depth;
     // scene bounding sphere (xyz = center, w = radius); envmap sample_le
     Vec4 scene_bound;
+    // LDS top-tree cache size in nodes, from the accelerator XML cache_level
+    // (reference semantics: top 2^level binary nodes cached; here 4-wide
+    // nodes).  0 = launcher default; HIPPT_TOPCACHE env still overrides.
+    int cache_nodes;
 
     HD EmitterGeom emitter_geom() const {
         return {prims, attrs, prim_obj, emitter_prims, emitter_cdf, textures,

@@ -215,11 +215,17 @@ int launch_render(const SceneView& sv, float* accum, float* var,
     // HIPPT_TOPCACHE = nodes of the tree top copied to LDS per block
     // (default 64 = 8 KB); the stack's LDS share shrinks to keep the same
     // per-block LDS budget (occupancy unchanged).
-    static int cache_req = [] {
+    static int cache_env = [] {
         const char* e = getenv("HIPPT_TOPCACHE");
-        return e ? atoi(e) : 64;
+        return e ? atoi(e) : -1;
     }();
-    const int n_cached = cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4;
+    // priority: env A/B hook > accelerator XML cache_level > measured default
+    int cache_req = cache_env >= 0 ? cache_env
+                  : (sv.cache_nodes > 0 ? sv.cache_nodes : 64);
+    int n_cached = cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4;
+    // keep at least 4 LDS stack entries per thread alongside the cache
+    int cache_cap = (20 * 256 * 8 - 4 * 256 * 8) / (int)sizeof(BVH4Node);
+    if (n_cached > cache_cap) n_cached = cache_cap;
     static int lds_budget = [] {
         const char* e = getenv("HIPPT_STACK");
         if (e && strcmp(e, "scratch") == 0) return 0;   // default: lds

@@ -682,10 +682,14 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     }();
     // LDS top-tree cache shares the block budget with the stacks (same
     // occupancy, fewer LDS stack entries; overflow spills to scratch)
-    static int cache_req = [] {
+    static int cache_env = [] {
         const char* e = getenv("HIPPT_TOPCACHE");
-        return e ? atoi(e) : 64;
+        return e ? atoi(e) : -1;
     }();
+    int cache_req = cache_env >= 0 ? cache_env
+                  : (sv.cache_nodes > 0 ? sv.cache_nodes : 64);
+    if (cache_req * (int)sizeof(BVH4Node) > lds_budget - 4 * WF_BLOCK * 8)
+        cache_req = (lds_budget - 4 * WF_BLOCK * 8) / (int)sizeof(BVH4Node);
     const int n_cached = lds_budget > 0
         ? (cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4) : 0;
     const int lds_n = lds_budget > 0

@@ -427,6 +427,9 @@ class Scene:
                                cfg.max_transmit, cfg.max_volume,
                                cfg.min_time, cfg.max_time, int(cfg.use_tof),
                                cfg.radiance_clamp)
+        # accelerator cache_level -> LDS top-tree cache nodes (honored by the
+        # kernel launchers; HIPPT_TOPCACHE env remains the A/B override)
+        self.native.set_cache_level(int(cfg.cache_level))
         self.native.cam_medium = d.cam_medium
         self.native.finalize()
 
