@@ -7,11 +7,14 @@ renderer switching, frame capture, FPS readout.
 
 MI355X-native substitution: the reference renders into a GL PBO via CUDA-GL
 interop under GLFW/ImGui.  MI355X nodes are headless (no GL/display), so the
-idiomatic equivalent is a web viewer: a FastAPI app serves the progressive
-framebuffer as PNG over HTTP plus JSON control endpoints, and a background
-thread keeps accumulating samples (the render loop).  The browser page
-implements WASD/mouse camera and parameter panels.  Everything is testable
-headlessly through the HTTP API (tests/test_viewer.py).
+idiomatic equivalent is a web viewer: a FastAPI app streams the progressive
+framebuffer over a client-paced BINARY WEBSOCKET (/ws/stream: 16-byte
+header + tonemapped RGB, drawn into a canvas via ImageData — no
+per-request PNG encode on the hot path; /frame.png stays for
+capture/fallback) plus JSON control endpoints, and a background thread
+keeps accumulating samples (the render loop).  The browser page implements
+WASD/mouse camera and parameter panels.  Everything is testable headlessly
+through the HTTP API (tests/test_viewer.py).
 """
 from __future__ import annotations
 
@@ -21,6 +24,14 @@ import time
 from typing import Optional
 
 import numpy as np
+
+# module-level so FastAPI can resolve the postponed `ws: WebSocket`
+# annotation (from __future__ import annotations stringifies it and the
+# lookup happens in module globals); guarded: the viewer is optional.
+try:
+    from fastapi import WebSocket
+except Exception:  # pragma: no cover - fastapi always ships in this image
+    WebSocket = None
 
 
 INDEX_HTML = """<!doctype html>
@@ -33,7 +44,7 @@ INDEX_HTML = """<!doctype html>
  .row { margin:4px 0 }
 </style></head>
 <body>
-<div><img id="img" tabindex="0"></div>
+<div><canvas id="cv" tabindex="0"></canvas><img id="img" style="display:none"></div>
 <div id="panel">
   <div class="row">fps: <span id="fps">-</span> spp: <span id="spp">-</span></div>
   <div class="row">renderer:
@@ -63,11 +74,30 @@ INDEX_HTML = """<!doctype html>
   <div class="row">WASD move, QE up/down, arrows look</div>
 </div>
 <script>
-const img = document.getElementById('img');
-function refresh(){ img.src = '/frame.png?' + Date.now(); }
-img.onload = () => setTimeout(refresh, 150);
-img.onerror = () => setTimeout(refresh, 500);
-refresh();
+const cv = document.getElementById('cv');
+const ctx = cv.getContext('2d');
+// client-paced binary stream: ask for a frame, draw it, ask again.
+function connect(){
+  const ws = new WebSocket((location.protocol==='https:'?'wss://':'ws://') +
+                           location.host + '/ws/stream');
+  ws.binaryType = 'arraybuffer';
+  ws.onopen = () => ws.send('next');
+  ws.onmessage = ev => {
+    const dv = new DataView(ev.data);
+    const w = dv.getUint32(0, true), h = dv.getUint32(4, true);
+    if (cv.width !== w) { cv.width = w; cv.height = h; }
+    const rgb = new Uint8Array(ev.data, 16);
+    const id = ctx.createImageData(w, h);
+    for (let i = 0, j = 0; j < rgb.length; i += 4, j += 3) {
+      id.data[i] = rgb[j]; id.data[i+1] = rgb[j+1];
+      id.data[i+2] = rgb[j+2]; id.data[i+3] = 255;
+    }
+    ctx.putImageData(id, 0, 0);
+    ws.send('next');
+  };
+  ws.onclose = () => setTimeout(connect, 800);
+}
+connect();
 setInterval(async () => {
   const s = await (await fetch('/api/stats')).json();
   document.getElementById('fps').textContent = s.fps.toFixed(1);
@@ -147,6 +177,30 @@ class ViewerApp:
             time.sleep(0.0005)
 
     # ------------------------------------------------------------- frames
+    def frame_raw(self, scale: int = 1) -> bytes:
+        """Tonemapped RGB frame as a binary packet: 16-byte header
+        (u32 width, u32 height, u32 spp, u32 reserved, little-endian) +
+        w*h*3 bytes.  This is the websocket streaming path — no PNG/zlib
+        work per frame (the round-1 viewer PNG-encoded every poll)."""
+        import struct
+        from ..utils.png import tonemap
+        with self.lock:
+            r = self.pyr.renderer
+            if self.denoise and getattr(r, "aux", None) is not None:
+                den = r.denoise()
+                den = den.cpu().numpy() if hasattr(den, "cpu") else np.asarray(den)
+                acc = np.concatenate([den, np.ones_like(den[..., :1])], axis=2)
+            else:
+                acc = (r.accum.cpu().numpy()
+                       if r.device is not None else r.accum.copy())
+            spp = r.accum_cnt
+        img = tonemap(acc)[..., :3]
+        if scale > 1:
+            img = img[::scale, ::scale]
+        img = np.ascontiguousarray(img)
+        h, w = img.shape[:2]
+        return struct.pack("<4I", w, h, int(spp), 0) + img.tobytes()
+
     def frame_png(self) -> bytes:
         from ..utils.png import tonemap, write_png
         import tempfile, os
@@ -335,6 +389,20 @@ def build_app(viewer: ViewerApp):
     @app.get("/frame.png")
     def frame():
         return Response(viewer.frame_png(), media_type="image/png")
+
+    @app.websocket("/ws/stream")
+    async def ws_stream(ws: WebSocket):
+        # client-paced: each received message yields one fresh frame, so a
+        # slow client never queues stale frames and a fast one streams at
+        # render speed
+        await ws.accept()
+        try:
+            while True:
+                msg = await ws.receive_text()
+                scale = 2 if msg == "next2" else 1
+                await ws.send_bytes(viewer.frame_raw(scale=scale))
+        except Exception:
+            pass
 
     @app.get("/capture.png")
     def capture():

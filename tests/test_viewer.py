@@ -117,3 +117,21 @@ def test_camera_params(client):
     time.sleep(0.3)
     png = c.get("/frame.png")
     assert png.status_code == 200
+
+
+def test_websocket_stream(client):
+    """Client-paced binary frame streaming: each 'next'
+    yields a 16-byte header + RGB payload; 'next2' halves resolution."""
+    import struct
+    c, viewer = client
+    with c.websocket_connect("/ws/stream") as ws:
+        ws.send_text("next")
+        data = ws.receive_bytes()
+        w, h, spp, _ = struct.unpack_from("<4I", data, 0)
+        acc = viewer.pyr.renderer.accum
+        assert (h, w) == tuple(acc.shape[:2])
+        assert len(data) == 16 + w * h * 3
+        ws.send_text("next2")
+        d2 = ws.receive_bytes()
+        w2, h2, _, _ = struct.unpack_from("<4I", d2, 0)
+        assert w2 == (w + 1) // 2 and len(d2) == 16 + w2 * h2 * 3
