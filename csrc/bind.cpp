@@ -125,6 +125,9 @@ struct SceneHolder {
         bvh4_depth = np_nodes4.ndim() == 2 && np_nodes4.shape(0) > 0
             ? bvh4_tree_depth((const BVH4Node*)np_nodes4.data(), (int)np_nodes4.shape(0))
             : 0;
+        nodes4q = np_nodes4.ndim() == 2 && np_nodes4.shape(0) > 0
+            ? quantize_bvh4((const BVH4Node*)np_nodes4.data(), (int)np_nodes4.shape(0))
+            : std::vector<BVH4NodeQ>();
         // scene bounding sphere (envmap sample_le disk origin)
         const Prim* pr = (const Prim*)np_prims.data();
         const uint32_t* po = np_prim_obj.data();
@@ -150,6 +153,7 @@ struct SceneHolder {
     int bvh4_depth = 0;
     Vec4 scene_bound{0.f, 0.f, 0.f, 1.f};
     int cache_nodes = 0;
+    std::vector<BVH4NodeQ> nodes4q;   // quantized mirror of np_nodes4
 
     void set_cache_level(int level) {
         // accelerator XML cache_level -> top-tree nodes in LDS (2^level,
@@ -387,6 +391,7 @@ struct SceneHolder {
         fill_common(host_sv);
         host_sv.nodes = (const BVHNode*)np_nodes.data();
         host_sv.nodes4 = host_sv.n_nodes4 > 0 ? (const BVH4Node*)np_nodes4.data() : nullptr;
+        host_sv.nodes4q = nodes4q.empty() ? nullptr : nodes4q.data();
         host_sv.nodes8 = host_sv.n_nodes8 > 0 ? (const BVH8Node*)np_nodes8.data() : nullptr;
         host_sv.prims = (const Prim*)np_prims.data();
         host_sv.attrs = (const PrimAttr*)np_attrs.data();
@@ -423,6 +428,8 @@ struct SceneHolder {
         dev_sv.nodes = upload_vec((const BVHNode*)np_nodes.data(), np_nodes.shape(0));
         dev_sv.nodes4 = dev_sv.n_nodes4 > 0
             ? upload_vec((const BVH4Node*)np_nodes4.data(), np_nodes4.shape(0)) : nullptr;
+        dev_sv.nodes4q = nodes4q.empty() ? nullptr
+            : upload_vec(nodes4q.data(), nodes4q.size());
         dev_sv.nodes8 = dev_sv.n_nodes8 > 0
             ? upload_vec((const BVH8Node*)np_nodes8.data(), np_nodes8.shape(0)) : nullptr;
         dev_sv.prims = upload_vec((const Prim*)np_prims.data(), np_prims.shape(0));
@@ -646,6 +653,33 @@ int py_bvh4_selftest(farr prims, uarr prim_obj, farr nodes, farr nodes4,
     return bad;
 }
 
+// Quantized-walk self-test: the Q tree must agree with the fp32 walk on
+// hit/miss, occlusion, and hit distance (quantized boxes are conservative,
+// so the same closest prim must be found).  Returns mismatch count.
+int py_bvh4q_selftest(farr prims, uarr prim_obj, farr nodes4,
+                      farr ray_o, farr ray_d, float tmax) {
+    const BVH4Node* n4 = (const BVH4Node*)nodes4.data();
+    int nn = (int)nodes4.shape(0);
+    std::vector<BVH4NodeQ> q = quantize_bvh4(n4, nn);
+    const Prim* pr = (const Prim*)prims.data();
+    const uint32_t* po = prim_obj.data();
+    int bad = 0;
+    int nr = (int)ray_o.shape(0);
+    for (int i = 0; i < nr; ++i) {
+        Ray r;
+        r.o = {ray_o.at(i, 0), ray_o.at(i, 1), ray_o.at(i, 2)};
+        r.d = {ray_d.at(i, 0), ray_d.at(i, 1), ray_d.at(i, 2)};
+        HitRecord a = ray_intersect_bvh4_ww(n4, pr, po, r, tmax);
+        HitRecord b = ray_intersect_bvh4q_ww(q.data(), pr, po, r, tmax);
+        if ((a.prim_idx < 0) != (b.prim_idx < 0) ||
+            (a.prim_idx >= 0 && fabsf(a.t - b.t) > 1e-5f * fmaxf(1.f, a.t))) { ++bad; continue; }
+        bool oa = occlusion_test_bvh4_ww(n4, pr, po, r, tmax);
+        bool ob = occlusion_test_bvh4q_ww(q.data(), pr, po, r, tmax);
+        if (oa != ob) ++bad;
+    }
+    return bad;
+}
+
 // Closest-hit query over the BVH4 tree: returns (t, prim) arrays for rays.
 py::tuple py_bvh4_hit(farr prims, uarr prim_obj, farr nodes4,
                       farr ray_o, farr ray_d) {
@@ -784,6 +818,9 @@ PYBIND11_MODULE(_C, m) {
     m.def("bvh4_hit", &py_bvh4_hit);
     m.def("bsdf_check", &py_bsdf_check);
     m.def("env_check", &py_env_check);
+    m.def("bvh4q_selftest", &py_bvh4q_selftest,
+          py::arg("prims"), py::arg("prim_obj"), py::arg("nodes4"),
+          py::arg("ray_o"), py::arg("ray_d"), py::arg("tmax"));
     m.def("bvh4_selftest", &py_bvh4_selftest,
           py::arg("prims"), py::arg("prim_obj"), py::arg("nodes"), py::arg("nodes4"),
           py::arg("ray_o"), py::arg("ray_d"), py::arg("tmax"),

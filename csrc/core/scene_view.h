@@ -46,6 +46,7 @@ struct SceneView {
     // the collapse source; traversal runs on the 4-wide tree when present)
     const BVHNode* nodes; int n_nodes;
     const BVH4Node* nodes4; int n_nodes4;
+    const BVH4NodeQ* nodes4q;               // quantized mirror (HIPPT_QBVH)
     const BVH8Node* nodes8; int n_nodes8;   // A/B: used when non-null
     const Prim* prims; const PrimAttr* attrs; const uint32_t* prim_obj; int n_prims;
     const ObjInfo* objs; int n_objs;
@@ -116,12 +117,30 @@ HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
 // Per-thread traversal-stack context: device kernels point lds_slot into a
 // __shared__ array (entry d at lds_slot[d*256]); host passes the default
 // (pure private stack).  See bvh4.h for why the stack lives in LDS.
+// Compile-time traversal-tree selection: HIPPT_QBVH makes the 64-byte
+// quantized node (bvh4.h BVH4NodeQ) the tree every kernel walks — a
+// runtime branch here was measured to cost ~3% even when dead (see the
+// BVH8 note above), so the choice is a build flag, A/B'd by rebuilding.
+#ifdef HIPPT_QBVH
+using TravNode = BVH4NodeQ;
+#else
+using TravNode = BVH4Node;
+#endif
+
 struct TravCtx {
     uint64_t* lds_slot = nullptr;
     int lds_n = 0;
-    const BVH4Node* top_cache = nullptr;  // LDS copy of nodes4[0..n_cached)
+    const TravNode* top_cache = nullptr;  // LDS copy of the tree top
     int n_cached = 0;
 };
+
+HD const TravNode* trav_nodes(const SceneView& sv) {
+#ifdef HIPPT_QBVH
+    return sv.nodes4q;
+#else
+    return sv.nodes4;
+#endif
+}
 
 // Closest hit = the while-while phase-batched walk (measured +25% megakernel
 // / +19% wavefront over the inline-leaf ordered walk, profiles/README.md);
@@ -135,15 +154,25 @@ struct TravCtx {
 // ray_intersect_bvh4_ww32 stays host-tested for the record.)
 HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                              float tmax = MAX_DIST, TravCtx tc = {}) {
+#ifdef HIPPT_QBVH
+    return ray_intersect_bvh4q_ww(sv.nodes4q, sv.prims, sv.prim_obj, ray, tmax,
+                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
+#else
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
+#endif
 }
 // Any-hit also runs the phase-batched form (+1-2% measured over the
 // inline-leaf walk; the inline walk stays for host self-tests).
 HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                        TravCtx tc = {}) {
+#ifdef HIPPT_QBVH
+    return occlusion_test_bvh4q_ww(sv.nodes4q, sv.prims, sv.prim_obj, ray, tmax,
+                                   tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
+#else
     return occlusion_test_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
                                   tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
+#endif
 }
 
 } // namespace hippt

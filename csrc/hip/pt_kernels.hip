@@ -48,11 +48,12 @@ void k_render(SceneView sv, float* __restrict__ accum, float* __restrict__ var,
     // The top of the tree is copied into LDS once per block: every walk's
     // first visits are nodes 0..n_cached, and at a 95% L2 hit rate the
     // bound is hit latency — LDS is ~4x closer.
-    BVH4Node* s_cache = (BVH4Node*)s_stk;
-    uint64_t* s_base = s_stk + (size_t)n_cached * 16;
+    constexpr int NW = (int)(sizeof(TravNode) / 8);   // u64 words per node
+    TravNode* s_cache = (TravNode*)s_stk;
+    uint64_t* s_base = s_stk + (size_t)n_cached * NW;
     const int tid = threadIdx.y * 16 + threadIdx.x;
-    for (int i = tid; i < n_cached * 16; i += 256)
-        ((uint64_t*)s_cache)[i] = ((const uint64_t*)sv.nodes4)[i];
+    for (int i = tid; i < n_cached * NW; i += 256)
+        ((uint64_t*)s_cache)[i] = ((const uint64_t*)trav_nodes(sv))[i];
     if (n_cached > 0) __syncthreads();
     TravCtx tc{&s_base[tid], lds_n, s_cache, n_cached};
     int tile = blockIdx.y * gridDim.x + blockIdx.x;
@@ -224,7 +225,7 @@ int launch_render(const SceneView& sv, float* accum, float* var,
                   : (sv.cache_nodes > 0 ? sv.cache_nodes : 64);
     int n_cached = cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4;
     // keep at least 4 LDS stack entries per thread alongside the cache
-    int cache_cap = (20 * 256 * 8 - 4 * 256 * 8) / (int)sizeof(BVH4Node);
+    int cache_cap = (20 * 256 * 8 - 4 * 256 * 8) / (int)sizeof(TravNode);
     if (n_cached > cache_cap) n_cached = cache_cap;
     static int lds_budget = [] {
         const char* e = getenv("HIPPT_STACK");
@@ -232,8 +233,8 @@ int launch_render(const SceneView& sv, float* accum, float* var,
         return (occ_v <= 3 ? 26 : occ_v == 4 ? 20 : occ_v == 5 ? 16 : 12) * 256 * 8;
     }();
     const int lds_n = lds_budget > 0
-        ? (lds_budget - n_cached * (int)sizeof(BVH4Node)) / (256 * 8) : 0;
-    const uint32_t shmem = (uint32_t)(lds_n * 256 * 8 + n_cached * (int)sizeof(BVH4Node));
+        ? (lds_budget - n_cached * (int)sizeof(TravNode)) / (256 * 8) : 0;
+    const uint32_t shmem = (uint32_t)(lds_n * 256 * 8 + n_cached * (int)sizeof(TravNode));
     const int w = sv.cam.w, h = sv.cam.h;
     if (y1 <= 0 || y1 > h) y1 = h;
     if (y0 < 0) y0 = 0;

@@ -77,10 +77,11 @@ __device__ __forceinline__ uint32_t pack_counts(int d, int s, int t, int b) {
 // (trace alone cost as much as the whole megakernel, profiles/README.md r02).
 __device__ __forceinline__ TravCtx wf_lds_ctx(const SceneView& sv, uint64_t* s_stk,
                                               int lds_n, int n_cached) {
-    BVH4Node* s_cache = (BVH4Node*)s_stk;
-    uint64_t* s_base = s_stk + (size_t)n_cached * 16;
-    for (int i = threadIdx.x; i < n_cached * 16; i += 256)
-        ((uint64_t*)s_cache)[i] = ((const uint64_t*)sv.nodes4)[i];
+    constexpr int NW = (int)(sizeof(TravNode) / 8);   // u64 words per node
+    TravNode* s_cache = (TravNode*)s_stk;
+    uint64_t* s_base = s_stk + (size_t)n_cached * NW;
+    for (int i = threadIdx.x; i < n_cached * NW; i += 256)
+        ((uint64_t*)s_cache)[i] = ((const uint64_t*)trav_nodes(sv))[i];
     if (n_cached > 0) __syncthreads();
     return TravCtx{&s_base[threadIdx.x], lds_n, s_cache, n_cached};
 }
@@ -688,13 +689,13 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     }();
     int cache_req = cache_env >= 0 ? cache_env
                   : (sv.cache_nodes > 0 ? sv.cache_nodes : 64);
-    if (cache_req * (int)sizeof(BVH4Node) > lds_budget - 4 * WF_BLOCK * 8)
-        cache_req = (lds_budget - 4 * WF_BLOCK * 8) / (int)sizeof(BVH4Node);
+    if (cache_req * (int)sizeof(TravNode) > lds_budget - 4 * WF_BLOCK * 8)
+        cache_req = (lds_budget - 4 * WF_BLOCK * 8) / (int)sizeof(TravNode);
     const int n_cached = lds_budget > 0
         ? (cache_req < sv.n_nodes4 ? cache_req : sv.n_nodes4) : 0;
     const int lds_n = lds_budget > 0
-        ? (lds_budget - n_cached * (int)sizeof(BVH4Node)) / (WF_BLOCK * 8) : 0;
-    const uint32_t shmem = (uint32_t)(lds_n * WF_BLOCK * 8 + n_cached * (int)sizeof(BVH4Node));
+        ? (lds_budget - n_cached * (int)sizeof(TravNode)) / (WF_BLOCK * 8) : 0;
+    const uint32_t shmem = (uint32_t)(lds_n * WF_BLOCK * 8 + n_cached * (int)sizeof(TravNode));
     static int wf_dual = [] {
         const char* e = getenv("HIPPT_WF_DUAL");
         return e ? atoi(e) : 0;

@@ -337,7 +337,8 @@ class Scene:
 
         self.native.set_geometry(prims, attrs, prim_obj, nodes, nodes4, nodes8)
         self.native.set_objects(objs)
-        self._np = dict(prims=prims, attrs=attrs, prim_obj=prim_obj, nodes=nodes, objs=objs)
+        self._np = dict(prims=prims, attrs=attrs, prim_obj=prim_obj, nodes=nodes,
+                        nodes4=nodes4, objs=objs)
 
         # ---- emitters (+ per-emitter area CDF over reordered prims)
         eprims: List[int] = []

@@ -43,6 +43,16 @@ def build(verbose=True):
         "-ffast-math", "-fno-finite-math-only",
         "-Wno-unused-result",
     ]
+    if os.environ.get("HIPPT_QBVH"):
+        # quantized 64-byte BVH4 nodes as the traversal tree (A/B build)
+        cflags.append("-DHIPPT_QBVH")
+    # the .o mtime cache is invalid when the flag set changes: stamp it
+    stamp = objdir / ".flags"
+    flags_now = " ".join(cflags)
+    if not stamp.exists() or stamp.read_text() != flags_now:
+        for o in objdir.glob("*.o"):
+            o.unlink()
+        stamp.write_text(flags_now)
     objs = []
     procs = []
     for src in SOURCES:

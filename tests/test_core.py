@@ -299,3 +299,22 @@ try:
             assert C.bvh4_selftest(pr, po, nodes, nodes4, o, d, 1e7) == 0
 except ImportError:  # pragma: no cover
     pass
+
+
+def test_bvh4_quantized_walk_matches_fp32():
+    """The 64-byte quantized node (BVH4NodeQ, HIPPT_QBVH traversal tree)
+    must agree with the fp32 walk on hit/miss, t, and occlusion for random
+    rays over the kitchen geometry (quantized bounds round outward, so the
+    closest hit is identical)."""
+    from hippt.scene.procedural import kitchen
+    from hippt.scene.scene import Scene
+    d = kitchen(width=32, height=32)
+    sc = Scene(d)
+    rng = np.random.default_rng(3)
+    n = 4000
+    o = rng.uniform(-3, 3, (n, 3)).astype(np.float32)
+    dirs = rng.normal(size=(n, 3)).astype(np.float32)
+    dirs /= np.linalg.norm(dirs, axis=1, keepdims=True)
+    bad = C.bvh4q_selftest(sc._np["prims"], sc._np["prim_obj"], sc._np["nodes4"],
+                           o, dirs, 1e7)
+    assert bad == 0, f"{bad}/{n} rays disagree"
