@@ -121,6 +121,12 @@ HD int pick_emitter(const SceneView& sv, Sampler& sp, float& pdf) {
 // quantized node (bvh4.h BVH4NodeQ) the tree every kernel walks — a
 // runtime branch here was measured to cost ~3% even when dead (see the
 // BVH8 note above), so the choice is a build flag, A/B'd by rebuilding.
+// MEASURED AND REJECTED as the default (kitchen 1080p megakernel: 99.6
+// vs 146.5 Msps): the walk is bound by dependent-load LATENCY, not
+// bandwidth, so halving node bytes buys nothing while the uint8
+// decompress (cvt+fma per bound) sits ON the critical path between the
+// node load and the slab tests.  The Q tree stays host-verified
+// (tests/test_core.py) as a recorded experiment.
 #ifdef HIPPT_QBVH
 using TravNode = BVH4NodeQ;
 #else
