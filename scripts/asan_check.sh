@@ -1,0 +1,13 @@
+#!/bin/bash
+# Host-AddressSanitizer pass over the native builders + CPU renderer
+# (beyond the reference's Debug -fsanitize=leak, CMakeLists.txt:17-27).
+# Builds the extension with ASAN, then runs the CPU test suite under the
+# preloaded runtime.  Usage: bash scripts/asan_check.sh [pytest args...]
+set -e
+cd "$(dirname "$0")/.."
+HIPPT_ASAN=1 python setup.py build_ext --inplace
+RT=$(/opt/rocm/bin/hipcc -print-file-name=libclang_rt.asan-x86_64.so)
+LD_PRELOAD=$RT ASAN_OPTIONS=detect_leaks=0 \
+    python -m pytest tests -m "not gpu" -q "${@:--x}"
+# restore the normal build
+python setup.py build_ext --inplace

@@ -46,6 +46,19 @@ def build(verbose=True):
     if os.environ.get("HIPPT_QBVH"):
         # quantized 64-byte BVH4 nodes as the traversal tree (A/B build)
         cflags.append("-DHIPPT_QBVH")
+    ldflags = []
+    if os.environ.get("HIPPT_DEBUG"):
+        # debuggable build (reference parity: Debug -G -g device debug,
+        # CMakeLists.txt:17-27): host -g + device line info, no fast math
+        cflags = [f for f in cflags if f not in ("-O3", "-ffast-math")]
+        cflags += ["-O1", "-g"]
+    if os.environ.get("HIPPT_ASAN"):
+        # host AddressSanitizer for the C++ builders/CPU renderer (beyond
+        # the reference's -fsanitize=leak).  The extension then needs
+        #   LD_PRELOAD=$(hipcc -print-file-name=libclang_rt.asan-x86_64.so)
+        # when imported into a stock (non-ASAN) python.
+        cflags += ["-fsanitize=address", "-shared-libasan"]
+        ldflags += ["-fsanitize=address", "-shared-libasan"]
     # the .o mtime cache is invalid when the flag set changes: stamp it
     stamp = objdir / ".flags"
     flags_now = " ".join(cflags)
@@ -73,7 +86,7 @@ def build(verbose=True):
     for p in procs:
         if p.wait() != 0:
             raise SystemExit(f"hipcc failed ({p.args[2]})")
-    link = [HIPCC, "-shared", "-fPIC", f"--offload-arch={ARCH}", "-o", str(out)] + objs
+    link = [HIPCC, "-shared", "-fPIC", f"--offload-arch={ARCH}", "-o", str(out)] + objs + ldflags
     if verbose:
         print("[hippt build]", " ".join(link), flush=True)
     subprocess.check_call(link)
