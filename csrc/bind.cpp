@@ -267,7 +267,13 @@ struct SceneHolder {
             // 1-voxel dilation (the stochastic-offset lookup samples up to
             // +-0.5 voxel outside the cell); scale is applied in the walk so
             // update_medium's scale changes need no rebuild
-            constexpr int SUP = 8;
+            // HIPPT_SUPER_N = voxels per supercell per axis (A/B: finer
+            // cells skip empty space tighter but cost more DDA steps)
+            static const int SUP = [] {
+                const char* e = getenv("HIPPT_SUPER_N");
+                int v = e ? atoi(e) : 8;
+                return v < 2 ? 2 : (v > 64 ? 64 : v);
+            }();
             if (getenv("HIPPT_NO_SUPER")) {
                 // A/B hook: single global majorant (round-1 behavior)
                 m.sx = m.sy = m.sz = 0;
