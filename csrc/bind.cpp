@@ -271,7 +271,7 @@ struct SceneHolder {
             // cells skip empty space tighter but cost more DDA steps)
             static const int SUP = [] {
                 const char* e = getenv("HIPPT_SUPER_N");
-                int v = e ? atoi(e) : 8;
+                int v = e ? atoi(e) : 16;  // smoke 1080p: sup4 74.4, sup8 84.9, sup16 89.6 Msps
                 return v < 2 ? 2 : (v > 64 ? 64 : v);
             }();
             if (getenv("HIPPT_NO_SUPER")) {
