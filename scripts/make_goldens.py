@@ -14,7 +14,8 @@ OUT = os.path.join(ROOT, "tests", "golden")
 JOBS = [("cornell-box.xml", 48, 24), ("balls.xml", 64, 24),
         ("grid-cbox.xml", 40, 16), ("diamonds.xml", 48, 24),
         ("env-balls.xml", 64, 16), ("point-cbox.xml", 40, 24),
-        ("hero.xml", 64, 16), ("grid-cbox-nvdb.xml", 40, 16)]
+        ("hero.xml", 64, 16), ("grid-cbox-nvdb.xml", 40, 16),
+        ("medium-cbox.xml", 40, 16), ("water-cbox.xml", 40, 16)]
 
 
 def render(name, w, spp):

@@ -14,7 +14,7 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 GOLD = os.path.join(ROOT, "tests", "golden")
 
 NAMES = ["cornell-box", "balls", "grid-cbox", "diamonds", "env-balls",
-         "point-cbox", "hero", "grid-cbox-nvdb"]
+         "point-cbox", "hero", "grid-cbox-nvdb", "medium-cbox", "water-cbox"]
 PROC = ["kitchen", "sports-car"]
 
 
