@@ -22,7 +22,9 @@ def main():
     xml_jobs = [("cornell-box.xml", 512 * m), ("balls.xml", 512 * m), ("grid-cbox.xml", 256 * m),
                 ("diamonds.xml", 512 * m), ("env-balls.xml", 256 * m), ("caustics-lt.xml", 512 * m),
                 ("tof-cbox.xml", 384 * m), ("point-cbox.xml", 256 * m),
-                ("dof-balls.xml", 256 * m), ("spot-cbox.xml", 256 * m)]
+                ("dof-balls.xml", 256 * m), ("spot-cbox.xml", 256 * m),
+                ("hero.xml", 384 * m), ("grid-cbox-nvdb.xml", 256 * m),
+                ("medium-cbox.xml", 256 * m), ("water-cbox.xml", 256 * m)]
     for name, spp in xml_jobs:
         d = parse_xml(os.path.join(ROOT, "scenes", name))
         if small:
