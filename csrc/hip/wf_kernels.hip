@@ -495,12 +495,34 @@ __device__ __forceinline__ void wf_store_ps(WfState& st, int i, const PathState&
     st.ray_d[i] = make_float4(ps.ray.d.x, ps.ray.d.y, ps.ray.d.z, ps.prev_pdf);
     st.thp[i] = make_float4(ps.thp.x, ps.thp.y, ps.thp.z,
                             uint_as_float(ps.prev_delta ? 1u : 0u));
-    float lw = st.L[i].w;
-    st.L[i] = make_float4(ps.L.x, ps.L.y, ps.L.z, lw);
+    st.L[i] = make_float4(ps.L.x, ps.L.y, ps.L.z, 0.f);
     st.prevn[i] = make_float4(ps.prev_n.x, ps.prev_n.y, ps.prev_n.z,
                               uint_as_float(pack_counts(ps.st.n_diffuse, ps.st.n_specular,
                                                         ps.st.n_transmit, ps.b)));
     st.rng[i] = sp.state;
+}
+
+// Shared shade(+trace) core of the fused step/primary kernels: shade the
+// in-register hit, trace the sampled ray, store payload + status.
+__device__ __forceinline__ void wf_step_core(const SceneView& sv, WfState& st, int i,
+                                             PathState& ps, Sampler& sp, TravCtx tc,
+                                             const HitRecord& hit, int do_trace) {
+    bool done = path_shade_hit(sv, ps, sp, tc, hit);
+    if (!done && ps.b >= sv.md.max_depth) done = true;  // path_step entry cap
+    uint32_t status = DEAD;
+    if (!done && do_trace) {
+        HitRecord nh = scene_intersect(sv, ps.ray, MAX_DIST, tc);
+        if (nh.prim_idx < 0) {
+            nh.t = MAX_DIST;
+            path_shade_hit(sv, ps, sp, tc, nh);  // miss branch: envmap + MIS
+        } else {
+            uint32_t oi = sv.prim_obj[nh.prim_idx] & PRIM_OBJ_MASK;
+            status = (uint32_t)(sv.objs[oi].bsdf_id & 0x3F);
+        }
+        st.hit[i] = make_float4(nh.t, nh.u, nh.v, int_as_float(nh.prim_idx));
+    }
+    wf_store_ps(st, i, ps, sp);
+    st.status[i] = (status << 24) | (uint32_t)i;
 }
 
 // ------------------------------------------------------- fused bounce step
@@ -526,22 +548,32 @@ void k_wf_step(SceneView sv, WfState st, const uint32_t* __restrict__ order,
     Sampler sp(0, 0);
     HitRecord hit;
     wf_load_ps(st, i, ps, sp, hit);
-    bool done = path_shade_hit(sv, ps, sp, tc, hit);
-    if (!done && ps.b >= sv.md.max_depth) done = true;  // path_step entry cap
-    uint32_t status = DEAD;
-    if (!done && do_trace) {
-        HitRecord nh = scene_intersect(sv, ps.ray, MAX_DIST, tc);
-        if (nh.prim_idx < 0) {
-            nh.t = MAX_DIST;
-            path_shade_hit(sv, ps, sp, tc, nh);  // miss branch: envmap + MIS
-        } else {
-            uint32_t oi = sv.prim_obj[nh.prim_idx] & PRIM_OBJ_MASK;
-            status = (uint32_t)(sv.objs[oi].bsdf_id & 0x3F);
-        }
-        st.hit[i] = make_float4(nh.t, nh.u, nh.v, int_as_float(nh.prim_idx));
-    }
-    wf_store_ps(st, i, ps, sp);
-    st.status[i] = (status << 24) | (uint32_t)i;
+    wf_step_core(sv, st, i, ps, sp, tc, hit, do_trace);
+}
+
+// ------------------------------------------------------ fused primary step
+// Camera ray + primary hit + bounce-0 shade + bounce-1 trace in ONE kernel
+// (replaces the separate raygen: no payload round trip or sort before the
+// first shade — primary rays are pixel-coherent anyway, and the
+// material-grouped shade was measured neutral).
+template <int MINW = 4>
+__global__ __launch_bounds__(256)
+__attribute__((amdgpu_waves_per_eu(MINW, MINW)))
+void k_wf_primary(SceneView sv, WfState st, int spp_idx, uint32_t seed,
+                  int do_trace, int lds_n, int n_cached) {
+    extern __shared__ uint64_t s_stk[];
+    TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
+    int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= st.n) return;
+    int px = i % st.w, py = i / st.w;
+    Sampler sp(uint32_t(i), uint32_t(spp_idx) * SEED_SCALER + seed);
+    Ray ray = sv.cam.gen_ray(px, py, sp, spp_idx);
+    PathState ps;
+    ps.reset(ray);
+    ps.iter = 1;
+    HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
+    if (hit.prim_idx < 0) hit.t = MAX_DIST;
+    wf_step_core(sv, st, i, ps, sp, tc, hit, do_trace);
 }
 
 // ------------------------------------------------------------- tail fuse
@@ -659,10 +691,17 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return occ < 3 ? 3 : (occ > 6 ? 6 : occ);
     }();
     using RaygenFn = void (*)(SceneView, WfState, int, uint32_t, int, int);
+    using PrimaryFn = void (*)(SceneView, WfState, int, uint32_t, int, int, int);
     using TraceFn = void (*)(SceneView, WfState, const uint32_t*, int, int, int);
     using ShadowFn = void (*)(SceneView, WfState, int, int);
     static RaygenFn f_raygen = occ_v == 3 ? k_wf_raygen<3> : occ_v == 4 ? k_wf_raygen<4>
                              : occ_v == 5 ? k_wf_raygen<5> : k_wf_raygen<6>;
+    static PrimaryFn f_primary = occ_v == 3 ? k_wf_primary<3> : occ_v == 4 ? k_wf_primary<4>
+                               : occ_v == 5 ? k_wf_primary<5> : k_wf_primary<6>;
+    static int prim_fuse = [] {
+        const char* e = getenv("HIPPT_WF_PRIMARY_FUSE");
+        return e ? atoi(e) : 1;
+    }();
     static TraceFn f_trace = occ_v == 3 ? k_wf_trace<3> : occ_v == 4 ? k_wf_trace<4>
                            : occ_v == 5 ? k_wf_trace<5> : k_wf_trace<6>;
     static TraceFn f_tail = occ_v == 3 ? k_wf_tail<3> : occ_v == 4 ? k_wf_tail<4>
@@ -715,8 +754,17 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return e ? atoi(e) : 0;
     }();
     for (int s = 0; s < nspp; ++s) {
-        hipLaunchKernelGGL(f_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed,
-                           lds_n, n_cached);
+        int bounce0;
+        if (prim_fuse && wf_fuse) {
+            // camera ray + primary hit + bounce-0 shade + bounce-1 trace fused
+            hipLaunchKernelGGL(f_primary, grd_n, blk, shmem, hs, sv, *st, spp0 + s,
+                               seed, sv.md.max_depth > 0 ? 1 : 0, lds_n, n_cached);
+            bounce0 = 1;
+        } else {
+            hipLaunchKernelGGL(f_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed,
+                               lds_n, n_cached);
+            bounce0 = 0;
+        }
         // Per bounce: build the compacted material-sorted live view, read the
         // live count back (4 bytes — the price the reference also pays,
         // wf_path_tracer.cu:199), then launch live-sized grids.  Bounce 0
@@ -726,7 +774,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         int prev_live = n;
         uint32_t* order_cur = st->order;
         uint32_t* order_prev = st->order2;
-        for (int bounce = 0; bounce < sv.md.max_depth + 1; ++bounce) {
+        for (int bounce = bounce0; bounce < sv.md.max_depth + 1; ++bounce) {
             const int scan_n = prev_live;
             const int nb = (scan_n + SORT_BLOCK * SORT_ITEMS - 1) / (SORT_BLOCK * SORT_ITEMS);
             (void)hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
