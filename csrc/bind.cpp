@@ -148,11 +148,15 @@ struct SceneHolder {
         Vec3 c = (sb.lo + sb.hi) * 0.5f;
         float rad = n > 0 ? (sb.hi - sb.lo).length() * 0.5f : 1.f;
         scene_bound = Vec4(c, rad);
+        tri_only = 1;
+        for (int i = 0; i < n; ++i)
+            if (po[i] & PRIM_SPHERE_BIT) { tri_only = 0; break; }
     }
 
     int bvh4_depth = 0;
     Vec4 scene_bound{0.f, 0.f, 0.f, 1.f};
     int cache_nodes = 0;
+    int tri_only = 0;
     std::vector<BVH4NodeQ> nodes4q;   // quantized mirror of np_nodes4
 
     void set_cache_level(int level) {
@@ -391,6 +395,7 @@ struct SceneHolder {
         sv.bvh4_depth = bvh4_depth;
         sv.scene_bound = scene_bound;
         sv.cache_nodes = cache_nodes;
+        sv.tri_only = tri_only;
     }
 
     void finalize() {

@@ -38,12 +38,16 @@ static_assert(sizeof(BVH4Node) == 128, "BVH4Node must be 128 bytes");
 
 constexpr int BVH4_STACK = 64;  // >= 3 * max collapsed depth; checked at build
 
-// Intersect prims [base, base+cnt) and tighten rec.
+// Intersect prims [base, base+cnt) and tighten rec.  tri_only (a scene-
+// uniform scalar) skips the per-prim prim_obj sphere-bit load entirely —
+// the reference's TRIANGLE_ONLY compile flag ("10% faster", defines.cuh:
+// 26-27) as a runtime-free SGPR branch.
 HD void bvh4_leaf_hit(const Prim* prims, const uint32_t* prim_obj,
-                      const Ray& ray, int base, int cnt, HitRecord& rec) {
+                      const Ray& ray, int base, int cnt, HitRecord& rec,
+                      bool tri_only = false) {
     for (int k = 0; k < cnt; ++k) {
         int pid = base + k;
-        bool sph = (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
+        bool sph = !tri_only && (prim_obj[pid] & PRIM_SPHERE_BIT) != 0;
         float u, v;
         float t = intersect_prim(prims[pid], sph, ray, u, v);
         if (t > EPSILON && t < rec.t) {
@@ -159,7 +163,7 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
                                    const Ray& ray, float tmax,
                                    uint64_t* lds_slot = nullptr, int lds_n = 0,
                                    const BVH4Node* top_cache = nullptr,
-                                   int n_cached = 0) {
+                                   int n_cached = 0, bool tri_only = false) {
     HitRecord rec;
     rec.t = tmax;
     const Vec3 inv_d = safe_rcp_dir(ray.d);
@@ -234,7 +238,7 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
         // ---- leaf phase: drain consecutive leaf entries
         while (cur >= 0x80000000u) {
             bvh4_leaf_hit(prims, prim_obj, ray, (int)(cur & 0x07ffffffu),
-                          (int)((cur >> 27) & 0xfu), rec);
+                          (int)((cur >> 27) & 0xfu), rec, tri_only);
             for (;;) {
                 if (sp == 0) { cur = DONE; break; }
                 --sp;
@@ -310,7 +314,7 @@ HD bool occlusion_test_bvh4_ww(const BVH4Node* nodes,
                                const Ray& ray, float tmax,
                                uint64_t* lds_slot = nullptr, int lds_n = 0,
                                const BVH4Node* top_cache = nullptr,
-                               int n_cached = 0) {
+                               int n_cached = 0, bool tri_only = false) {
     const Vec3 inv_d = safe_rcp_dir(ray.d);
     const Vec3 o_div = ray.o * inv_d;
     uint64_t stack[BVH4_STACK];

@@ -73,6 +73,9 @@ struct SceneView {
     int bvh4_depth;
     // scene bounding sphere (xyz = center, w = radius); envmap sample_le
     Vec4 scene_bound;
+    // no spheres in the scene: leaf tests skip the per-prim prim_obj load
+    // (reference TRIANGLE_ONLY compile flag as a scene-uniform branch)
+    int tri_only;
     // LDS top-tree cache size in nodes, from the accelerator XML cache_level
     // (reference semantics: top 2^level binary nodes cached; here 4-wide
     // nodes).  0 = launcher default; HIPPT_TOPCACHE env still overrides.
@@ -165,7 +168,8 @@ HD HitRecord scene_intersect(const SceneView& sv, const Ray& ray,
                                   tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
 #else
     return ray_intersect_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                                 tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
+                                 tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached,
+                                 sv.tri_only != 0);
 #endif
 }
 // Any-hit also runs the phase-batched form (+1-2% measured over the
@@ -177,7 +181,8 @@ HD bool scene_occluded(const SceneView& sv, const Ray& ray, float tmax,
                                    tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
 #else
     return occlusion_test_bvh4_ww(sv.nodes4, sv.prims, sv.prim_obj, ray, tmax,
-                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached);
+                                  tc.lds_slot, tc.lds_n, tc.top_cache, tc.n_cached,
+                                  sv.tri_only != 0);
 #endif
 }
 
