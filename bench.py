@@ -151,7 +151,8 @@ def main():
                 "spp_per_step_per_gpu": args.spp_per_step,
                 "renderer": rend.kind,
                 "n_prims": int(r.info()["n_prims"]),
-                "parallelism": f"dp{max(world_size,1)} (sample-split, RCCL all-reduce)",
+                "parallelism": f"dp{max(world_size,1)} (sample-split, "
+                               f"{'RCCL' if backend == 'nccl' else backend} all-reduce)",
             },
         }
         print(json.dumps(out), flush=True)
