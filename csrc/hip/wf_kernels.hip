@@ -506,8 +506,14 @@ __device__ __forceinline__ void wf_store_ps(WfState& st, int i, const PathState&
 // in-register hit, trace the sampled ray, store payload + status.
 __device__ __forceinline__ void wf_step_core(const SceneView& sv, WfState& st, int i,
                                              PathState& ps, Sampler& sp, TravCtx tc,
-                                             const HitRecord& hit, int do_trace) {
+                                             const HitRecord& hit, int extra_bounces,
+                                             int do_trace) {
     bool done = path_shade_hit(sv, ps, sp, tc, hit);
+    // multi-bounce span: keep shading+tracing in registers while live
+    // fractions are high (kitchen stays ~99% live through bounce 6), so
+    // the payload round trip + sort only happen every `span` bounces
+    for (int b = 0; b < extra_bounces && !done; ++b)
+        done = path_step(sv, ps, sp, tc);
     if (!done && ps.b >= sv.md.max_depth) done = true;  // path_step entry cap
     uint32_t status = DEAD;
     if (!done && do_trace) {
@@ -538,7 +544,7 @@ template <int MINW = 4>
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(MINW, MINW)))
 void k_wf_step(SceneView sv, WfState st, const uint32_t* __restrict__ order,
-               int live, int do_trace, int lds_n, int n_cached) {
+               int live, int extra_bounces, int do_trace, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
     int k = blockIdx.x * blockDim.x + threadIdx.x;
@@ -548,7 +554,7 @@ void k_wf_step(SceneView sv, WfState st, const uint32_t* __restrict__ order,
     Sampler sp(0, 0);
     HitRecord hit;
     wf_load_ps(st, i, ps, sp, hit);
-    wf_step_core(sv, st, i, ps, sp, tc, hit, do_trace);
+    wf_step_core(sv, st, i, ps, sp, tc, hit, extra_bounces, do_trace);
 }
 
 // ------------------------------------------------------ fused primary step
@@ -560,7 +566,7 @@ template <int MINW = 4>
 __global__ __launch_bounds__(256)
 __attribute__((amdgpu_waves_per_eu(MINW, MINW)))
 void k_wf_primary(SceneView sv, WfState st, int spp_idx, uint32_t seed,
-                  int do_trace, int lds_n, int n_cached) {
+                  int extra_bounces, int do_trace, int lds_n, int n_cached) {
     extern __shared__ uint64_t s_stk[];
     TravCtx tc = wf_lds_ctx(sv, s_stk, lds_n, n_cached);
     int i = blockIdx.x * blockDim.x + threadIdx.x;
@@ -573,7 +579,7 @@ void k_wf_primary(SceneView sv, WfState st, int spp_idx, uint32_t seed,
     ps.iter = 1;
     HitRecord hit = scene_intersect(sv, ray, MAX_DIST, tc);
     if (hit.prim_idx < 0) hit.t = MAX_DIST;
-    wf_step_core(sv, st, i, ps, sp, tc, hit, do_trace);
+    wf_step_core(sv, st, i, ps, sp, tc, hit, extra_bounces, do_trace);
 }
 
 // ------------------------------------------------------------- tail fuse
@@ -691,7 +697,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return occ < 3 ? 3 : (occ > 6 ? 6 : occ);
     }();
     using RaygenFn = void (*)(SceneView, WfState, int, uint32_t, int, int);
-    using PrimaryFn = void (*)(SceneView, WfState, int, uint32_t, int, int, int);
+    using PrimaryFn = void (*)(SceneView, WfState, int, uint32_t, int, int, int, int);
     using TraceFn = void (*)(SceneView, WfState, const uint32_t*, int, int, int);
     using ShadowFn = void (*)(SceneView, WfState, int, int);
     static RaygenFn f_raygen = occ_v == 3 ? k_wf_raygen<3> : occ_v == 4 ? k_wf_raygen<4>
@@ -706,12 +712,18 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
                            : occ_v == 5 ? k_wf_trace<5> : k_wf_trace<6>;
     static TraceFn f_tail = occ_v == 3 ? k_wf_tail<3> : occ_v == 4 ? k_wf_tail<4>
                           : occ_v == 5 ? k_wf_tail<5> : k_wf_tail<6>;
-    using StepFn = void (*)(SceneView, WfState, const uint32_t*, int, int, int, int);
+    using StepFn = void (*)(SceneView, WfState, const uint32_t*, int, int, int, int, int);
     static StepFn f_step = occ_v == 3 ? k_wf_step<3> : occ_v == 4 ? k_wf_step<4>
                          : occ_v == 5 ? k_wf_step<5> : k_wf_step<6>;
     static int wf_fuse = [] {
         const char* e = getenv("HIPPT_WF_FUSE");
         return e ? atoi(e) : 1;
+    }();
+    static int wf_span = [] {
+        // shades per fused launch between sorts/compactions
+        const char* e = getenv("HIPPT_WF_SPAN");
+        int v = e ? atoi(e) : 1;
+        return v < 1 ? 1 : (v > 8 ? 8 : v);
     }();
     static ShadowFn f_shadow = occ_v == 3 ? k_wf_shadow<3> : occ_v == 4 ? k_wf_shadow<4>
                              : occ_v == 5 ? k_wf_shadow<5> : k_wf_shadow<6>;
@@ -755,11 +767,13 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     }();
     for (int s = 0; s < nspp; ++s) {
         int bounce0;
+        const int n_stage = sv.md.max_depth + 1;   // total shades per path
         if (prim_fuse && wf_fuse) {
-            // camera ray + primary hit + bounce-0 shade + bounce-1 trace fused
+            // camera ray + primary hit + span shades + next trace fused
+            int cnt = wf_span < n_stage ? wf_span : n_stage;
             hipLaunchKernelGGL(f_primary, grd_n, blk, shmem, hs, sv, *st, spp0 + s,
-                               seed, sv.md.max_depth > 0 ? 1 : 0, lds_n, n_cached);
-            bounce0 = 1;
+                               seed, cnt - 1, cnt < n_stage ? 1 : 0, lds_n, n_cached);
+            bounce0 = cnt;
         } else {
             hipLaunchKernelGGL(f_raygen, grd_n, blk, shmem, hs, sv, *st, spp0 + s, seed,
                                lds_n, n_cached);
@@ -774,7 +788,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         int prev_live = n;
         uint32_t* order_cur = st->order;
         uint32_t* order_prev = st->order2;
-        for (int bounce = bounce0; bounce < sv.md.max_depth + 1; ++bounce) {
+        for (int bounce = bounce0; bounce < n_stage; bounce += (wf_fuse ? wf_span : 1)) {
             const int scan_n = prev_live;
             const int nb = (scan_n + SORT_BLOCK * SORT_ITEMS - 1) / (SORT_BLOCK * SORT_ITEMS);
             (void)hipMemsetAsync(st->hist, 0, 256 * sizeof(uint32_t), hs);
@@ -800,8 +814,10 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
                 break;
             }
             if (wf_fuse) {
+                int cnt = wf_span < n_stage - bounce ? wf_span : n_stage - bounce;
                 hipLaunchKernelGGL(f_step, grd_live, blk, shmem, hs, sv, *st, order_cur,
-                                   live, bounce < sv.md.max_depth ? 1 : 0, lds_n, n_cached);
+                                   live, cnt - 1, bounce + cnt < n_stage ? 1 : 0,
+                                   lds_n, n_cached);
             } else {
             (void)hipMemsetAsync(st->sh_cnt, 0, sizeof(int), hs);
             hipLaunchKernelGGL(k_wf_shade, grd_live, blk, 0, hs, sv, *st, order_cur,
