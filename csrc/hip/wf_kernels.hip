@@ -693,7 +693,7 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
     // LDS stack + cache entirely (A/B hook).
     static int occ_v = [] {
         const char* e = getenv("HIPPT_WF_OCC");
-        int occ = e ? atoi(e) : 3;   // measured best for the fused step (r02)
+        int occ = e ? atoi(e) : 4;   // span>=4 ladder: occ4 143-146 vs occ3 136
         return occ < 3 ? 3 : (occ > 6 ? 6 : occ);
     }();
     using RaygenFn = void (*)(SceneView, WfState, int, uint32_t, int, int);
@@ -720,10 +720,15 @@ int launch_render_wavefront(WfState* st, const SceneView& sv, float* accum, floa
         return e ? atoi(e) : 1;
     }();
     static int wf_span = [] {
-        // shades per fused launch between sorts/compactions
+        // Shades per fused launch between sorts/compactions.  The kitchen
+        // live-count curve (~99% live through bounce 6) makes long spans
+        // pay: measured span1 124.5, span4 142.4, span8 145.9 Msps at occ4
+        // — the optimal sort/compaction cadence on this latency-bound
+        // hardware is about once per 8 bounces.  span=1 restores the
+        // classic per-bounce wavefront.
         const char* e = getenv("HIPPT_WF_SPAN");
-        int v = e ? atoi(e) : 1;
-        return v < 1 ? 1 : (v > 8 ? 8 : v);
+        int v = e ? atoi(e) : 8;
+        return v < 1 ? 1 : (v > 16 ? 16 : v);
     }();
     static ShadowFn f_shadow = occ_v == 3 ? k_wf_shadow<3> : occ_v == 4 ? k_wf_shadow<4>
                              : occ_v == 5 ? k_wf_shadow<5> : k_wf_shadow<6>;
