@@ -11,22 +11,28 @@
 // material id} over a 24-bit pixel id, sorted so same-material rays shade
 // together and dead rays compact to the tail.
 //
-// MI355X-native design: the sort is a 3-kernel one-byte counting sort
-// (per-block LDS histograms -> single-block scan -> scatter), entirely
-// memory-bound.  The live count is produced on-device by the scan and read
-// back once per bounce (4 bytes, like the reference's post-sort
-// lower_bound readback, wf_path_tracer.cu:199) so that
-//   * the bounce loop BREAKS the moment live == 0,
-//   * shade/shadow/trace launch live-sized grids instead of full-width
-//     ones, and
-//   * from bounce 1 on, the sort passes GATHER through the previous
-//     bounce's compacted order array — their cost scales with the live
-//     count, not with W*H.
-// Below HIPPT_WF_TAIL live rays (default 1.5M, measured: per-bounce pass
-// overhead outweighs the divergence it avoids well before the chip runs
-// dry), one fused tail kernel finishes every surviving path
-// megakernel-style (integrator.h path_shade_hit/path_step), eliminating
-// all remaining per-bounce sorts and launches.
+// MI355X-native design (round-2 final shape, every step A/B-measured —
+// profiles/README.md):
+//   * The sort is a 3-kernel one-byte counting sort (per-block LDS
+//     histograms -> single-block scan -> scatter), memory-bound; the live
+//     count is produced on-device and read back once per stage (4 bytes,
+//     like the reference's post-sort lower_bound readback,
+//     wf_path_tracer.cu:199).  The pipeline BREAKS at live==0, launches
+//     live-sized grids, and gathers later sorts through the previous
+//     compacted view so sort cost scales with the live count.
+//   * Stages are FUSED: k_wf_primary / k_wf_step run `span` whole
+//     shade+trace bounces in registers (single-source integrator
+//     path_shade_hit/path_step, NEE occlusion inline) between payload
+//     round trips.  The kitchen live-ray curve stays ~99% live through
+//     bounce 6, so per-bounce sorting is pure overhead; the measured
+//     optimum cadence is ~one sort/compaction per 8 bounces
+//     (span1 124.5 -> span8 145.9 Msps, megakernel 149.5; and span8
+//     BEATS span16-no-sort-at-all 144.3 — the compaction at ~40% live
+//     measurably pays).
+//   * Below HIPPT_WF_TAIL live rays (default 1.5M) one fused tail kernel
+//     finishes every surviving path megakernel-style.
+//   * HIPPT_WF_FUSE=0 / HIPPT_WF_SPAN=1 restore the classic split
+//     shade / shadow-queue / trace per-bounce pipeline for A/B.
 #include <hip/hip_runtime.h>
 #include "kernels.h"
 #include <cstdlib>
