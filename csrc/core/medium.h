@@ -32,8 +32,9 @@ struct alignas(16) MediumParams {
     Vec4 grid_inv_extent;  // 1 / (hi - lo)
     const float* density;      // nx*ny*nz density voxels (nullptr for homogeneous)
     const float* temperature;  // optional emission temperature grid
-    // majorant supergrid: per-supercell RAW density max (8^3 voxels/cell,
-    // dilated by 1 voxel for the stochastic-offset lookup); the tracking
+    // majorant supergrid: per-supercell RAW density max (HIPPT_SUPER_N
+    // voxels/cell per axis, default 16; dilated by 1 voxel for the
+    // stochastic-offset lookup); the tracking
     // walks DDA over it so null collisions never happen in empty space
     // (reference uses one global majorant from tree extrema, vol_grid.cu:131)
     const float* super;
