@@ -9,11 +9,12 @@
 // at Tr<0.1 :177-198, residual-ratio variant :153-175, stochastic-offset
 // nearest-neighbor lookup, blackbody emission :97-102).
 //
-// MI355X-native change: instead of NanoVDB trees (long scalar-dependent
-// pointer chases, and no .nvdb assets ship with the reference anyway) we use
-// a dense brick-friendly float grid in a flat device buffer with an explicit
-// majorant, which delta/ratio tracking consume directly; the host layer
-// (hippt/scene/volume.py) loads/creates grids.
+// MI355X-native change: instead of walking NanoVDB trees on-device (long
+// scalar-dependent pointer chases) we use a dense float grid in a flat
+// device buffer plus a majorant SUPERGRID that delta/ratio tracking DDA
+// over; the host layer loads grids from .npy / .nvdb files
+// (hippt/scene/nvdb.py converts NanoVDB trees to dense on the host) or
+// procedural generators.
 #pragma once
 #include "phase.h"
 #include "spectrum.h"
