@@ -318,3 +318,33 @@ def test_bvh4_quantized_walk_matches_fp32():
     bad = C.bvh4q_selftest(sc._np["prims"], sc._np["prim_obj"], sc._np["nodes4"],
                            o, dirs, 1e7)
     assert bad == 0, f"{bad}/{n} rays disagree"
+
+
+def test_sbvh_degenerate_leaves_capped_at_15():
+    """Advisor r01 (medium): SBVH fallbacks (depth cap, degenerate split)
+    used to emit leaves with arbitrary prim counts, silently wrapping mod 16
+    in the 4-bit traversal packing.  100 coincident triangles force every
+    degenerate path; all leaves must now stay <= 15 and traversal must agree
+    with brute force."""
+    n = 100
+    tri = np.array([[0, 0, 0], [1, 0, 0], [0, 1, 0]], np.float32)
+    prims = np.zeros((n, 12), np.float32)
+    prims[:, 0:3] = tri[0]
+    prims[:, 4:7] = tri[1] - tri[0]
+    prims[:, 8:11] = tri[2] - tri[0]
+    prim_obj = np.zeros(n, np.uint32)
+    nodes, order, stats = C.build_bvh(prims, prim_obj, 8, 0.6, True, True, 1.0)
+    # binary leaves carry prim_cnt in hi.w as INT BITS (positive = leaf)
+    cnts = nodes[:, 7].copy().view(np.int32)
+    leaf_cnts = cnts[cnts > 0]
+    assert len(leaf_cnts) > 0 and leaf_cnts.max() <= 15, \
+        f"oversized SBVH leaf: {leaf_cnts.max() if len(leaf_cnts) else 0} prims"
+    nodes4, depth4 = C.collapse_bvh4(nodes)
+    rp = np.ascontiguousarray(prims[order])
+    rpo = np.ascontiguousarray(prim_obj[order])
+    o = np.array([[0.2, 0.2, -1.0], [0.2, 0.2, 1.0], [5, 5, 5]], np.float32)
+    d = np.array([[0, 0, 1], [0, 0, -1], [0, 0, 1]], np.float32)
+    t, p = C.bvh4_hit(rp, rpo, nodes4, o, d)
+    assert p[0] >= 0 and abs(t[0] - 1.0) < 1e-5
+    assert p[1] >= 0 and abs(t[1] - 1.0) < 1e-5
+    assert p[2] < 0
