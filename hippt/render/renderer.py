@@ -146,7 +146,7 @@ class Renderer:
     def aov(self):
         """dict(normal (h,w,3), depth (h,w), albedo (h,w,3)) — means."""
         a = self.aux
-        cnt = a[:, :, 7:8] if self.device is not None else a[:, :, 7:8]
+        cnt = a[:, :, 7:8]
         c = cnt.clip(1e-9, None) if self.device is None else cnt.clamp(min=1e-9)
         return {"normal": a[:, :, 0:3] / c, "depth": a[:, :, 3] / c[:, :, 0],
                 "albedo": a[:, :, 4:7] / c}
@@ -157,7 +157,7 @@ class Renderer:
         if getattr(self, "aux", None) is None:
             raise RuntimeError("call enable_aov() before denoise()")
         from ..utils.denoise import atrous_denoise
-        img = self.raw()[:, :, :3] if self.device is None else self.raw()[:, :, :3]
+        img = self.raw()[:, :, :3]
         g = self.aov()
         return atrous_denoise(img, g["normal"], g["depth"], g["albedo"],
                               iterations=iterations, **kw)
@@ -251,9 +251,9 @@ class Renderer:
         self.accum_cnt = 0
 
     def update_camera(self, **kw):
+        # no device re-upload needed: render_device refreshes the camera
+        # from the holder on every call (hot-reload path)
         self.scene.update_camera(**kw)
-        if self.device is not None:
-            self.scene.native.upload(self.device) if False else None
         self.reset()
 
     def counter(self) -> int:
