@@ -172,8 +172,6 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
     int sp = 0;
     constexpr uint32_t DONE = 0x7fffffffu;
     uint32_t cur = 0;
-    BVH4Node nd_pre;
-    bool use_pre = false;
     for (;;) {
         // ---- node phase: walk internal nodes until a leaf surfaces
         while (cur < 0x80000000u && cur != DONE) {
@@ -181,8 +179,7 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
             // the root; 95% L2 hit rate means the bound is hit LATENCY, and
             // LDS is ~4x closer than L2).  Pointer select, single flat load.
             const BVH4Node* nsrc = (int)cur < n_cached ? top_cache : nodes;
-            const BVH4Node nd = use_pre ? nd_pre : nsrc[cur];
-            use_pre = false;
+            const BVH4Node nd = nsrc[cur];
             uint32_t keys[4];
             int nhit = 0;
 #pragma unroll
@@ -211,27 +208,8 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
                     }
                 }
             }
-            // Software prefetch: resolve the NEAREST child first and start
-            // its node load, then push the far entries while the load is in
-            // flight — the push/cull bookkeeping (~tens of VALU/LDS ops)
-            // overlaps the dependent L2 hit instead of serializing after it.
             uint32_t next = DONE;
-            if (nhit > 0) {
-                int c0 = (int)(keys[0] & 3u);
-                int ch0 = nd.child[c0];
-                int pc0 = nd.cnt[c0];
-                if (!(ch0 < 0 && pc0 == 0))
-                    next = ch0 < 0
-                        ? (0x80000000u | ((uint32_t)pc0 << 27) | (uint32_t)(~ch0))
-                        : (uint32_t)ch0;
-            }
-            BVH4Node pre;
-            bool have_pre = false;
-            if (next != DONE && next < 0x80000000u) {
-                pre = ((int)next < n_cached ? top_cache : nodes)[next];
-                have_pre = true;
-            }
-            for (int k = nhit - 1; k >= 1; --k) {  // far -> near so near pops first
+            for (int k = nhit - 1; k >= 0; --k) {  // far -> near so near pops first
                 int c = (int)(keys[k] & 3u);
                 int ch = nd.child[c];
                 int pc = nd.cnt[c];
@@ -239,16 +217,16 @@ HD HitRecord ray_intersect_bvh4_ww(const BVH4Node* nodes,
                 uint32_t lo = ch < 0
                     ? (0x80000000u | ((uint32_t)pc << 27) | (uint32_t)(~ch))
                     : (uint32_t)ch;
-                uint64_t e = ((uint64_t)(keys[k] & ~3u) << 32) | lo;
-                if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = e;
-                else stack[sp - lds_n] = e;
-                ++sp;
+                if (k == 0) {
+                    next = lo;
+                } else {
+                    uint64_t e = ((uint64_t)(keys[k] & ~3u) << 32) | lo;
+                    if (sp < lds_n) lds_slot[sp * BVH4_LDS_STRIDE] = e;
+                    else stack[sp - lds_n] = e;
+                    ++sp;
+                }
             }
-            if (next != DONE) {
-                cur = next;
-                if (have_pre) { nd_pre = pre; use_pre = true; }
-                continue;
-            }
+            if (next != DONE) { cur = next; continue; }
             for (;;) {  // pop, culling stale subtrees
                 if (sp == 0) { cur = DONE; break; }
                 --sp;
