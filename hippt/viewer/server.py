@@ -126,6 +126,23 @@ function setDepth(){
     body:JSON.stringify({max_depth:+document.getElementById('maxd').value})});
 }
 function capture(){ window.open('/capture.png'); }
+// mouse-drag orbit (reference cpt mouse camera): drag on the canvas
+// sends yaw/pitch steps through the same hot-reload camera API
+let drag = null;
+cv.addEventListener('mousedown', ev => { drag = [ev.clientX, ev.clientY]; });
+window.addEventListener('mouseup', () => { drag = null; });
+window.addEventListener('mousemove', ev => {
+  if (!drag) return;
+  const dx = ev.clientX - drag[0], dy = ev.clientY - drag[1];
+  if (Math.abs(dx) < 8 && Math.abs(dy) < 8) return;
+  drag = [ev.clientX, ev.clientY];
+  const keys = [];
+  if (dx > 0) keys.push('yaw+'); else if (dx < 0) keys.push('yaw-');
+  if (dy > 0) keys.push('pitch-'); else if (dy < 0) keys.push('pitch+');
+  for (const k of keys)
+    fetch('/api/camera/move',{method:'POST',
+      headers:{'Content-Type':'application/json'},body:JSON.stringify({key:k})});
+});
 window.addEventListener('keydown', ev => {
   const map = {w:'w',a:'a',s:'s',d:'d',q:'q',e:'e',
                ArrowLeft:'yaw-',ArrowRight:'yaw+',ArrowUp:'pitch+',ArrowDown:'pitch-'};
