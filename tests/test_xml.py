@@ -3,6 +3,7 @@ grammar: brdf/emitter/shape/sensor/renderer/accelerator elements)."""
 import os
 
 import numpy as np
+import pytest
 
 import hippt
 from hippt.scene.xml_parser import parse_xml, parse_rgb
@@ -168,3 +169,22 @@ def test_multi_material_obj_hero():
     groups, mats = load_obj_multi(os.path.join(SCENES, "meshes", "hero", "hero.obj"))
     assert [g[0] for g in groups] == ["ceramic", "metal", "glass", "wood"]
     assert set(mats) == {"ceramic", "metal", "glass", "wood"}
+
+
+REF = "/root/reference/scene/xml"
+
+
+@pytest.mark.skipif(not os.path.isdir(REF), reason="reference repo not present")
+@pytest.mark.parametrize("name", ["cornell-box", "whiskey", "bunny"])
+def test_reference_scene_xmls_parse_and_render(name):
+    """Grammar parity, proven on the reference's OWN scene files: parse
+    /root/reference/scene/xml/<name>.xml (hex colors, metal presets by name,
+    hflip, accelerator block, relative ../meshes paths, its shipped OBJ
+    assets) and render a small frame.  (vader.xml's mesh is a
+    .MISSING_LARGE_BLOBS placeholder in the reference repo itself.)"""
+    d = parse_xml(os.path.join(REF, name + ".xml"))
+    assert len(d.objects) >= 5
+    d.camera.width = d.camera.height = 48
+    img = hippt.PythonRenderer(d, device_id=-1).render(spp=4).numpy()
+    assert np.isfinite(img).all()
+    assert img[..., :3].mean() > 0.01
