@@ -175,16 +175,19 @@ REF = "/root/reference/scene/xml"
 
 
 @pytest.mark.skipif(not os.path.isdir(REF), reason="reference repo not present")
-@pytest.mark.parametrize("name", ["cornell-box", "whiskey", "bunny"])
+@pytest.mark.parametrize("name", ["cornell-box", "whiskey", "bunny", "balls",
+                                  "env-balls", "medium-cbox", "point"])
 def test_reference_scene_xmls_parse_and_render(name):
     """Grammar parity, proven on the reference's OWN scene files: parse
     /root/reference/scene/xml/<name>.xml (hex colors, metal presets by name,
     hflip, accelerator block, relative ../meshes paths, its shipped OBJ
     assets) and render a small frame.  (vader.xml's mesh is a
-    .MISSING_LARGE_BLOBS placeholder in the reference repo itself.)"""
+    .MISSING_LARGE_BLOBS placeholder in the reference repo itself, as are
+    the meshes of the other 9 scenes — these 7 are every reference scene
+    whose assets ship.)"""
     d = parse_xml(os.path.join(REF, name + ".xml"))
     assert len(d.objects) >= 5
     d.camera.width = d.camera.height = 48
     img = hippt.PythonRenderer(d, device_id=-1).render(spp=4).numpy()
     assert np.isfinite(img).all()
-    assert img[..., :3].mean() > 0.01
+    assert img[..., :3].mean() > (0.001 if name == "point" else 0.01)
