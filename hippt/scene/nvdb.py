@@ -10,7 +10,8 @@ a host-side conversion: sparse tree -> dense array + world-space bounds.
 
 Scope (documented subset, checked loudly):
   * NanoVDB ABI 32.x float grids (LevelSet/FogVolume/Unknown classes),
-  * file codec NONE (uncompressed); ZIP/BLOSC raise with a clear message,
+  * file codecs NONE and ZIP (zlib per-grid blobs); BLOSC raises with a
+    clear message,
   * single-key root tiles (the library default NANOVDB_USE_SINGLE_ROOT_KEY).
 The writer emits the same subset (one grid per file), so any file this
 module writes it also reads back bit-exactly; scripts/make_scenes.py ships
@@ -76,7 +77,7 @@ class NvdbError(ValueError):
 # --------------------------------------------------------------------- write
 def write_nvdb(path, density, voxel_size=1.0, origin=(0, 0, 0),
                grid_name="density", grid_class=GC_FOG,
-               world_origin=(0.0, 0.0, 0.0)):
+               world_origin=(0.0, 0.0, 0.0), codec="none"):
     """Write a dense (nz, ny, nx) float32 array as a single-grid .nvdb file
     (codec NONE).  `origin` is the index-space coordinate of voxel [0,0,0];
     world transform is a uniform scale by `voxel_size` plus a translation by
@@ -270,14 +271,21 @@ def write_nvdb(path, density, voxel_size=1.0, origin=(0, 0, 0),
     meta += struct.pack("<I", len(name_file))
     meta += struct.pack("<4I", n_leaf, n_lower, n_upper, 1)   # node counts (leaf,lower,upper,root)
     meta += struct.pack("<3I", 0, 0, 0)                       # tile counts
-    meta += struct.pack("<HH", CODEC_NONE, 0)
-    meta += struct.pack("<I", _version())
+    # codec/version appended below (depends on the codec argument)
+    codec_id = CODEC_ZIP if codec == "zip" else CODEC_NONE
+    blob = bytes(buf)
+    if codec_id == CODEC_ZIP:
+        import zlib
+        blob = zlib.compress(blob, 6)
+    meta = meta[:168] + struct.pack("<HH", codec_id, 0) + struct.pack("<I", _version())
+    # fileSize field (offset 8) = on-disk blob size (compressed when zipped)
+    meta = meta[:8] + struct.pack("<Q", len(blob)) + meta[16:]
     assert len(meta) == 176, len(meta)
     with open(path, "wb") as f:
-        f.write(struct.pack("<QIHH", MAGIC, _version(), 1, CODEC_NONE))
+        f.write(struct.pack("<QIHH", MAGIC, _version(), 1, codec_id))
         f.write(meta)
         f.write(name_file)
-        f.write(bytes(buf))
+        f.write(blob)
 
 
 # ---------------------------------------------------------------------- read
@@ -373,18 +381,26 @@ def read_nvdb(path):
     magic, version, grid_count, codec = struct.unpack_from("<QIHH", data, 0)
     if magic != MAGIC:
         raise NvdbError(f"{path}: not a NanoVDB file (magic 0x{magic:x})")
-    if codec != CODEC_NONE:
-        raise NvdbError(f"{path}: codec {('NONE','ZIP','BLOSC')[codec]} is not "
-                        "supported — re-export uncompressed (nanovdb_convert "
-                        "without -z/-b)")
+    if codec not in (CODEC_NONE, CODEC_ZIP):
+        raise NvdbError(f"{path}: codec BLOSC is not supported — re-export "
+                        "uncompressed or zlib (nanovdb_convert without -b)")
     grids = []
     off = 16
     for _ in range(grid_count):
         meta = data[off:off + 176]
-        grid_size, _fsize, _key, _nvox = struct.unpack_from("<4Q", meta, 0)
+        grid_size, file_size, _key, _nvox = struct.unpack_from("<4Q", meta, 0)
         name_size, = struct.unpack_from("<I", meta, 136)
         off += 176
         off += name_size
-        grids.append(_read_grid(data[off:off + grid_size]))
-        off += grid_size
+        if codec == CODEC_ZIP:
+            import zlib
+            blob = zlib.decompress(data[off:off + file_size])
+            if len(blob) != grid_size:
+                raise NvdbError(f"{path}: zip blob decompressed to "
+                                f"{len(blob)} bytes, expected {grid_size}")
+            off += file_size
+        else:
+            blob = data[off:off + grid_size]
+            off += grid_size
+        grids.append(_read_grid(blob))
     return grids

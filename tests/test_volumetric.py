@@ -337,3 +337,21 @@ class TestNvdb:
         img = r.render(spp=4).numpy()
         assert np.isfinite(img).all()
         assert img[..., :3].mean() > 0.01
+
+    def test_zip_codec_roundtrip(self, tmp_path):
+        """ZIP-codec .nvdb (zlib per-grid blob) reads back losslessly; BLOSC
+        still raises."""
+        from hippt.scene.nvdb import write_nvdb, read_nvdb, NvdbError
+        rng = np.random.default_rng(4)
+        d = (rng.random((30, 20, 25), np.float32) *
+             (rng.random((30, 20, 25)) > 0.5)).astype(np.float32)
+        p = str(tmp_path / "z.nvdb")
+        write_nvdb(p, d, voxel_size=0.3, codec="zip")
+        g = read_nvdb(p)[0]
+        np.testing.assert_array_equal(g["dense"], d)
+        raw = bytearray(open(p, "rb").read())
+        raw[14] = 2  # BLOSC
+        bad = str(tmp_path / "b.nvdb")
+        open(bad, "wb").write(bytes(raw))
+        with pytest.raises(NvdbError, match="BLOSC"):
+            read_nvdb(bad)
