@@ -394,7 +394,10 @@ def read_nvdb(path):
         off += name_size
         if codec == CODEC_ZIP:
             import zlib
-            blob = zlib.decompress(data[off:off + file_size])
+            try:
+                blob = zlib.decompress(data[off:off + file_size])
+            except zlib.error as e:
+                raise NvdbError(f"{path}: corrupt ZIP grid blob ({e})")
             if len(blob) != grid_size:
                 raise NvdbError(f"{path}: zip blob decompressed to "
                                 f"{len(blob)} bytes, expected {grid_size}")

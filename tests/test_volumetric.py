@@ -314,9 +314,9 @@ class TestNvdb:
         open(bad, "wb").write(b"\x00" * 8 + bytes(raw[8:]))
         with pytest.raises(NvdbError, match="magic"):
             read_nvdb(bad)
-        raw[14] = 1  # codec = ZIP
+        raw[14] = 1  # codec byte says ZIP but the blob is raw -> corrupt
         open(bad, "wb").write(bytes(raw))
-        with pytest.raises(NvdbError, match="codec|ZIP"):
+        with pytest.raises(NvdbError, match="ZIP|corrupt"):
             read_nvdb(bad)
 
     def test_grid_cbox_nvdb_scene_renders(self):
