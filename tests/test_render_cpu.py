@@ -553,3 +553,38 @@ def test_convex_sphere_albedo_oracle():
     img = hippt.PythonRenderer(d, device_id=-1).render(spp=256).numpy()
     center = img[18:30, 18:30, :3].mean()
     assert abs(center - rho) < 0.015, (center, rho)
+
+
+def test_analytic_sphere_light_oracle():
+    """Third independent oracle: a uniformly emitting SPHERE over a
+    lambertian floor.  Directly below the sphere center the irradiance has
+    the closed form E = L * pi * sin^2(theta_max), sin(theta_max) = R/d, so
+    the pixel radiance is rho * L * (R/d)^2 — numpy-only expectation.
+    Exercises the sphere-primitive emitter path (uniform-surface-area
+    sampling + solid-angle pdf) that the rectangle oracle does not."""
+    from hippt.scene.scene import (SceneDesc, ObjectDesc, BsdfDesc, EmitterDesc,
+                                   CameraDesc, RenderConfig)
+    L, rho, R, h = 4.0, 0.55, 0.3, 1.6
+
+    def quad(p0, p1, p2, p3):
+        return np.array([[p0, p1, p2], [p0, p2, p3]], np.float32)
+
+    d = SceneDesc()
+    d.bsdfs = [BsdfDesc(type="lambertian", kd=(rho,) * 3),
+               BsdfDesc(type="lambertian", kd=(0.0, 0.0, 0.0))]
+    d.emitters = [EmitterDesc(type="area", emission=(L, L, L), scale=1.0)]
+    d.objects = [
+        ObjectDesc(tris=quad((-20, 0, -20), (-20, 0, 20), (20, 0, 20), (20, 0, -20)),
+                   bsdf=0),
+        ObjectDesc(spheres=np.array([[0.0, h, 0.0, R]], np.float32),
+                   bsdf=1, emitter=0),
+    ]
+    # camera straight down at the origin (the point below the sphere center)
+    d.camera = CameraDesc(pos=(0.0, 0.9, -1.8), lookat=(0, 0, 0), fov=25,
+                          width=48, height=48)
+    d.config = RenderConfig(renderer="pt", max_depth=3)
+    img = hippt.PythonRenderer(d, device_id=-1).render(spp=768).numpy()
+    # the lookat point projects to the image center
+    got = img[23:26, 23:26, :3].mean()
+    expect = rho * L * (R / h) ** 2
+    assert abs(got - expect) < 0.06 * expect + 0.005, (got, expect)
