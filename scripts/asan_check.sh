@@ -7,7 +7,10 @@ set -e
 cd "$(dirname "$0")/.."
 HIPPT_ASAN=1 python setup.py build_ext --inplace
 RT=$(/opt/rocm/bin/hipcc -print-file-name=libclang_rt.asan-x86_64.so)
+# golden-image tests are excluded: they assert near-bit-exact output and
+# the ASAN build legitimately perturbs FP contraction (one flipped MC
+# branch diverges a whole sample) — every other test must pass clean.
 LD_PRELOAD=$RT ASAN_OPTIONS=detect_leaks=0 \
-    python -m pytest tests -m "not gpu" -q "${@:--x}"
+    python -m pytest tests -m "not gpu" --ignore=tests/test_golden.py -q "${@:--x}"
 # restore the normal build
 python setup.py build_ext --inplace
