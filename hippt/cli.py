@@ -28,6 +28,8 @@ def main(argv=None):
     ap.add_argument("--width", type=int, default=None)
     ap.add_argument("--height", type=int, default=None)
     ap.add_argument("--gamma", type=float, default=2.1)
+    ap.add_argument("--exposure", type=float, default=1.0,
+                    help="radiance scale before gamma (bright scenes)")
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("--colormap", default="plasma", help="depth/bvh-cost false color")
     ap.add_argument("--variance", default=None, help="also write the variance map here")
@@ -102,7 +104,7 @@ def main(argv=None):
         acc = np.concatenate([den, np.ones_like(den[..., :1])], axis=2)
         write_png(args.output, tonemap(acc, gamma=args.gamma))
     else:
-        r.save(args.output, gamma=args.gamma)
+        r.renderer.save(args.output, gamma=args.gamma, exposure=args.exposure)
     print(f"[hippt] wrote {args.output} ({time.perf_counter() - t0:.1f}s total)")
 
     if args.variance:

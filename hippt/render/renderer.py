@@ -262,10 +262,10 @@ class Renderer:
     def avg_frame_time(self) -> float:
         return self.timer.avg()
 
-    def save(self, path: str, gamma: float = 2.1):
+    def save(self, path: str, gamma: float = 2.1, exposure: float = 1.0):
         acc = self.accum.cpu().numpy() if self.device is not None else self.accum
-        write_png(path, tonemap(acc, gamma))
+        write_png(path, tonemap(acc, gamma, exposure))
 
-    def image(self, gamma: float = 2.1) -> np.ndarray:
+    def image(self, gamma: float = 2.1, exposure: float = 1.0) -> np.ndarray:
         acc = self.accum.cpu().numpy() if self.device is not None else self.accum
-        return tonemap(acc, gamma)
+        return tonemap(acc, gamma, exposure)

@@ -110,10 +110,13 @@ def read_png(path: str) -> np.ndarray:
     return out
 
 
-def tonemap(accum: np.ndarray, gamma: float = 2.1) -> np.ndarray:
+def tonemap(accum: np.ndarray, gamma: float = 2.1,
+            exposure: float = 1.0) -> np.ndarray:
     """accum: (h,w,4) radiance sums + counts -> (h,w,3) uint8 (reference
-    DeviceImage::export_cpu + to_int gamma 1/2.1)."""
+    DeviceImage::export_cpu + to_int gamma 1/2.1).  `exposure` scales
+    radiance before gamma (bright-light scenes like the reference's
+    bunny.xml, scaler 50, clip at exposure 1)."""
     cnt = np.maximum(accum[:, :, 3:4], 1e-9)
-    rgb = accum[:, :, :3] / cnt
+    rgb = accum[:, :, :3] / cnt * exposure
     rgb = np.clip(rgb, 0.0, None) ** (1.0 / gamma)
     return (np.clip(rgb, 0.0, 1.0) * 255.0 + 0.5).astype(np.uint8)
