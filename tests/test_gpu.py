@@ -316,3 +316,31 @@ def test_envmap_light_tracing_gpu():
     pt_c = pt[22:42, 22:42, :3].mean()
     assert lt_c > 1e-3, "GPU LT from envmap produced a black image"
     assert abs(lt_c - pt_c) < 0.25 * pt_c, (lt_c, pt_c)
+
+
+def test_wavefront_split_pipeline_env():
+    """The classic per-bounce split pipeline (HIPPT_WF_FUSE=0: shade /
+    shadow-queue / trace kernels + per-bounce sort) stays correct — it is
+    the measured A/B baseline for the fused-span design.  Env is read at
+    first launch, so run in a subprocess."""
+    import os
+    import subprocess
+    import sys
+    code = (
+        "import hippt, numpy as np\n"
+        "from hippt.scene.procedural import kitchen\n"
+        "d = kitchen(width=192, height=108, renderer='wfpt')\n"
+        "r = hippt.PythonRenderer(d, device_id=0)\n"
+        "img = r.render(spp=8).cpu().numpy()\n"
+        "assert np.isfinite(img).all() and img[..., :3].mean() > 0.01\n"
+        "d2 = kitchen(width=192, height=108, renderer='pt')\n"
+        "p = hippt.PythonRenderer(d2, device_id=0).render(spp=8).cpu().numpy()\n"
+        "ratio = img[..., :3].mean() / p[..., :3].mean()\n"
+        "assert 0.9 < ratio < 1.1, ratio\n"
+        "print('split ok', ratio)\n")
+    env = dict(os.environ, HIPPT_WF_FUSE="0", HIPPT_WF_SPAN="1")
+    out = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                         text=True, timeout=240, env=env,
+                         cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-1500:]
+    assert "split ok" in out.stdout
