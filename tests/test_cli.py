@@ -53,3 +53,13 @@ def test_cli_depth_false_color(tmp_path):
     img = read(out)
     # false-colored depth: non-trivial chroma
     assert img[..., :3].std() > 0.01
+
+
+def test_soak_script_cpu_smoke():
+    """scripts/soak.py (mixed-workload stability driver) runs its CPU mode:
+    renderer cycling + hot-reload churn + checkpoint round trips."""
+    r = subprocess.run([sys.executable, "scripts/soak.py", "--minutes", "0.05",
+                       "--cpu", "--width", "64", "--height", "36"],
+                      capture_output=True, text=True, timeout=300, cwd=ROOT)
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+    assert "OK" in r.stdout
