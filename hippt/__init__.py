@@ -7,7 +7,7 @@ BSDF system, Mitsuba-like XML scenes, a PythonRenderer returning PyTorch-ROCm
 tensors, and multi-GPU sample-split DDP rendering over RCCL/xGMI.
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 import os
 
