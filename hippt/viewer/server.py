@@ -47,6 +47,13 @@ INDEX_HTML = """<!doctype html>
 <div><canvas id="cv" tabindex="0"></canvas><img id="img" style="display:none"></div>
 <div id="panel">
   <div class="row">fps: <span id="fps">-</span> spp: <span id="spp">-</span></div>
+  <div class="row">scene:
+    <select id="scene" onchange="fetch('/api/scene',{method:'POST',
+      headers:{'Content-Type':'application/json'},
+      body:JSON.stringify({scene:this.value})})">
+      <option>cornell</option><option>kitchen</option>
+      <option>sports-car</option><option>smoke</option>
+    </select></div>
   <div class="row">renderer:
     <select id="renderer">
       <option>pt</option><option>pt-dyn</option><option>wfpt</option><option>vpt</option>
