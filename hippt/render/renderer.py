@@ -18,6 +18,10 @@ from .. import C
 from ..scene.scene import Scene
 from ..utils.png import tonemap, write_png
 
+# On the CPU reference path, "wfpt" and "pt-dyn" run the same single-source
+# integrator as "pt" (the wavefront pipeline and persistent scheduler are
+# GPU execution strategies, not different estimators) — GPU wfpt numerics
+# tests compare against exactly this estimator.
 RENDERER_IDS = {
     "pt": C.R_MEGAKERNEL_PT,
     "pt-dyn": C.R_MEGAKERNEL_PT_DYN,  # persistent-tile scheduler (GPU)
