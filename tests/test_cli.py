@@ -63,3 +63,19 @@ def test_soak_script_cpu_smoke():
                       capture_output=True, text=True, timeout=300, cwd=ROOT)
     assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
     assert "OK" in r.stdout
+
+
+def test_gallery_script_all_scenes_cpu(tmp_path):
+    """render_gallery --small --cpu pushes EVERY shipped scene (14 XMLs +
+    2 procedural) through parse -> build -> render -> PNG in one pass —
+    the broadest end-to-end sweep in the suite (~11 s)."""
+    out = str(tmp_path / "gal")
+    r = subprocess.run([sys.executable, "scripts/render_gallery.py", out,
+                        "--cpu", "--small"],
+                       capture_output=True, text=True, timeout=600, cwd=ROOT)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    pngs = [f for f in os.listdir(out) if f.endswith(".png")]
+    assert len(pngs) >= 16, pngs
+    for f in pngs:
+        img = read(os.path.join(out, f))
+        assert np.isfinite(img).all(), f
