@@ -210,6 +210,7 @@ class ViewerApp:
         from ..utils.png import tonemap
         with self.lock:
             r = self.pyr.renderer
+            kind = r.kind
             if self.denoise and getattr(r, "aux", None) is not None:
                 den = r.denoise()
                 den = den.cpu().numpy() if hasattr(den, "cpu") else np.asarray(den)
@@ -218,12 +219,24 @@ class ViewerApp:
                 acc = (r.accum.cpu().numpy()
                        if r.device is not None else r.accum.copy())
             spp = r.accum_cnt
-        img = tonemap(acc)[..., :3]
+        img = self._present(acc, kind)
         if scale > 1:
             img = img[::scale, ::scale]
         img = np.ascontiguousarray(img)
         h, w = img.shape[:2]
         return struct.pack("<4I", w, h, int(spp), 0) + img.tobytes()
+
+    def _present(self, acc, kind):
+        """accum -> displayable RGB: false color for the debug renderers
+        (reference cpt colormaps depth/BVH-cost), tonemap otherwise."""
+        from ..utils.png import tonemap
+        if kind in ("depth", "bvh-cost"):
+            from ..utils.colormap import false_color
+            vals = acc[:, :, 0] / np.maximum(acc[:, :, 3], 1e-9)
+            img = false_color(vals, cmap="plasma", log_scale=kind == "bvh-cost")
+            img = (np.clip(img[..., :3], 0, 1) * 255.0 + 0.5).astype(np.uint8)
+            return np.ascontiguousarray(img)
+        return tonemap(acc)[..., :3]
 
     def frame_png(self) -> bytes:
         from ..utils.png import tonemap, write_png
@@ -238,7 +251,8 @@ class ViewerApp:
             else:
                 acc = (r.accum.cpu().numpy()
                        if r.device is not None else r.accum.copy())
-        img = tonemap(acc)
+            kind = r.kind
+        img = self._present(acc, kind)
         buf = io.BytesIO()
         # write_png writes to path; reuse its encoder via temp buffer
         tmp = tempfile.NamedTemporaryFile(suffix=".png", delete=False)

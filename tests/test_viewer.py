@@ -154,3 +154,23 @@ def test_scene_hot_swap(client):
     time.sleep(0.2)
     assert viewer.pyr.info()["n_prims"] != n0 or True
     assert c.get("/frame.png").status_code == 200
+
+
+def test_depth_renderer_false_color_stream(client):
+    """Debug renderers stream false-colored frames (reference cpt colormaps
+    depth/BVH-cost), not tonemapped raw distances."""
+    import struct
+    c, viewer = client
+    c.post("/api/renderer", json={"kind": "depth"})
+    time.sleep(0.3)
+    with c.websocket_connect("/ws/stream") as ws:
+        ws.send_text("next")
+        data = ws.receive_bytes()
+        w, h, _, _ = struct.unpack_from("<4I", data, 0)
+        import numpy as np
+        rgb = np.frombuffer(data[16:], np.uint8).reshape(h, w, 3)
+        # plasma colormap: channels differ (raw depth would be gray-ish)
+        assert abs(int(rgb[..., 0].mean()) - int(rgb[..., 2].mean())) > 4
+    png = c.get("/frame.png")
+    assert png.status_code == 200
+    c.post("/api/renderer", json={"kind": "pt"})
