@@ -372,7 +372,8 @@ class Scene:
                 if etype == EM_AREA_SPOT:
                     aux[3] = e.cos_max
             self.native.add_emitter(etype, list(e.emission), e.scale, aux, obj_id,
-                                    e.tex_id, prim_base, prim_cnt, inv_area)
+                                    -1 if e.tex_id is None else e.tex_id,
+                                    prim_base, prim_cnt, inv_area)
         self.native.set_emitter_prims(np.asarray(eprims, np.int32),
                                       np.asarray(ecdf, np.float32))
 
