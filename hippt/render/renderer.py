@@ -60,6 +60,12 @@ class Renderer:
         self.spec_constraint = cfg.spec_constraint
         self.caustic_scaling = cfg.caustic_scaling
         self.bidirectional = cfg.bidirectional or self.kind == "bdpt"
+        if cfg.use_tof and self.kind not in ("pt", "pt-dyn", "vpt"):
+            # wavefront / light tracing do not track path time; gating
+            # min_time > 0 there would silently produce black frames
+            raise ValueError(
+                f"ToF (transient) rendering supports pt/pt-dyn/vpt, not "
+                f"{self.kind!r} (reference scope: vpt, megakernel_vpt.cu)")
         self.accum_cnt = 0
         self.timer = FrameTimer()
         self.device = device

@@ -639,3 +639,15 @@ def test_textured_area_emitter_nee_agrees():
     # same mean power -> same mean image within MC noise
     assert abs(m - mr) < 0.12 * mr, (m, mr)
     assert np.isfinite(img).all()
+
+
+def test_tof_refuses_untracked_renderers():
+    """use_tof with a renderer that does not track path time must fail
+    loudly instead of gating everything to black."""
+    from hippt.scene.procedural import cornell_box
+    d = cornell_box(width=16, height=16, renderer="wfpt")
+    d.config.use_tof = True
+    d.config.min_time = 2.0
+    d.config.max_time = 4.0
+    with pytest.raises(ValueError, match="ToF"):
+        hippt.PythonRenderer(d, device_id=-1)
