@@ -176,6 +176,14 @@ class Renderer:
         """Accumulate spp more samples (reference render_raw semantics).
         y0/y1 restrict rendering to the row band [y0, y1) — tile-split DP
         (megakernel renderers only; 0,0 = full frame)."""
+        if self.rid not in (C.R_MEGAKERNEL_PT, C.R_MEGAKERNEL_PT_DYN, C.R_VOLUME_PT,
+                            C.R_DEPTH, C.R_BVH_COST):
+            # frame-wide pipelines (wavefront, light tracing) would silently
+            # ignore per-pixel budgets / row bands — refuse instead
+            if spp_map is not None:
+                raise ValueError(f"spp_map (adaptive) unsupported for {self.kind!r}")
+            if (y0, y1) != (0, 0):
+                raise ValueError(f"row-band rendering unsupported for {self.kind!r}")
         t0 = time.perf_counter()
         if self.device is not None:
             stream = self.torch.cuda.current_stream().cuda_stream
