@@ -331,6 +331,19 @@ def parse_xml(path: str) -> SceneDesc:
         emitter_ids[eid] = len(d.emitters) - 1
 
     # ---- shapes
+
+    def resolve_ref(refs, kind, table):
+        """Loud ref resolution: a PRESENT-but-unknown id is a scene bug
+        (silently binding material 0 was round-1-class config divergence);
+        an absent ref keeps the documented default."""
+        rid = refs.get(kind, "")
+        if not rid:
+            return None
+        if rid not in table:
+            raise KeyError(f"unknown {kind} ref id '{rid}' "
+                           f"(known: {sorted(table)})")
+        return table[rid]
+
     def bsdf_from_mtl(name: str, mat: dict) -> int:
         """Map a Wavefront MTL material onto the BSDF matrix (multi-material
         OBJ hero assets; reference loads materials via tinyobjloader,
@@ -378,7 +391,8 @@ def parse_xml(path: str) -> SceneDesc:
                 elif mname in mats:
                     o.bsdf = bsdf_from_mtl(mname, mats[mname])
                 else:
-                    o.bsdf = bsdf_ids.get(refs.get("material", ""), 0)
+                    rb = resolve_ref(refs, "material", bsdf_ids)
+                    o.bsdf = 0 if rb is None else rb
                 o.cullable = bool(props.get("cullable", False))
                 d.objects.append(o)
             continue
@@ -393,11 +407,14 @@ def parse_xml(path: str) -> SceneDesc:
                 raise FileNotFoundError(f"mesh not found: {fn}")
             tris, normals, uvs = load_obj(fn)
             o.tris, o.normals, o.uvs = tris, normals, uvs
-        o.bsdf = bsdf_ids.get(refs.get("material", ""), 0)
-        if "emitter" in refs:
-            o.emitter = emitter_ids[refs["emitter"]]
-        if "medium" in refs:
-            o.medium_in = medium_ids[refs["medium"]]
+        rb = resolve_ref(refs, "material", bsdf_ids)
+        o.bsdf = 0 if rb is None else rb
+        re_ = resolve_ref(refs, "emitter", emitter_ids)
+        if re_ is not None:
+            o.emitter = re_
+        rm = resolve_ref(refs, "medium", medium_ids)
+        if rm is not None:
+            o.medium_in = rm
         o.cullable = bool(props.get("cullable", False))
         d.objects.append(o)
 

@@ -128,15 +128,18 @@ class TestParserRobustness:
         with _pt.raises(FileNotFoundError):
             parse_xml(str(p))
 
-    def test_unknown_refs_dont_crash(self, tmp_path):
+    def test_unknown_refs_raise_loudly(self, tmp_path):
+        # round-2 contract change: a present-but-unknown ref id raises with
+        # the known ids listed (round 1 silently bound material 0 — the
+        # same silent-config-divergence class the cache_level fix closed)
         p = tmp_path / "r.xml"
         p.write_text("""<scene version='1.2'>
           <brdf type='lambertian' id='w'><rgb name='k_d' value='0.5'/></brdf>
           <shape type='sphere'><point name='center' value='0,0,0'/>
             <float name='radius' value='1'/>
             <ref type='material' id='does-not-exist'/></shape></scene>""")
-        d = parse_xml(str(p))   # falls back to bsdf 0
-        assert len(d.objects) == 1 and d.objects[0].bsdf == 0
+        with pytest.raises(KeyError, match="does-not-exist"):
+            parse_xml(str(p))
 
 
 def test_dof_balls_xml():
@@ -191,3 +194,4 @@ def test_reference_scene_xmls_parse_and_render(name):
     img = hippt.PythonRenderer(d, device_id=-1).render(spp=4).numpy()
     assert np.isfinite(img).all()
     assert img[..., :3].mean() > (0.001 if name == "point" else 0.01)
+
